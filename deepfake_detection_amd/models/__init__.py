@@ -1,0 +1,24 @@
+"""Model zoo (registry + factory + architectures)."""
+
+from .registry import (  # noqa: F401
+    register_model,
+    list_models,
+    is_model,
+    model_entrypoint,
+    list_modules,
+    is_model_in_modules,
+)
+from .helpers import (  # noqa: F401
+    load_state_dict,
+    load_checkpoint,
+    resume_checkpoint,
+    load_pretrained,
+)
+from . import efficientnet  # noqa: F401  (registers entrypoints)
+from .factory import (  # noqa: F401
+    create_model,
+    create_deepfake_model,
+    create_deepfake_model_v3,
+    create_deepfake_model_v4,
+)
+from .efficientnet import EfficientNet, EfficientNetFeatures  # noqa: F401

@@ -1,0 +1,467 @@
+"""Core layer helpers: conv dispatch, TF-SAME padding, pooling, activations,
+stochastic depth.
+
+Capability parity with the reference layer set (reference
+dfd/timm/models/layers/: create_conv2d.py:11-30, conv2d_same.py:14-31,
+mixed_conv2d.py:20, cond_conv2d.py:33-121, adaptive_avgmax_pool.py:70-97,
+drop.py:24-100, activations.py:19-163) — re-implemented for current
+PyTorch-ROCm; all hot activation/norm work is fused into the HIP kernels in
+deepfake_detection_amd/ops at run time, these modules define parameters and
+the CPU-reference semantics.
+"""
+
+import math
+from functools import partial
+from typing import List, Tuple
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+__all__ = [
+    "get_padding",
+    "get_same_padding",
+    "pad_same",
+    "conv2d_same",
+    "Conv2dSame",
+    "create_conv2d_pad",
+    "MixedConv2d",
+    "CondConv2d",
+    "get_condconv_initializer",
+    "create_conv2d",
+    "sigmoid",
+    "swish",
+    "mish",
+    "hard_swish",
+    "hard_sigmoid",
+    "Swish",
+    "Mish",
+    "HardSwish",
+    "HardSigmoid",
+    "adaptive_avgmax_pool2d",
+    "adaptive_catavgmax_pool2d",
+    "select_adaptive_pool2d",
+    "SelectAdaptivePool2d",
+    "drop_path",
+    "DropPath",
+    "DropBlock2d",
+    "drop_block_2d",
+]
+
+
+# ---------------------------------------------------------------------------
+# Padding helpers (reference dfd/timm/models/layers/padding.py)
+# ---------------------------------------------------------------------------
+
+def get_padding(kernel_size: int, stride: int = 1, dilation: int = 1) -> int:
+    """Symmetric static padding that keeps `out = in // stride`."""
+    return ((stride - 1) + dilation * (kernel_size - 1)) // 2
+
+
+def get_same_padding(x: int, k: int, s: int, d: int) -> int:
+    """Dynamic TF-SAME total padding along one dim."""
+    return max((math.ceil(x / s) - 1) * s + (k - 1) * d + 1 - x, 0)
+
+
+def _is_static_pad(kernel_size: int, stride: int = 1, dilation: int = 1, **_) -> bool:
+    return stride == 1 and (dilation * (kernel_size - 1)) % 2 == 0
+
+
+def pad_same(x, k: List[int], s: List[int], d: List[int] = (1, 1), value: float = 0):
+    ih, iw = x.size()[-2:]
+    pad_h = get_same_padding(ih, k[0], s[0], d[0])
+    pad_w = get_same_padding(iw, k[1], s[1], d[1])
+    if pad_h > 0 or pad_w > 0:
+        x = F.pad(
+            x,
+            [pad_w // 2, pad_w - pad_w // 2, pad_h // 2, pad_h - pad_h // 2],
+            value=value,
+        )
+    return x
+
+
+def conv2d_same(x, weight, bias=None, stride=(1, 1), padding=(0, 0), dilation=(1, 1), groups=1):
+    x = pad_same(x, weight.shape[-2:], stride, dilation)
+    return F.conv2d(x, weight, bias, stride, (0, 0), dilation, groups)
+
+
+class Conv2dSame(nn.Conv2d):
+    """Conv2d with TF-SAME dynamic (possibly asymmetric) padding
+    (reference conv2d_same.py:21-31)."""
+
+    def __init__(self, in_channels, out_channels, kernel_size, stride=1,
+                 padding=0, dilation=1, groups=1, bias=True):
+        super().__init__(in_channels, out_channels, kernel_size, stride, 0, dilation, groups, bias)
+
+    def forward(self, x):
+        return conv2d_same(x, self.weight, self.bias, self.stride,
+                           self.padding, self.dilation, self.groups)
+
+
+def get_padding_value(padding, kernel_size, **kwargs) -> Tuple[object, bool]:
+    stride = kwargs.get("stride", 1)
+    dilation = kwargs.get("dilation", 1)
+    dynamic = False
+    if isinstance(padding, str):
+        padding = padding.lower()
+        if padding == "same":
+            if _is_static_pad(kernel_size, stride, dilation):
+                padding = get_padding(kernel_size, stride, dilation)
+            else:
+                padding = 0
+                dynamic = True
+        elif padding == "valid":
+            padding = 0
+        else:
+            padding = get_padding(kernel_size, stride, dilation)
+    return padding, dynamic
+
+
+def create_conv2d_pad(in_chs, out_chs, kernel_size, **kwargs):
+    padding = kwargs.pop("padding", "")
+    kwargs.setdefault("bias", False)
+    padding, is_dynamic = get_padding_value(padding, kernel_size, **kwargs)
+    if is_dynamic:
+        return Conv2dSame(in_chs, out_chs, kernel_size, **kwargs)
+    return nn.Conv2d(in_chs, out_chs, kernel_size, padding=padding, **kwargs)
+
+
+# ---------------------------------------------------------------------------
+# MixedConv2d (reference mixed_conv2d.py:20) — per-group kernel sizes
+# ---------------------------------------------------------------------------
+
+def _split_channels(num_chan, num_groups):
+    split = [num_chan // num_groups for _ in range(num_groups)]
+    split[0] += num_chan - sum(split)
+    return split
+
+
+class MixedConv2d(nn.ModuleDict):
+    """Mixed grouped conv with per-group kernel size (MixNet)."""
+
+    def __init__(self, in_channels, out_channels, kernel_size=3,
+                 stride=1, padding="", dilation=1, depthwise=False, **kwargs):
+        super().__init__()
+        kernel_size = kernel_size if isinstance(kernel_size, list) else [kernel_size]
+        num_groups = len(kernel_size)
+        in_splits = _split_channels(in_channels, num_groups)
+        out_splits = _split_channels(out_channels, num_groups)
+        self.in_channels = sum(in_splits)
+        self.out_channels = sum(out_splits)
+        for idx, (k, in_ch, out_ch) in enumerate(zip(kernel_size, in_splits, out_splits)):
+            conv_groups = out_ch if depthwise else 1
+            self.add_module(
+                str(idx),
+                create_conv2d_pad(
+                    in_ch, out_ch, k, stride=stride, padding=padding,
+                    dilation=dilation, groups=conv_groups, **kwargs),
+            )
+        self.splits = in_splits
+
+    def forward(self, x):
+        x_split = torch.split(x, self.splits, 1)
+        x_out = [c(x_split[i]) for i, c in enumerate(self.values())]
+        return torch.cat(x_out, 1)
+
+
+# ---------------------------------------------------------------------------
+# CondConv2d (reference cond_conv2d.py:33-121) — per-sample expert mixing
+# ---------------------------------------------------------------------------
+
+def get_condconv_initializer(initializer, num_experts, expert_shape):
+    def condconv_initializer(weight):
+        num_params = 1
+        for d in expert_shape:
+            num_params *= d
+        if weight.dim() != 2 or weight.shape[0] != num_experts or weight.shape[1] != num_params:
+            raise ValueError("CondConv variables must have shape [num_experts, num_params]")
+        for i in range(num_experts):
+            initializer(weight[i].view(expert_shape))
+
+    return condconv_initializer
+
+
+class CondConv2d(nn.Module):
+    """Conditionally-parameterized conv: per-sample expert-weighted kernels,
+    evaluated via the grouped-conv trick."""
+
+    __constants__ = ["in_channels", "out_channels", "dynamic_padding"]
+
+    def __init__(self, in_channels, out_channels, kernel_size=3,
+                 stride=1, padding="", dilation=1, groups=1, bias=False, num_experts=4):
+        super().__init__()
+        self.in_channels = in_channels
+        self.out_channels = out_channels
+        self.kernel_size = (kernel_size, kernel_size) if isinstance(kernel_size, int) else tuple(kernel_size)
+        self.stride = (stride, stride) if isinstance(stride, int) else tuple(stride)
+        padding_val, is_padding_dynamic = get_padding_value(
+            padding, kernel_size, stride=stride, dilation=dilation)
+        self.dynamic_padding = is_padding_dynamic
+        self.padding = (padding_val, padding_val) if isinstance(padding_val, int) else tuple(padding_val)
+        self.dilation = (dilation, dilation) if isinstance(dilation, int) else tuple(dilation)
+        self.groups = groups
+        self.num_experts = num_experts
+
+        self.weight_shape = (self.out_channels, self.in_channels // self.groups) + self.kernel_size
+        weight_num_param = 1
+        for wd in self.weight_shape:
+            weight_num_param *= wd
+        self.weight = nn.Parameter(torch.Tensor(self.num_experts, weight_num_param))
+
+        if bias:
+            self.bias_shape = (self.out_channels,)
+            self.bias = nn.Parameter(torch.Tensor(self.num_experts, self.out_channels))
+        else:
+            self.register_parameter("bias", None)
+
+        self.reset_parameters()
+
+    def reset_parameters(self):
+        init_weight = get_condconv_initializer(
+            partial(nn.init.kaiming_uniform_, a=math.sqrt(5)), self.num_experts, self.weight_shape)
+        init_weight(self.weight)
+        if self.bias is not None:
+            fan_in = self.weight_shape[1] * self.weight_shape[2] * self.weight_shape[3]
+            bound = 1 / math.sqrt(fan_in)
+            init_bias = get_condconv_initializer(
+                partial(nn.init.uniform_, a=-bound, b=bound), self.num_experts, self.bias_shape)
+            init_bias(self.bias)
+
+    def forward(self, x, routing_weights):
+        B, C, H, W = x.shape
+        weight = torch.matmul(routing_weights, self.weight)
+        new_weight_shape = (B * self.out_channels, self.in_channels // self.groups) + self.kernel_size
+        weight = weight.view(new_weight_shape)
+        bias = None
+        if self.bias is not None:
+            bias = torch.matmul(routing_weights, self.bias).view(B * self.out_channels)
+        x = x.reshape(1, B * C, H, W)
+        if self.dynamic_padding:
+            out = conv2d_same(
+                x, weight, bias, stride=self.stride, padding=self.padding,
+                dilation=self.dilation, groups=self.groups * B)
+        else:
+            out = F.conv2d(
+                x, weight, bias, stride=self.stride, padding=self.padding,
+                dilation=self.dilation, groups=self.groups * B)
+        return out.permute([1, 0, 2, 3]).view(B, self.out_channels, out.shape[-2], out.shape[-1])
+
+
+# ---------------------------------------------------------------------------
+# create_conv2d dispatch (reference create_conv2d.py:11-30)
+# ---------------------------------------------------------------------------
+
+def create_conv2d(in_chs, out_chs, kernel_size, **kwargs):
+    """Dispatch to MixedConv2d / CondConv2d / Conv2dSame / Conv2d."""
+    if isinstance(kernel_size, list):
+        assert "num_experts" not in kwargs
+        assert "groups" not in kwargs
+        return MixedConv2d(in_chs, out_chs, kernel_size, **kwargs)
+    depthwise = kwargs.pop("depthwise", False)
+    groups = out_chs if depthwise else kwargs.pop("groups", 1)
+    if "num_experts" in kwargs and kwargs["num_experts"] > 0:
+        return CondConv2d(in_chs, out_chs, kernel_size, groups=groups, **kwargs)
+    return create_conv2d_pad(in_chs, out_chs, kernel_size, groups=groups, **kwargs)
+
+
+# ---------------------------------------------------------------------------
+# Activations (reference activations.py). On GPU these are fused into the
+# producing HIP kernel (ops/bn_act.py); the modules here define CPU-reference
+# semantics and hold no state.
+# ---------------------------------------------------------------------------
+
+def sigmoid(x, inplace: bool = False):
+    return x.sigmoid_() if inplace else x.sigmoid()
+
+
+def swish(x, inplace: bool = False):
+    """x * sigmoid(x). The reference's memory-efficient autograd variant
+    (activations.py:19-48) is unnecessary on torch>=2: F.silu is fused."""
+    return F.silu(x, inplace=inplace)
+
+
+def mish(x, inplace: bool = False):
+    return F.mish(x, inplace=inplace)
+
+
+def hard_sigmoid(x, inplace: bool = False):
+    if inplace:
+        return x.add_(3.0).clamp_(0.0, 6.0).div_(6.0)
+    return F.relu6(x + 3.0) / 6.0
+
+
+def hard_swish(x, inplace: bool = False):
+    return F.hardswish(x, inplace=inplace)
+
+
+class Swish(nn.Module):
+    def __init__(self, inplace: bool = False):
+        super().__init__()
+        self.inplace = inplace
+
+    def forward(self, x):
+        return swish(x, self.inplace)
+
+
+class Mish(nn.Module):
+    def __init__(self, inplace: bool = False):
+        super().__init__()
+        self.inplace = inplace
+
+    def forward(self, x):
+        return mish(x, self.inplace)
+
+
+class HardSwish(nn.Module):
+    def __init__(self, inplace: bool = False):
+        super().__init__()
+        self.inplace = inplace
+
+    def forward(self, x):
+        return hard_swish(x, self.inplace)
+
+
+class HardSigmoid(nn.Module):
+    def __init__(self, inplace: bool = False):
+        super().__init__()
+        self.inplace = inplace
+
+    def forward(self, x):
+        return hard_sigmoid(x, self.inplace)
+
+
+# ---------------------------------------------------------------------------
+# Adaptive pooling (reference adaptive_avgmax_pool.py:70-97)
+# ---------------------------------------------------------------------------
+
+def adaptive_pool_feat_mult(pool_type="avg"):
+    return 2 if pool_type == "catavgmax" else 1
+
+
+def adaptive_avgmax_pool2d(x, output_size=1):
+    x_avg = F.adaptive_avg_pool2d(x, output_size)
+    x_max = F.adaptive_max_pool2d(x, output_size)
+    return 0.5 * (x_avg + x_max)
+
+
+def adaptive_catavgmax_pool2d(x, output_size=1):
+    x_avg = F.adaptive_avg_pool2d(x, output_size)
+    x_max = F.adaptive_max_pool2d(x, output_size)
+    return torch.cat((x_avg, x_max), 1)
+
+
+def select_adaptive_pool2d(x, pool_type="avg", output_size=1):
+    if pool_type == "avg":
+        return F.adaptive_avg_pool2d(x, output_size)
+    if pool_type == "avgmax":
+        return adaptive_avgmax_pool2d(x, output_size)
+    if pool_type == "catavgmax":
+        return adaptive_catavgmax_pool2d(x, output_size)
+    if pool_type == "max":
+        return F.adaptive_max_pool2d(x, output_size)
+    raise AssertionError("Invalid pool type: %s" % pool_type)
+
+
+class AdaptiveAvgMaxPool2d(nn.Module):
+    def __init__(self, output_size=1):
+        super().__init__()
+        self.output_size = output_size
+
+    def forward(self, x):
+        return adaptive_avgmax_pool2d(x, self.output_size)
+
+
+class AdaptiveCatAvgMaxPool2d(nn.Module):
+    def __init__(self, output_size=1):
+        super().__init__()
+        self.output_size = output_size
+
+    def forward(self, x):
+        return adaptive_catavgmax_pool2d(x, self.output_size)
+
+
+class SelectAdaptivePool2d(nn.Module):
+    """Selectable global pooling with dynamic input size."""
+
+    def __init__(self, output_size=1, pool_type="avg"):
+        super().__init__()
+        self.output_size = output_size
+        self.pool_type = pool_type
+        if pool_type == "avgmax":
+            self.pool = AdaptiveAvgMaxPool2d(output_size)
+        elif pool_type == "catavgmax":
+            self.pool = AdaptiveCatAvgMaxPool2d(output_size)
+        elif pool_type == "max":
+            self.pool = nn.AdaptiveMaxPool2d(output_size)
+        else:
+            assert pool_type == "avg", "Invalid pool type: %s" % pool_type
+            self.pool = nn.AdaptiveAvgPool2d(output_size)
+
+    def forward(self, x):
+        return self.pool(x)
+
+    def feat_mult(self):
+        return adaptive_pool_feat_mult(self.pool_type)
+
+    def __repr__(self):
+        return self.__class__.__name__ + " (output_size=" + str(self.output_size) \
+            + ", pool_type=" + self.pool_type + ")"
+
+
+# ---------------------------------------------------------------------------
+# Stochastic depth / DropBlock (reference drop.py:24-100)
+# ---------------------------------------------------------------------------
+
+def drop_path(x, drop_prob: float = 0.0, training: bool = False):
+    """Per-sample stochastic depth: zero the whole residual branch with
+    probability `drop_prob`, scale survivors by 1/keep."""
+    if drop_prob == 0.0 or not training:
+        return x
+    keep_prob = 1 - drop_prob
+    shape = (x.shape[0],) + (1,) * (x.ndim - 1)
+    random_tensor = keep_prob + torch.rand(shape, dtype=x.dtype, device=x.device)
+    random_tensor.floor_()
+    return x.div(keep_prob) * random_tensor
+
+
+class DropPath(nn.Module):
+    def __init__(self, drop_prob=None):
+        super().__init__()
+        self.drop_prob = drop_prob
+
+    def forward(self, x):
+        return drop_path(x, self.drop_prob, self.training)
+
+
+def drop_block_2d(x, drop_prob=0.1, block_size=7, gamma_scale=1.0, drop_with_noise=False):
+    """DropBlock (https://arxiv.org/abs/1810.12890) — structured dropout."""
+    _, _, H, W = x.shape
+    total_size = W * H
+    clipped_block_size = min(block_size, min(W, H))
+    gamma = gamma_scale * drop_prob * total_size / clipped_block_size ** 2 / (
+        (W - block_size + 1) * (H - block_size + 1))
+    block_mask = torch.rand_like(x) < gamma
+    block_mask = F.max_pool2d(
+        block_mask.to(x.dtype), kernel_size=clipped_block_size,
+        stride=1, padding=clipped_block_size // 2)
+    if drop_with_noise:
+        normal_noise = torch.randn_like(x)
+        x = x * (1.0 - block_mask) + normal_noise * block_mask
+    else:
+        normalize_scale = block_mask.numel() / (block_mask.to(torch.float32).sum() + 1e-7)
+        x = x * (1.0 - block_mask) * normalize_scale.to(x.dtype)
+    return x
+
+
+class DropBlock2d(nn.Module):
+    def __init__(self, drop_prob=0.1, block_size=7, gamma_scale=1.0, with_noise=False):
+        super().__init__()
+        self.drop_prob = drop_prob
+        self.gamma_scale = gamma_scale
+        self.block_size = block_size
+        self.with_noise = with_noise
+
+    def forward(self, x):
+        if not self.training or not self.drop_prob:
+            return x
+        return drop_block_2d(x, self.drop_prob, self.block_size, self.gamma_scale, self.with_noise)

@@ -1,0 +1,78 @@
+"""Dispatch layer: model blocks call these; on ROCm devices they run the
+hand-written gfx950 HIP kernels, on CPU the PyTorch reference path.
+
+The HIP path is NOT optional on a GPU machine: if a CUDA(ROCm) tensor
+arrives and the extension is missing, we raise (see ops/extension.py).
+"""
+
+import torch
+import torch.nn.functional as F
+
+from . import reference
+from .extension import gpu_ops_required
+
+_FUSED_ACTS = ("none", "relu", "silu")
+
+
+def act_name_of(module) -> str:
+    """Map an activation module instance to a fused-kernel act name."""
+    import torch.nn as nn
+
+    from ..models import layers as L
+
+    if module is None or isinstance(module, nn.Identity):
+        return "none"
+    if isinstance(module, (nn.SiLU, L.Swish)):
+        return "silu"
+    if isinstance(module, (nn.ReLU, nn.ReLU6)):
+        return "relu"
+    return "other"
+
+
+def bn_act(x, bn, act: str = "silu"):
+    """BatchNorm2d + activation, fused on GPU (HIP kernel, NHWC, bf16 I/O,
+    fp32 stats — SURVEY.md §2.6 item 5), torch ops on CPU."""
+    if x.is_cuda and gpu_ops_required() and act in _FUSED_ACTS:
+        from .bn_act import fused_bn_act
+
+        return fused_bn_act(
+            x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
+            bn.training, bn.momentum, bn.eps, act,
+        )
+    y = F.batch_norm(
+        x, bn.running_mean, bn.running_var, bn.weight, bn.bias,
+        bn.training, bn.momentum if bn.momentum is not None else 0.1, bn.eps,
+    )
+    if act == "silu":
+        return F.silu(y)
+    if act == "relu":
+        return F.relu(y)
+    return y
+
+
+def se(x, conv_reduce, act_module, conv_expand):
+    """Squeeze-excite chain (pool -> 1x1 -> act -> 1x1 -> sigmoid -> mul).
+
+    GPU: fused HIP kernel chain (ops/hip/se.hip); CPU: reference path.
+    Falls back to composable torch ops for non-silu/relu gates.
+    """
+    act = act_name_of(act_module)
+    if x.is_cuda and gpu_ops_required() and act in _FUSED_ACTS:
+        from .se import fused_se
+
+        return fused_se(x, conv_reduce.weight, conv_reduce.bias,
+                        conv_expand.weight, conv_expand.bias, act)
+    s = x.mean(dim=(2, 3), keepdim=True)
+    s = conv_reduce(s)
+    s = act_module(s)
+    s = conv_expand(s)
+    return x * torch.sigmoid(s)
+
+
+def global_avg_pool(x):
+    """Global average pool NxCxHxW -> NxC (head pooling)."""
+    if x.is_cuda and gpu_ops_required():
+        from .pool import fused_global_avg_pool
+
+        return fused_global_avg_pool(x)
+    return x.mean(dim=(2, 3))

@@ -1,0 +1,119 @@
+"""Model-zoo structural tests: param parity with the reference vendored-timm
+(SURVEY.md §2.3 verified numbers), state-dict key layout, registry/factory."""
+
+import pytest
+import torch
+
+import deepfake_detection_amd as dfd
+from deepfake_detection_amd.models import list_models, is_model
+from deepfake_detection_amd.models.blocks import make_divisible, round_channels
+
+
+def test_registry_basic():
+    assert is_model("efficientnet_deepfake_v4")
+    assert "efficientnet_b0" in list_models()
+    assert list_models("efficientnet_b*")[0] == "efficientnet_b0"
+
+
+def test_round_channels():
+    # reference semantics (efficientnet_blocks.py:55-69)
+    assert make_divisible(32) == 32
+    assert round_channels(16, 2.0) == 32
+    assert round_channels(128, 2.0) == 256
+    assert round_channels(1280, 1.4) == 1792
+    assert round_channels(320, 2.0) == 640
+
+
+def test_deepfake_v4_shape_parity():
+    """Verified reference shape: 62,373,826 params, 1200 keys, stem
+    Conv2d(12,256,3x3,s2), head Conv2d(640,256,1x1), blocks [4,7,7,10,10,13,4]
+    (SURVEY.md §2.3)."""
+    m = dfd.create_deepfake_model_v4("efficientnet_deepfake_v4", num_classes=2, in_chans=12)
+    assert sum(p.numel() for p in m.parameters()) == 62373826
+    sd = list(m.state_dict().keys())
+    assert len(sd) == 1200
+    assert sd[0] == "conv_stem.weight"
+    assert sd[1].startswith("bn1.")
+    assert sd[-2:] == ["classifier.weight", "classifier.bias"]
+    assert [len(s) for s in m.blocks] == [4, 7, 7, 10, 10, 13, 4]
+    assert m.conv_stem.in_channels == 12 and m.conv_stem.out_channels == 256
+    assert m.conv_head.in_channels == 640 and m.conv_head.out_channels == 256
+
+
+def test_deepfake_v4_block_key_names():
+    m = dfd.create_deepfake_model_v4("efficientnet_deepfake_v4", num_classes=2, in_chans=12)
+    sd = m.state_dict()
+    # IR block key layout (reference efficientnet_blocks.py:260-348)
+    for k in ["blocks.1.0.conv_pw.weight", "blocks.1.0.bn1.weight",
+              "blocks.1.0.conv_dw.weight", "blocks.1.0.bn2.running_mean",
+              "blocks.1.0.se.conv_reduce.weight", "blocks.1.0.se.conv_expand.bias",
+              "blocks.1.0.conv_pwl.weight", "blocks.1.0.bn3.bias"]:
+        assert k in sd, k
+    # DS block key layout (stage 0)
+    for k in ["blocks.0.0.conv_dw.weight", "blocks.0.0.bn1.weight",
+              "blocks.0.0.se.conv_reduce.weight", "blocks.0.0.conv_pw.weight",
+              "blocks.0.0.bn2.weight"]:
+        assert k in sd, k
+
+
+@pytest.mark.parametrize("name,expect", [
+    ("efficientnet_b0", 5288548),
+    ("efficientnet_b4", 19341616),
+])
+def test_imagenet_param_parity(name, expect):
+    m = dfd.create_model(name)
+    assert sum(p.numel() for p in m.parameters()) == expect
+
+
+def test_forward_shapes():
+    m = dfd.create_model("efficientnet_b0", num_classes=10)
+    m.eval()
+    with torch.no_grad():
+        y = m(torch.randn(2, 3, 64, 64))
+    assert y.shape == (2, 10)
+
+
+def test_deepfake_v4_forward_backward():
+    m = dfd.create_deepfake_model_v4("efficientnet_deepfake_v4", num_classes=2, in_chans=12)
+    x = torch.randn(2, 12, 64, 64)
+    y = m(x)
+    assert y.shape == (2, 2)
+    y.sum().backward()
+    assert m.conv_stem.weight.grad is not None
+
+
+def test_factory_asserts_model_name():
+    with pytest.raises(AssertionError):
+        dfd.create_deepfake_model_v4("efficientnet_b0")
+
+
+def test_bn_momentum_plumbs():
+    m = dfd.create_deepfake_model_v4(
+        "efficientnet_deepfake_v4", num_classes=2, in_chans=12, bn_momentum=0.001)
+    assert abs(m.bn1.momentum - 0.001) < 1e-12
+    assert abs(m.blocks[3][2].bn2.momentum - 0.001) < 1e-12
+
+
+def test_checkpoint_roundtrip(tmp_path):
+    m = dfd.create_deepfake_model_v4("efficientnet_deepfake_v4", num_classes=2, in_chans=12)
+    p = tmp_path / "ckpt.pth.tar"
+    # module.-prefixed save must load (reference helpers.py:17-20)
+    sd = {"module." + k: v for k, v in m.state_dict().items()}
+    torch.save({"state_dict": sd, "epoch": 3, "version": 2, "arch": "efficientnet_deepfake_v4"}, p)
+    m2 = dfd.create_deepfake_model_v4("efficientnet_deepfake_v4", num_classes=2, in_chans=12)
+    from deepfake_detection_amd.models import resume_checkpoint
+
+    other, epoch = resume_checkpoint(m2, str(p))
+    assert epoch == 4  # version>=2 -> epoch+1 (reference helpers.py:62-66)
+    for k in m.state_dict():
+        assert torch.equal(m.state_dict()[k], m2.state_dict()[k])
+
+
+def test_nonstrict_load_drops_mismatched(tmp_path):
+    m = dfd.create_deepfake_model_v4("efficientnet_deepfake_v4", num_classes=2, in_chans=12)
+    p = tmp_path / "c.pth.tar"
+    torch.save({"state_dict": m.state_dict()}, p)
+    # different head size: non-strict load drops the mismatched classifier
+    m3 = dfd.create_deepfake_model_v4(
+        "efficientnet_deepfake_v4", num_classes=5, in_chans=12, checkpoint_path=str(p))
+    assert m3.classifier.out_features == 5
