@@ -1,0 +1,46 @@
+"""Loss tests: label-smoothing vs plain CE limit, soft-target equivalence,
+JSD structure."""
+
+import torch
+import torch.nn.functional as F
+
+from deepfake_detection_amd.loss import (
+    JsdCrossEntropy,
+    LabelSmoothingCrossEntropy,
+    SoftTargetCrossEntropy,
+)
+
+
+def test_label_smoothing_zero_equals_ce():
+    torch.manual_seed(0)
+    x = torch.randn(8, 4)
+    t = torch.randint(0, 4, (8,))
+    ls = LabelSmoothingCrossEntropy(smoothing=1e-9)
+    assert torch.allclose(ls(x, t), F.cross_entropy(x, t), atol=1e-5)
+
+
+def test_soft_target_equals_ce_on_onehot():
+    torch.manual_seed(0)
+    x = torch.randn(8, 4)
+    t = torch.randint(0, 4, (8,))
+    onehot = F.one_hot(t, 4).float()
+    st = SoftTargetCrossEntropy()
+    assert torch.allclose(st(x, onehot), F.cross_entropy(x, t), atol=1e-6)
+
+
+def test_label_smoothing_value():
+    x = torch.tensor([[10.0, 0.0]])
+    t = torch.tensor([0])
+    ls = LabelSmoothingCrossEntropy(smoothing=0.1)
+    logprobs = F.log_softmax(x, dim=-1)
+    expect = -0.9 * logprobs[0, 0] - 0.1 * logprobs.mean()
+    assert torch.allclose(ls(x, t), expect, atol=1e-6)
+
+
+def test_jsd_runs():
+    torch.manual_seed(0)
+    x = torch.randn(12, 4)  # 3 splits of 4
+    t = torch.randint(0, 4, (4,))
+    jsd = JsdCrossEntropy(num_splits=3, alpha=12, smoothing=0.1)
+    loss = jsd(x, t)
+    assert loss.ndim == 0 and torch.isfinite(loss)

@@ -1,0 +1,66 @@
+"""Scheduler behavior tests (step decay + warmup, cosine cycle, factory)."""
+
+import types
+
+import torch
+
+from deepfake_detection_amd.scheduler import (
+    CosineLRScheduler,
+    StepLRScheduler,
+    create_scheduler,
+)
+
+
+def _opt(lr=1.0):
+    p = torch.nn.Parameter(torch.zeros(1))
+    return torch.optim.SGD([p], lr=lr)
+
+
+def test_step_lr_decay():
+    opt = _opt(1.0)
+    sched = StepLRScheduler(opt, decay_t=2, decay_rate=0.5)
+    lrs = []
+    for epoch in range(6):
+        sched.step(epoch)
+        lrs.append(opt.param_groups[0]["lr"])
+    assert lrs == [1.0, 1.0, 0.5, 0.5, 0.25, 0.25]
+
+
+def test_step_lr_warmup():
+    opt = _opt(1.0)
+    sched = StepLRScheduler(opt, decay_t=10, decay_rate=0.9, warmup_t=4, warmup_lr_init=0.2)
+    assert abs(opt.param_groups[0]["lr"] - 0.2) < 1e-9  # init to warmup_lr
+    sched.step(2)
+    assert abs(opt.param_groups[0]["lr"] - 0.6) < 1e-9  # halfway up
+    sched.step(4)
+    assert abs(opt.param_groups[0]["lr"] - 1.0) < 1e-9
+
+
+def test_cosine_cycle():
+    opt = _opt(1.0)
+    sched = CosineLRScheduler(opt, t_initial=10, lr_min=0.0)
+    sched.step(0)
+    assert abs(opt.param_groups[0]["lr"] - 1.0) < 1e-9
+    sched.step(5)
+    assert abs(opt.param_groups[0]["lr"] - 0.5) < 1e-6
+
+
+def test_factory_step():
+    args = types.SimpleNamespace(
+        epochs=10, sched="step", decay_epochs=2, decay_rate=0.92,
+        warmup_lr=1e-4, warmup_epochs=0, min_lr=1e-5, cooldown_epochs=0,
+        lr_noise=None, seed=42, eval_metric="loss")
+    opt = _opt(0.5)
+    sched, num_epochs = create_scheduler(args, opt)
+    assert isinstance(sched, StepLRScheduler)
+    assert num_epochs == 10
+
+
+def test_factory_cosine_adds_cooldown():
+    args = types.SimpleNamespace(
+        epochs=10, sched="cosine", decay_epochs=2, decay_rate=0.1,
+        warmup_lr=1e-4, warmup_epochs=0, min_lr=1e-5, cooldown_epochs=3,
+        lr_noise=None, seed=42, eval_metric="loss")
+    opt = _opt(0.5)
+    sched, num_epochs = create_scheduler(args, opt)
+    assert num_epochs == 13
