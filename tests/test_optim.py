@@ -100,8 +100,13 @@ def test_lookahead_sync():
     for _ in range(2):
         p.grad = torch.tensor([1.0])
         opt.step()
-    # fast: 1 -> 0.9 -> 0.8; slow sync at k=2: 1 + 0.5*(0.8-1) = 0.9
-    assert abs(p.item() - 0.9) < 1e-6
+    # fast: 1 -> 0.9 -> 0.8; slow buffer lazily initialized at first sync
+    assert abs(p.item() - 0.8) < 1e-6
+    for _ in range(2):
+        p.grad = torch.tensor([1.0])
+        opt.step()
+    # fast: 0.8 -> 0.7 -> 0.6; sync: slow = 0.8 + 0.5*(0.6-0.8) = 0.7
+    assert abs(p.item() - 0.7) < 1e-6
     opt.sync_lookahead()
 
 
