@@ -1,0 +1,22 @@
+from .config import resolve_data_config  # noqa: F401
+from .constants import *  # noqa: F401,F403
+from .dataset import (  # noqa: F401
+    Dataset,
+    DeepFakeDataset_v3,
+    SyntheticDeepFakeDataset,
+)
+from .distributed_sampler import OrderedDistributedSampler  # noqa: F401
+from .loader import (  # noqa: F401
+    PrefetchLoader,
+    PrefetchLoader_v3,
+    create_deepfake_loader_v3,
+    create_loader,
+    fast_collate,
+)
+from .mixup import FastCollateMixup, mixup_batch, mixup_target, one_hot  # noqa: F401
+from .random_erasing import RandomErasing  # noqa: F401
+from .transforms_factory import (  # noqa: F401
+    create_transform,
+    transforms_deepfake_eval_v3,
+    transforms_deepfake_train_v3,
+)
