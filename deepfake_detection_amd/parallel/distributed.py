@@ -1,0 +1,109 @@
+"""Distributed runtime: RCCL process group over xGMI, metric reduction,
+BN-buffer sync, DDP wrapping.
+
+Capability parity with reference collective call sites (SURVEY.md §2.6
+items 18-22): `reduce_tensor` (reference timm/utils.py:256-260),
+`distribute_bn` (:263-274), process-group init (reference
+train.py:275-282), DDP wrap (train.py:401-406).
+
+MI355X-native design: one process per GPU; backend "nccl" IS RCCL on
+ROCm. Gradient reduction uses torch DDP's bucketed all-reduce overlapped
+with backward (gradient_as_bucket_view; bucket size tuned for the 7-link
+xGMI p2p fabric — per-link ring bandwidth ≈153 GB/s favours fewer, larger
+buckets than NVSwitch defaults) instead of the reference's apex
+`delay_allreduce` whole-model reduce.
+"""
+
+import logging
+import os
+
+import torch
+import torch.distributed as dist
+
+_logger = logging.getLogger(__name__)
+
+# xGMI: ring all-reduce is per-link bound; 60 MB buckets keep the pipeline
+# deep enough to overlap with backward while avoiding per-bucket latency.
+XGMI_BUCKET_CAP_MB = 60
+
+
+def init_distributed(backend=None, init_method=None, world_size=None, rank=None,
+                     local_rank=None, device_type=None):
+    """Initialize the process group from torchrun-style env or explicit args.
+
+    Returns (device, world_size, rank, local_rank). Single-process when no
+    WORLD_SIZE in env and no explicit world_size.
+    """
+    world_size = world_size if world_size is not None else int(os.environ.get("WORLD_SIZE", 1))
+    rank = rank if rank is not None else int(os.environ.get("RANK", 0))
+    local_rank = local_rank if local_rank is not None else int(os.environ.get("LOCAL_RANK", rank))
+
+    use_cuda = torch.cuda.is_available() if device_type is None else device_type == "cuda"
+    if use_cuda:
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
+
+    if world_size > 1 and not dist.is_initialized():
+        backend = backend or ("nccl" if use_cuda else "gloo")
+        kwargs = dict(backend=backend, world_size=world_size, rank=rank)
+        if init_method:
+            kwargs["init_method"] = init_method
+        elif "MASTER_ADDR" not in os.environ:
+            # file-store rendezvous fallback (reference train.py:279-282)
+            kwargs["init_method"] = "file:///tmp/dfd_amd_pg_init"
+        dist.init_process_group(**kwargs)
+        _logger.info("Initialized %s process group: rank %d/%d (local %d)",
+                     backend, rank, world_size, local_rank)
+    return device, world_size, rank, local_rank
+
+
+def is_primary(rank=None):
+    if rank is not None:
+        return rank == 0
+    return (not dist.is_initialized()) or dist.get_rank() == 0
+
+
+def reduce_tensor(tensor, n):
+    """clone -> all_reduce(SUM) -> /n (metric averaging)."""
+    rt = tensor.clone()
+    dist.all_reduce(rt, op=dist.ReduceOp.SUM)
+    rt /= n
+    return rt
+
+
+def distribute_bn(model, world_size, reduce=False):
+    """Per-epoch BN-buffer sync: all-reduce-mean or rank-0 broadcast of
+    every running_mean/running_var."""
+    from ..utils.model import unwrap_model
+
+    for bn_name, bn_buf in unwrap_model(model).named_buffers(recurse=True):
+        if ("running_mean" in bn_name) or ("running_var" in bn_name):
+            if reduce:
+                dist.all_reduce(bn_buf, op=dist.ReduceOp.SUM)
+                bn_buf /= float(world_size)
+            else:
+                dist.broadcast(bn_buf, 0)
+
+
+def wrap_ddp(model, device, find_unused_parameters=False, bucket_cap_mb=XGMI_BUCKET_CAP_MB,
+             static_graph=True):
+    """Wrap in torch DDP configured for RCCL/xGMI: bucketed all-reduce
+    overlapped with backward, gradients viewed into flat buckets (no extra
+    copy), static graph (fixed CNN) enables bucket-order capture."""
+    device_ids = [device.index] if device.type == "cuda" else None
+    ddp = torch.nn.parallel.DistributedDataParallel(
+        model,
+        device_ids=device_ids,
+        bucket_cap_mb=bucket_cap_mb,
+        gradient_as_bucket_view=True,
+        find_unused_parameters=find_unused_parameters,
+        static_graph=static_graph,
+    )
+    return ddp
+
+
+def convert_sync_batchnorm(model):
+    """Optional SyncBatchNorm conversion (reference train.py:387-398)."""
+    return torch.nn.SyncBatchNorm.convert_sync_batchnorm(model)
