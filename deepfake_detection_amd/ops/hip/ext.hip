@@ -1,0 +1,48 @@
+// Python bindings for the gfx950 kernel set (module: _hip_ops).
+
+#include <torch/extension.h>
+
+#include <vector>
+
+std::vector<at::Tensor> bn_act_fwd(at::Tensor x, at::Tensor weight, at::Tensor bias,
+                                   at::Tensor running_mean, at::Tensor running_var,
+                                   bool training, double momentum, double eps,
+                                   std::string act);
+std::vector<at::Tensor> bn_act_bwd(at::Tensor dy, at::Tensor x, at::Tensor weight,
+                                   at::Tensor bias, at::Tensor save_mean,
+                                   at::Tensor save_invstd, bool training, std::string act);
+at::Tensor normalize_uint8_nhwc(at::Tensor x, at::Tensor mean, at::Tensor std,
+                                std::string dtype, bool channels_last);
+at::Tensor global_avg_pool_fwd(at::Tensor x);
+at::Tensor global_avg_pool_bwd(at::Tensor dy, long long N, long long C, long long H,
+                               long long W);
+std::vector<at::Tensor> se_fwd(at::Tensor x, at::Tensor w1, at::Tensor b1, at::Tensor w2,
+                               at::Tensor b2, std::string act);
+std::vector<at::Tensor> se_bwd_reduce(at::Tensor dy, at::Tensor x, at::Tensor g);
+void se_bwd_add_pool(at::Tensor dx, at::Tensor ds);
+void rmsprop_tf_multi_tensor(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+                             std::vector<at::Tensor> square_avgs,
+                             std::vector<at::Tensor> momentum_buffers, double lr,
+                             double alpha, double eps, double momentum, double weight_decay,
+                             bool decoupled_decay, bool lr_in_momentum);
+void adamw_multi_tensor(std::vector<at::Tensor> params, std::vector<at::Tensor> grads,
+                        std::vector<at::Tensor> exp_avgs, std::vector<at::Tensor> exp_avg_sqs,
+                        long long step, double lr, double beta1, double beta2, double eps,
+                        double weight_decay);
+void ema_multi_tensor(std::vector<at::Tensor> ema_params, std::vector<at::Tensor> model_params,
+                      double decay);
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "deepfake_detection_amd gfx950 (MI355X/CDNA4) kernels";
+  m.def("bn_act_fwd", &bn_act_fwd, "fused BatchNorm+act forward (NHWC)");
+  m.def("bn_act_bwd", &bn_act_bwd, "fused BatchNorm+act backward (NHWC)");
+  m.def("normalize_uint8_nhwc", &normalize_uint8_nhwc, "uint8 NCHW -> norm NHWC");
+  m.def("global_avg_pool_fwd", &global_avg_pool_fwd, "global avg pool fwd (NHWC)");
+  m.def("global_avg_pool_bwd", &global_avg_pool_bwd, "global avg pool bwd (NHWC)");
+  m.def("se_fwd", &se_fwd, "fused squeeze-excite forward (NHWC)");
+  m.def("se_bwd_reduce", &se_bwd_reduce, "SE backward: dx-direct + dgate");
+  m.def("se_bwd_add_pool", &se_bwd_add_pool, "SE backward: add pooled ds");
+  m.def("rmsprop_tf_multi_tensor", &rmsprop_tf_multi_tensor, "fused RMSpropTF step");
+  m.def("adamw_multi_tensor", &adamw_multi_tensor, "fused AdamW step");
+  m.def("ema_multi_tensor", &ema_multi_tensor, "fused EMA update");
+}

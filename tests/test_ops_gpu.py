@@ -1,0 +1,283 @@
+"""GPU numerics tests: every HIP kernel vs a plain PyTorch fp32 reference
+(SURVEY.md §4 strategy). All marked @pytest.mark.gpu."""
+
+import os
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _cl(x):
+    return x.contiguous(memory_format=torch.channels_last)
+
+
+@pytest.fixture(scope="module")
+def ext():
+    from deepfake_detection_amd.ops.extension import load_extension
+
+    return load_extension()
+
+
+# ---------------------------------------------------------------------------
+# normalize
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("dtype", ["float32", "bfloat16", "float16"])
+def test_normalize_uint8(ext, dtype):
+    torch.manual_seed(0)
+    x = torch.randint(0, 256, (3, 12, 37, 41), dtype=torch.uint8, device="cuda")
+    mean = torch.rand(12, device="cuda") * 255
+    std = torch.rand(12, device="cuda") * 100 + 20
+    out = ext.normalize_uint8_nhwc(x, mean, std, dtype, True)
+    ref = (x.float() - mean.view(1, 12, 1, 1)) / std.view(1, 12, 1, 1)
+    assert out.is_contiguous(memory_format=torch.channels_last)
+    tol = 1e-5 if dtype == "float32" else (0.02 if dtype == "bfloat16" else 0.005)
+    assert torch.allclose(out.float(), ref, atol=tol, rtol=tol)
+
+
+# ---------------------------------------------------------------------------
+# fused BN + act
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("training", [True, False])
+@pytest.mark.parametrize("act", ["silu", "relu", "none"])
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_bn_act_forward(ext, training, act, dtype):
+    torch.manual_seed(1)
+    N, C, H, W = 4, 48, 17, 19
+    x = _cl(torch.randn(N, C, H, W, device="cuda", dtype=dtype))
+    weight = torch.randn(C, device="cuda") * 0.5 + 1
+    bias = torch.randn(C, device="cuda") * 0.1
+    rm = torch.randn(C, device="cuda") * 0.1
+    rv = torch.rand(C, device="cuda") + 0.5
+    rm_ref, rv_ref = rm.clone(), rv.clone()
+    momentum, eps = 0.01, 1e-3
+
+    y, mean, invstd = ext.bn_act_fwd(x, weight, bias, rm, rv, training, momentum, eps, act)
+
+    ref = torch.nn.functional.batch_norm(
+        x.float(), rm_ref, rv_ref, weight, bias, training, momentum, eps)
+    if act == "silu":
+        ref = torch.nn.functional.silu(ref)
+    elif act == "relu":
+        ref = torch.nn.functional.relu(ref)
+
+    tol = 1e-4 if dtype == torch.float32 else 0.05
+    assert torch.allclose(y.float(), ref, atol=tol, rtol=tol)
+    # running stats parity (fp32 path, exact math)
+    assert torch.allclose(rm, rm_ref, atol=1e-4, rtol=1e-4)
+    assert torch.allclose(rv, rv_ref, atol=1e-3, rtol=1e-3)
+
+
+@pytest.mark.parametrize("training", [True, False])
+@pytest.mark.parametrize("act", ["silu", "none"])
+def test_bn_act_backward(ext, training, act):
+    torch.manual_seed(2)
+    N, C, H, W = 3, 32, 13, 11
+    x = _cl(torch.randn(N, C, H, W, device="cuda"))
+    weight = torch.randn(C, device="cuda") * 0.5 + 1
+    bias = torch.randn(C, device="cuda") * 0.1
+    rm = torch.zeros(C, device="cuda")
+    rv = torch.ones(C, device="cuda")
+    dy = _cl(torch.randn(N, C, H, W, device="cuda"))
+    eps = 1e-3
+
+    y, mean, invstd = ext.bn_act_fwd(x, weight, bias, rm.clone(), rv.clone(),
+                                     training, 0.1, eps, act)
+    dx, dgamma, dbeta = ext.bn_act_bwd(dy, x, weight, bias, mean, invstd, training, act)
+
+    # torch autograd reference in fp32
+    x_ref = x.float().detach().requires_grad_(True)
+    w_ref = weight.detach().requires_grad_(True)
+    b_ref = bias.detach().requires_grad_(True)
+    ref = torch.nn.functional.batch_norm(
+        x_ref, rm.clone(), rv.clone(), w_ref, b_ref, training, 0.1, eps)
+    if act == "silu":
+        ref = torch.nn.functional.silu(ref)
+    ref.backward(dy.float())
+
+    assert torch.allclose(y.float(), ref.detach(), atol=1e-4, rtol=1e-4)
+    assert torch.allclose(dx.float(), x_ref.grad, atol=1e-3, rtol=1e-3)
+    assert torch.allclose(dgamma, w_ref.grad, atol=1e-2, rtol=1e-3)
+    assert torch.allclose(dbeta, b_ref.grad, atol=1e-2, rtol=1e-3)
+
+
+def test_bn_act_autograd_function():
+    from deepfake_detection_amd.ops.bn_act import fused_bn_act
+
+    torch.manual_seed(3)
+    N, C, H, W = 2, 24, 9, 9
+    x = _cl(torch.randn(N, C, H, W, device="cuda")).requires_grad_(True)
+    bn = torch.nn.BatchNorm2d(C, momentum=0.01, eps=1e-3).cuda()
+    y = fused_bn_act(x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
+                     True, 0.01, 1e-3, "silu")
+    loss = (y ** 2).sum()
+    loss.backward()
+    assert x.grad is not None and torch.isfinite(x.grad).all()
+    assert bn.weight.grad is not None and torch.isfinite(bn.weight.grad).all()
+
+
+# ---------------------------------------------------------------------------
+# SE
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("dtype", [torch.float32, torch.bfloat16])
+def test_se_forward_backward(ext, dtype):
+    from deepfake_detection_amd.ops.se import fused_se
+
+    torch.manual_seed(4)
+    N, C, H, W, Cr = 3, 64, 11, 13, 16
+    x0 = torch.randn(N, C, H, W, device="cuda", dtype=dtype)
+    w1 = torch.randn(Cr, C, 1, 1, device="cuda") * 0.1
+    b1 = torch.randn(Cr, device="cuda") * 0.1
+    w2 = torch.randn(C, Cr, 1, 1, device="cuda") * 0.1
+    b2 = torch.randn(C, device="cuda") * 0.1
+
+    x = _cl(x0.clone()).requires_grad_(True)
+    w1_p = w1.clone().requires_grad_(True)
+    b1_p = b1.clone().requires_grad_(True)
+    w2_p = w2.clone().requires_grad_(True)
+    b2_p = b2.clone().requires_grad_(True)
+    y = fused_se(x, w1_p, b1_p, w2_p, b2_p, "silu")
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    # fp32 torch reference
+    xr = x0.float().detach().requires_grad_(True)
+    w1r = w1.detach().requires_grad_(True)
+    b1r = b1.detach().requires_grad_(True)
+    w2r = w2.detach().requires_grad_(True)
+    b2r = b2.detach().requires_grad_(True)
+    s = xr.mean(dim=(2, 3), keepdim=True)
+    t = torch.nn.functional.conv2d(s, w1r, b1r)
+    t = torch.nn.functional.silu(t)
+    t = torch.nn.functional.conv2d(t, w2r, b2r)
+    yr = xr * torch.sigmoid(t)
+    yr.backward(dy.float())
+
+    tol = 1e-4 if dtype == torch.float32 else 0.05
+    assert torch.allclose(y.float(), yr.detach(), atol=tol, rtol=tol)
+    assert torch.allclose(x.grad.float(), xr.grad, atol=tol * 10, rtol=tol * 10)
+    assert torch.allclose(w1_p.grad.float(), w1r.grad, atol=0.05, rtol=0.02)
+    assert torch.allclose(w2_p.grad.float(), w2r.grad, atol=0.05, rtol=0.02)
+    assert torch.allclose(b1_p.grad.float(), b1r.grad, atol=0.05, rtol=0.02)
+    assert torch.allclose(b2_p.grad.float(), b2r.grad, atol=0.05, rtol=0.02)
+
+
+# ---------------------------------------------------------------------------
+# global avg pool
+# ---------------------------------------------------------------------------
+def test_global_avg_pool(ext):
+    from deepfake_detection_amd.ops.pool import fused_global_avg_pool
+
+    torch.manual_seed(5)
+    x = _cl(torch.randn(4, 96, 19, 19, device="cuda")).requires_grad_(True)
+    y = fused_global_avg_pool(x)
+    assert y.shape == (4, 96)
+    ref = x.float().mean(dim=(2, 3))
+    assert torch.allclose(y.float(), ref, atol=1e-5, rtol=1e-5)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    xr = x.detach().clone().requires_grad_(True)
+    xr.mean(dim=(2, 3)).backward(dy)
+    assert torch.allclose(x.grad, xr.grad, atol=1e-6)
+
+
+# ---------------------------------------------------------------------------
+# fused optimizers
+# ---------------------------------------------------------------------------
+def test_rmsproptf_fused_matches_cpu():
+    from deepfake_detection_amd.optim import RMSpropTF
+
+    torch.manual_seed(6)
+    shapes = [(37,), (16, 8), (128,), (5, 5, 3, 3)]
+    cpu_params = [torch.nn.Parameter(torch.randn(s)) for s in shapes]
+    gpu_params = [torch.nn.Parameter(p.detach().clone().cuda()) for p in cpu_params]
+    kw = dict(lr=0.05, alpha=0.9, eps=1e-3, momentum=0.9, weight_decay=1e-4)
+    opt_cpu = RMSpropTF(cpu_params, **kw)
+    opt_gpu = RMSpropTF(gpu_params, **kw)
+    for step in range(5):
+        torch.manual_seed(100 + step)
+        grads = [torch.randn(s) for s in shapes]
+        for p, g in zip(cpu_params, grads):
+            p.grad = g.clone()
+        for p, g in zip(gpu_params, grads):
+            p.grad = g.clone().cuda()
+        opt_cpu.step()
+        opt_gpu.step()
+    for pc, pg in zip(cpu_params, gpu_params):
+        assert torch.allclose(pc.detach(), pg.detach().cpu(), atol=1e-5, rtol=1e-5)
+
+
+def test_adamw_fused_matches_cpu():
+    from deepfake_detection_amd.optim import AdamW
+
+    torch.manual_seed(7)
+    shapes = [(41,), (8, 8)]
+    cpu_params = [torch.nn.Parameter(torch.randn(s)) for s in shapes]
+    gpu_params = [torch.nn.Parameter(p.detach().clone().cuda()) for p in cpu_params]
+    kw = dict(lr=0.01, weight_decay=0.1)
+    opt_cpu = AdamW(cpu_params, **kw)
+    opt_gpu = AdamW(gpu_params, **kw)
+    for step in range(4):
+        torch.manual_seed(200 + step)
+        grads = [torch.randn(s) for s in shapes]
+        for p, g in zip(cpu_params, grads):
+            p.grad = g.clone()
+        for p, g in zip(gpu_params, grads):
+            p.grad = g.clone().cuda()
+        opt_cpu.step()
+        opt_gpu.step()
+    for pc, pg in zip(cpu_params, gpu_params):
+        assert torch.allclose(pc.detach(), pg.detach().cpu(), atol=1e-5, rtol=1e-5)
+
+
+def test_ema_fused(ext):
+    e = [torch.ones(100, device="cuda"), torch.full((31,), 2.0, device="cuda")]
+    m = [torch.full((100,), 2.0, device="cuda"), torch.full((31,), 4.0, device="cuda")]
+    ext.ema_multi_tensor(e, m, 0.9)
+    assert torch.allclose(e[0], torch.full((100,), 1.1, device="cuda"))
+    assert torch.allclose(e[1], torch.full((31,), 2.2, device="cuda"))
+
+
+# ---------------------------------------------------------------------------
+# end-to-end model A/B: fused HIP path vs plain torch ops on the same device
+# ---------------------------------------------------------------------------
+def test_model_fused_vs_torch_path():
+    import deepfake_detection_amd as dfd
+
+    torch.manual_seed(8)
+    model = dfd.create_model("efficientnet_b0", num_classes=2).cuda().eval()
+    model = model.to(memory_format=torch.channels_last)
+    x = torch.randn(2, 3, 224, 224, device="cuda").contiguous(
+        memory_format=torch.channels_last)
+
+    with torch.no_grad():
+        y_fused = model(x)
+        os.environ["DFD_AMD_FORCE_TORCH_OPS"] = "1"
+        try:
+            y_torch = model(x)
+        finally:
+            del os.environ["DFD_AMD_FORCE_TORCH_OPS"]
+    assert torch.allclose(y_fused, y_torch, atol=1e-3, rtol=1e-3)
+
+
+def test_train_step_gpu():
+    """One forward+backward+step of the production model family on GPU,
+    bf16 autocast, fused ops + fused optimizer."""
+    import deepfake_detection_amd as dfd
+    from deepfake_detection_amd.optim import RMSpropTF
+
+    torch.manual_seed(9)
+    model = dfd.create_model("efficientnet_b0", num_classes=2, in_chans=12).cuda()
+    model = model.to(memory_format=torch.channels_last)
+    opt = RMSpropTF(model.parameters(), lr=1e-4, alpha=0.9, eps=1e-3, momentum=0.9)
+    x = torch.randn(4, 12, 64, 64, device="cuda").contiguous(
+        memory_format=torch.channels_last)
+    t = torch.randint(0, 2, (4,), device="cuda")
+    with torch.autocast("cuda", torch.bfloat16):
+        out = model(x)
+        loss = torch.nn.functional.cross_entropy(out, t)
+    loss.backward()
+    opt.step()
+    torch.cuda.synchronize()
+    assert torch.isfinite(loss)
