@@ -15,6 +15,9 @@ from .helpers import (  # noqa: F401
     load_pretrained,
 )
 from . import efficientnet  # noqa: F401  (registers entrypoints)
+from . import mobilenetv3  # noqa: F401
+from . import resnet  # noqa: F401
+from . import xception  # noqa: F401
 from .factory import (  # noqa: F401
     create_model,
     create_deepfake_model,

@@ -137,7 +137,9 @@ void rmsprop_tf_multi_tensor(std::vector<at::Tensor> params, std::vector<at::Ten
                              double lr, double alpha, double eps, double momentum,
                              double weight_decay, bool decoupled_decay, bool lr_in_momentum) {
   TORCH_CHECK(!params.empty());
-  for (auto& p : params) TORCH_CHECK(p.scalar_type() == at::kFloat && p.is_contiguous());
+  for (auto& p : params)
+    TORCH_CHECK(p.scalar_type() == at::kFloat && p.is_non_overlapping_and_dense(),
+                "rmsprop_tf_multi_tensor: dense fp32 params required");
   auto table = build_chunks(params, grads, square_avgs, momentum_buffers, params[0].options());
   auto stream = at::cuda::getCurrentHIPStream();
   hipLaunchKernelGGL(rmsprop_tf_kernel, dim3(std::min(table.nchunks, kMaxGrid)), dim3(256), 0,
@@ -151,7 +153,9 @@ void adamw_multi_tensor(std::vector<at::Tensor> params, std::vector<at::Tensor> 
                         long long step, double lr, double beta1, double beta2,
                         double eps, double weight_decay) {
   TORCH_CHECK(!params.empty());
-  for (auto& p : params) TORCH_CHECK(p.scalar_type() == at::kFloat && p.is_contiguous());
+  for (auto& p : params)
+    TORCH_CHECK(p.scalar_type() == at::kFloat && p.is_non_overlapping_and_dense(),
+                "adamw_multi_tensor: dense fp32 params required");
   auto table = build_chunks(params, grads, exp_avgs, exp_avg_sqs, params[0].options());
   auto stream = at::cuda::getCurrentHIPStream();
   const float bias_c1 = 1.f - powf((float)beta1, (float)step);

@@ -55,7 +55,9 @@ class AdamW(Optimizer):
                         state["max_exp_avg_sq"] = torch.zeros_like(p)
                 state["step"] += 1
 
-                if p.is_cuda and not amsgrad:
+                from .rmsprop_tf import _fusable
+
+                if p.is_cuda and not amsgrad and _fusable(p, grad, state):
                     fused_bucket.append((p, grad, state))
                     continue
 

@@ -16,6 +16,20 @@ import torch
 from torch.optim import Optimizer
 
 
+def _fusable(p, grad, state):
+    """Fused path needs param/grad/state to share one dense memory layout
+    (plain or channels_last contiguous — the elementwise update is
+    layout-agnostic as long as all tensors agree)."""
+    if not (p.dtype == torch.float32 and p.is_non_overlapping_and_dense()):
+        return False
+    if grad.stride() != p.stride():
+        return False
+    for v in state.values():
+        if isinstance(v, torch.Tensor) and v.stride() != p.stride():
+            return False
+    return True
+
+
 class RMSpropTF(Optimizer):
     def __init__(self, params, lr=1e-2, alpha=0.9, eps=1e-10, weight_decay=0,
                  momentum=0.0, centered=False, decoupled_decay=False, lr_in_momentum=True):
@@ -69,7 +83,7 @@ class RMSpropTF(Optimizer):
                         state["grad_avg"] = torch.zeros_like(p)
                 state["step"] += 1
 
-                if p.is_cuda and not group["centered"]:
+                if p.is_cuda and not group["centered"] and _fusable(p, grad, state):
                     fused_bucket.append((p, grad, state))
                     continue
 
