@@ -1,7 +1,19 @@
 from .config import resolve_data_config  # noqa: F401
 from .constants import *  # noqa: F401,F403
+from .auto_augment import (  # noqa: F401
+    AutoAugment,
+    RandAugment,
+    AugMixAugment,
+    auto_augment_transform,
+    rand_augment_transform,
+    augment_and_mix_transform,
+)
 from .dataset import (  # noqa: F401
+    AugMixDataset,
+    ConcatDataset,
     Dataset,
+    DatasetTar,
+    DeepFakeDataset_v1,
     DeepFakeDataset_v3,
     SyntheticDeepFakeDataset,
 )

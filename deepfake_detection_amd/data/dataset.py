@@ -241,6 +241,152 @@ class DeepFakeDataset_v3(data.Dataset):
         self.transform = transform
 
 
+class DeepFakeDataset_v1(data.Dataset):
+    """Legacy pair-file dataset (reference dataset.py:531): `fake:real:rotated`
+    lines; even indices yield the fake image (label 0), odd the real (1)."""
+
+    def __init__(self, root, result_file, load_bytes=False, transform=None,
+                 transform_rotateds=None, class_map=""):
+        self.root = root
+        self.load_bytes = load_bytes
+        self.transform = transform
+        self.transform_rotateds = transform_rotateds
+        self.results = []
+        with open(result_file, "r") as f:
+            for line in f.readlines():
+                line_s = line.strip().split(":")
+                if len(line_s) != 3:
+                    continue
+                self.results.append((line_s[0], line_s[1], int(line_s[2])))
+        if len(self.results) == 0:
+            raise RuntimeError(f"Found 0 entries in {result_file}")
+
+    def __getitem__(self, index):
+        target = index % 2
+        result_index = index // 2
+        img_path = self.results[result_index][target]
+        rotated = self.results[result_index][2]
+        img = open(img_path, "rb").read() if self.load_bytes else Image.open(img_path).convert("RGB")
+        if self.transform_rotateds is not None:
+            img = self.transform_rotateds[rotated](img)
+        if self.transform is not None:
+            img = self.transform(img)
+        return img, target
+
+    def __len__(self):
+        return 2 * len(self.results)
+
+    def set_transform(self, transform, transform_rotateds=None):
+        self.transform = transform
+        if transform_rotateds is not None:
+            self.transform_rotateds = transform_rotateds
+
+
+class ConcatDataset(data.ConcatDataset):
+    """Concatenation with transform pass-through (reference dataset.py:229)."""
+
+    def set_transform(self, transform):
+        for ds in self.datasets:
+            if hasattr(ds, "set_transform"):
+                ds.set_transform(transform)
+            else:
+                ds.transform = transform
+
+
+class DatasetTar(data.Dataset):
+    """Tar-archive image dataset (reference dataset.py:602): folder name
+    inside the archive is the class label; the tar handle is reopened
+    lazily per worker process."""
+
+    def __init__(self, root, load_bytes=False, transform=None, class_map=""):
+        import os as _os
+        import tarfile
+
+        assert _os.path.isfile(root)
+        self.root = root
+        class_to_idx = load_class_map(class_map) if class_map else None
+        with tarfile.open(root) as tf:
+            files, labels = [], []
+            for ti in tf.getmembers():
+                if not ti.isfile():
+                    continue
+                dirname, basename = _os.path.split(ti.path)
+                ext = _os.path.splitext(basename)[1]
+                if ext.lower() in IMG_EXTENSIONS:
+                    files.append(ti)
+                    labels.append(_os.path.basename(dirname))
+            if class_to_idx is None:
+                sorted_labels = sorted(set(labels), key=natural_key)
+                class_to_idx = {c: idx for idx, c in enumerate(sorted_labels)}
+            self.samples = sorted(
+                zip(files, [class_to_idx[l] for l in labels]),
+                key=lambda k: natural_key(k[0].path))
+            self.class_to_idx = class_to_idx
+        self.tarfile = None  # lazy re-open per worker
+        self.load_bytes = load_bytes
+        self.transform = transform
+
+    def __getitem__(self, index):
+        import tarfile
+
+        if self.tarfile is None:
+            self.tarfile = tarfile.open(self.root)
+        tarinfo, target = self.samples[index]
+        iob = self.tarfile.extractfile(tarinfo)
+        img = iob.read() if self.load_bytes else Image.open(iob).convert("RGB")
+        if self.transform is not None:
+            img = self.transform(img)
+        return img, target
+
+    def __len__(self):
+        return len(self.samples)
+
+    def set_transform(self, transform):
+        self.transform = transform
+
+
+class AugMixDataset(data.Dataset):
+    """Clean + augmented split wrapper for AugMix JSD training
+    (reference dataset.py:633): transform is a (base, augmentation,
+    normalize) triple; item = (clean, aug1, ..., augN-1), label."""
+
+    def __init__(self, dataset, num_splits=2):
+        self.augmentation = None
+        self.normalize = None
+        self.dataset = dataset
+        if self.dataset.transform is not None:
+            self._set_transforms(self.dataset.transform)
+        self.num_splits = num_splits
+
+    def _set_transforms(self, x):
+        assert isinstance(x, (list, tuple)) and len(x) == 3, \
+            "Expecting a tuple/list of 3 transforms"
+        self.dataset.transform = x[0]
+        self.augmentation = x[1]
+        self.normalize = x[2]
+
+    @property
+    def transform(self):
+        return self.dataset.transform
+
+    @transform.setter
+    def transform(self, x):
+        self._set_transforms(x)
+
+    def _normalize(self, x):
+        return x if self.normalize is None else self.normalize(x)
+
+    def __getitem__(self, i):
+        x, y = self.dataset[i]
+        x_list = [self._normalize(x)]  # clean split
+        for _ in range(self.num_splits - 1):
+            x_list.append(self._normalize(self.augmentation(x)))
+        return tuple(x_list), y
+
+    def __len__(self):
+        return len(self.dataset)
+
+
 class SyntheticDeepFakeDataset(data.Dataset):
     """Synthetic face-crop dataset for benchmarking (no disk, no network):
     deterministic pseudo-random uint8 frame groups shaped like

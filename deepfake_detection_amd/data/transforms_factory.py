@@ -71,12 +71,37 @@ def transforms_deepfake_eval_v3(img_size=224, use_prefetcher=True):
 
 def transforms_imagenet_train(
         img_size=224, scale=(0.08, 1.0), color_jitter=0.4, interpolation="random",
-        random_erasing=0.0, random_erasing_mode="const", use_prefetcher=False,
-        mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD):
+        auto_augment=None, random_erasing=0.0, random_erasing_mode="const",
+        use_prefetcher=False, mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD):
     tfl = [
         RandomResizedCropAndInterpolation(img_size, scale=scale, interpolation=interpolation),
         transforms.RandomHorizontalFlip(),
     ]
+    if auto_augment:
+        # AA/RandAugment/AugMix replace color jitter
+        # (reference transforms_factory.py:269-287)
+        from .auto_augment import (
+            augment_and_mix_transform,
+            auto_augment_transform,
+            rand_augment_transform,
+        )
+
+        assert isinstance(auto_augment, str)
+        img_size_min = min(img_size) if isinstance(img_size, (tuple, list)) else img_size
+        aa_params = dict(
+            translate_const=int(img_size_min * 0.45),
+            img_mean=tuple(int(round(255 * x)) for x in mean),
+        )
+        if interpolation and interpolation != "random":
+            aa_params["interpolation"] = _pil_interp(interpolation)
+        if auto_augment.startswith("rand"):
+            tfl += [rand_augment_transform(auto_augment, aa_params)]
+        elif auto_augment.startswith("augmix"):
+            aa_params["translate_pct"] = 0.3
+            tfl += [augment_and_mix_transform(auto_augment, aa_params)]
+        else:
+            tfl += [auto_augment_transform(auto_augment, aa_params)]
+        color_jitter = None
     if color_jitter is not None:
         if isinstance(color_jitter, (list, tuple)):
             assert len(color_jitter) in (3, 4)
@@ -125,8 +150,8 @@ def transforms_imagenet_eval(
 
 def create_transform(
         input_size, is_training=False, use_prefetcher=False, color_jitter=0.4,
-        interpolation="bilinear", mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD,
-        crop_pct=None, tf_preprocessing=False):
+        auto_augment=None, interpolation="bilinear", mean=IMAGENET_DEFAULT_MEAN,
+        std=IMAGENET_DEFAULT_STD, crop_pct=None, tf_preprocessing=False):
     if isinstance(input_size, tuple):
         img_size = input_size[-2:]
     else:
@@ -134,7 +159,7 @@ def create_transform(
 
     if is_training:
         return transforms_imagenet_train(
-            img_size, color_jitter=color_jitter,
+            img_size, color_jitter=color_jitter, auto_augment=auto_augment,
             interpolation=interpolation if interpolation != "bilinear" else "random",
             use_prefetcher=use_prefetcher, mean=mean, std=std)
     return transforms_imagenet_eval(

@@ -258,8 +258,16 @@ def main(rank, args, args_text, world_size=None, start_rank=0):
             args.data, args.class_names, train_split=True,
             train_ratio=1.0 - args.validation_frac, random_state=args.seed,
             is_training=False, label_balance=args.label_balance)
+        # collate-time mixup when the device prefetcher owns normalization
+        # (reference train.py:442-445)
+        collate_fn = None
+        if args.prefetcher and args.mixup > 0:
+            from ..data import FastCollateMixup
+
+            collate_fn = FastCollateMixup(args.mixup, args.smoothing, args.num_classes)
         loader_train = create_deepfake_loader_v3(
             dataset_train, input_size=data_config["input_size"], batch_size=args.batch_size,
+            collate_fn=collate_fn,
             is_training=True, use_prefetcher=args.prefetcher, re_prob=args.reprob,
             re_mode=args.remode, re_count=args.recount, re_split=args.resplit,
             re_max=args.remax, color_jitter=args.color_jitter,
@@ -314,6 +322,11 @@ def main(rank, args, args_text, world_size=None, start_rank=0):
 
             if args.distributed and args.dist_bn in ("broadcast", "reduce"):
                 distribute_bn(model, world_size, args.dist_bn == "reduce")
+
+            # mixup cutoff (reference train.py:597-599)
+            if args.prefetcher and args.mixup > 0 and getattr(loader_train, "mixup_enabled", False):
+                if args.mixup_off_epoch and epoch >= args.mixup_off_epoch:
+                    loader_train.mixup_enabled = False
 
             train_metrics = train_epoch(
                 epoch, model, loader_train, optimizer, train_loss_fn, args, device,

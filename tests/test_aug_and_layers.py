@@ -1,0 +1,150 @@
+"""AutoAugment engine + auxiliary layer tests."""
+
+import numpy as np
+import pytest
+import torch
+from PIL import Image
+
+from deepfake_detection_amd.data.auto_augment import (
+    AugmentOp,
+    augment_and_mix_transform,
+    auto_augment_transform,
+    rand_augment_transform,
+)
+from deepfake_detection_amd.models.layers_extra import (
+    AvgPool2dSame,
+    CbamModule,
+    EcaModule,
+    FeatureHooks,
+    MedianPool2d,
+    SelectiveKernelConv,
+    SEModule,
+    SplitBatchNorm2d,
+    TestTimePoolHead,
+    convert_splitbn_model,
+)
+
+
+def _img(seed=0, size=32):
+    rng = np.random.RandomState(seed)
+    return Image.fromarray(rng.randint(0, 255, (size, size, 3), dtype=np.uint8))
+
+
+def test_augment_ops_run():
+    img = _img()
+    for name in ["Rotate", "ShearX", "TranslateXRel", "Solarize", "SolarizeAdd",
+                 "PosterizeTpu", "Color", "Contrast", "Brightness", "Sharpness",
+                 "AutoContrast", "Equalize", "Invert"]:
+        out = AugmentOp(name, prob=1.0, magnitude=7)(img)
+        assert out.size == img.size
+
+
+@pytest.mark.parametrize("cfg,builder", [
+    ("original", auto_augment_transform),
+    ("v0-mstd0.5", auto_augment_transform),
+    ("rand-m9-n2-mstd0.5", rand_augment_transform),
+    ("rand-m7-n3-w0", rand_augment_transform),
+    ("augmix-m3-w3", augment_and_mix_transform),
+])
+def test_policy_configs(cfg, builder):
+    t = builder(cfg, {"translate_const": 10, "img_mean": (128, 128, 128)})
+    out = t(_img())
+    assert out.size == (32, 32)
+
+
+def test_create_transform_with_aa():
+    from deepfake_detection_amd.data import create_transform
+
+    t = create_transform((3, 48, 48), is_training=True, auto_augment="rand-m5-n2")
+    out = t(_img(size=64))
+    assert out.shape == (3, 48, 48)
+
+
+def test_split_batchnorm():
+    bn = SplitBatchNorm2d(8, num_splits=2)
+    x = torch.randn(4, 8, 5, 5)
+    y = bn(x)
+    assert y.shape == x.shape
+    m = torch.nn.Sequential(torch.nn.Conv2d(3, 8, 3), torch.nn.BatchNorm2d(8))
+    m2 = convert_splitbn_model(m, num_splits=2)
+    assert isinstance(m2[1], SplitBatchNorm2d)
+    # eval path uses main stats only
+    m2.eval()
+    assert m2(torch.randn(2, 3, 8, 8)).shape == (2, 8, 6, 6)
+
+
+@pytest.mark.parametrize("mod", [
+    lambda: SEModule(32),
+    lambda: EcaModule(32),
+    lambda: CbamModule(32),
+])
+def test_attention_modules(mod):
+    m = mod()
+    x = torch.randn(2, 32, 7, 7)
+    y = m(x)
+    assert y.shape == x.shape
+
+
+def test_selective_kernel():
+    m = SelectiveKernelConv(16, 32, stride=2)
+    y = m(torch.randn(2, 16, 16, 16))
+    assert y.shape == (2, 32, 8, 8)
+
+
+def test_avg_pool_same():
+    m = AvgPool2dSame(3, stride=2)
+    assert m(torch.randn(1, 4, 7, 7)).shape == (1, 4, 4, 4)
+
+
+def test_median_pool():
+    m = MedianPool2d(3, same=True)
+    x = torch.randn(1, 2, 9, 9)
+    assert m(x).shape == x.shape
+
+
+def test_test_time_pool_head():
+    import deepfake_detection_amd as dfd
+
+    model = dfd.create_model("efficientnet_lite0", num_classes=5)
+    head = TestTimePoolHead(model, original_pool=7)
+    y = head(torch.randn(1, 3, 256, 256))
+    assert y.shape == (1, 5)
+
+
+def test_feature_hooks():
+    m = torch.nn.Sequential(torch.nn.Conv2d(3, 4, 3), torch.nn.Conv2d(4, 8, 3))
+    hooks = FeatureHooks([{"name": "0"}, {"name": "1"}], m.named_modules())
+    x = torch.randn(1, 3, 16, 16)
+    m(x)
+    outs = hooks.get_output(x.device)
+    assert len(outs) == 2 and outs[0].shape[1] == 4 and outs[1].shape[1] == 8
+
+
+def test_legacy_datasets(tmp_path):
+    import tarfile
+
+    from deepfake_detection_amd.data import DatasetTar, DeepFakeDataset_v1
+
+    # DeepFakeDataset_v1 pair file
+    fake = tmp_path / "f.jpg"
+    real = tmp_path / "r.jpg"
+    _img(1).save(fake)
+    _img(2).save(real)
+    pair_file = tmp_path / "pairs.txt"
+    pair_file.write_text(f"{fake}:{real}:0\n")
+    ds = DeepFakeDataset_v1(str(tmp_path), str(pair_file))
+    assert len(ds) == 2
+    img0, y0 = ds[0]
+    img1, y1 = ds[1]
+    assert (y0, y1) == (0, 1)
+
+    # DatasetTar
+    tar_path = tmp_path / "data.tar"
+    d = tmp_path / "cls_a"
+    d.mkdir()
+    _img(3).save(d / "0.jpg")
+    with tarfile.open(tar_path, "w") as tf:
+        tf.add(d / "0.jpg", arcname="cls_a/0.jpg")
+    dt = DatasetTar(str(tar_path))
+    img, y = dt[0]
+    assert y == 0 and img.size == (32, 32)
