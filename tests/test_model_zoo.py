@@ -42,3 +42,13 @@ def test_xception_fp16_inference_path():
     with torch.no_grad():
         y = m(torch.randn(1, 3, 299, 299).half())
     assert y.shape == (1, 2)
+
+
+@pytest.mark.parametrize("name,expect", [
+    ("densenet121", 7978856),
+    ("densenet161", 28681000),
+    ("inception_v3", 23834568),
+])
+def test_param_parity_extra(name, expect):
+    m = dfd.create_model(name)
+    assert sum(p.numel() for p in m.parameters()) == expect
