@@ -20,7 +20,9 @@ def _fusable(p, grad, state):
     """Fused path needs param/grad/state to share one dense memory layout
     (plain or channels_last contiguous — the elementwise update is
     layout-agnostic as long as all tensors agree)."""
-    if not (p.dtype == torch.float32 and p.is_non_overlapping_and_dense()):
+    dense = p.is_contiguous() or p.is_contiguous(memory_format=torch.channels_last) \
+        or (p.dim() == 5 and p.is_contiguous(memory_format=torch.channels_last_3d))
+    if not (p.dtype == torch.float32 and dense):
         return False
     if grad.stride() != p.stride():
         return False

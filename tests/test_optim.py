@@ -121,3 +121,16 @@ def test_factory_dispatch_and_wd_split():
     opt2 = create_optimizer(_args(opt="lookahead_adamw"), m)
     assert isinstance(opt2, Lookahead)
     assert isinstance(opt2.base_optimizer, AdamW)
+
+
+def test_fusable_runs_on_cpu_layouts():
+    from deepfake_detection_amd.optim.rmsprop_tf import _fusable
+
+    p = torch.nn.Parameter(torch.randn(4, 8, 3, 3))
+    g = torch.randn_like(p)
+    state = {"step": 1, "square_avg": torch.ones_like(p)}
+    assert _fusable(p, g, state)
+    p_cl = torch.nn.Parameter(p.detach().to(memory_format=torch.channels_last))
+    g_cl = g.to(memory_format=torch.channels_last)
+    assert _fusable(p_cl, g_cl, {"square_avg": torch.ones_like(p_cl)})
+    assert not _fusable(p_cl, g, {"square_avg": torch.ones_like(p_cl)})  # mismatched layouts
