@@ -24,6 +24,7 @@ from . import senet  # noqa: F401
 from . import dpn  # noqa: F401
 from . import res2net  # noqa: F401
 from . import sknet  # noqa: F401
+from . import inception_v4  # noqa: F401
 from .factory import (  # noqa: F401
     create_model,
     create_deepfake_model,
