@@ -24,6 +24,7 @@ __all__ = [
     "pad_same",
     "conv2d_same",
     "Conv2dSame",
+    "DepthwiseConv2d",
     "create_conv2d_pad",
     "MixedConv2d",
     "CondConv2d",
@@ -80,8 +81,23 @@ def pad_same(x, k: List[int], s: List[int], d: List[int] = (1, 1), value: float 
     return x
 
 
+def _dw_hip_path(x, weight, stride, padding, dilation, groups) -> bool:
+    """True when this conv should run on the hand-written gfx950 depthwise
+    kernels (ops/dwconv.py) instead of MIOpen grouped conv."""
+    if not x.is_cuda:
+        return False
+    from ..ops.dwconv import dw_supported
+    from ..ops.extension import gpu_ops_required
+
+    return gpu_ops_required() and dw_supported(weight, stride, padding, dilation, groups)
+
+
 def conv2d_same(x, weight, bias=None, stride=(1, 1), padding=(0, 0), dilation=(1, 1), groups=1):
     x = pad_same(x, weight.shape[-2:], stride, dilation)
+    if _dw_hip_path(x, weight, stride, (0, 0), dilation, groups):
+        from ..ops.dwconv import dw_conv2d
+
+        return dw_conv2d(x, weight, bias, stride, (0, 0), dilation)
     return F.conv2d(x, weight, bias, stride, (0, 0), dilation, groups)
 
 
@@ -96,6 +112,19 @@ class Conv2dSame(nn.Conv2d):
     def forward(self, x):
         return conv2d_same(x, self.weight, self.bias, self.stride,
                            self.padding, self.dilation, self.groups)
+
+
+class DepthwiseConv2d(nn.Conv2d):
+    """nn.Conv2d with groups == in == out, routed to the gfx950 HIP depthwise
+    kernels on ROCm devices (same state_dict as nn.Conv2d)."""
+
+    def forward(self, x):
+        if _dw_hip_path(x, self.weight, self.stride, self.padding, self.dilation, self.groups):
+            from ..ops.dwconv import dw_conv2d
+
+            return dw_conv2d(x, self.weight, self.bias, self.stride, self.padding,
+                             self.dilation)
+        return super().forward(x)
 
 
 def get_padding_value(padding, kernel_size, **kwargs) -> Tuple[object, bool]:
@@ -123,6 +152,8 @@ def create_conv2d_pad(in_chs, out_chs, kernel_size, **kwargs):
     padding, is_dynamic = get_padding_value(padding, kernel_size, **kwargs)
     if is_dynamic:
         return Conv2dSame(in_chs, out_chs, kernel_size, **kwargs)
+    if (kwargs.get("groups", 1) == in_chs and in_chs == out_chs and in_chs > 1):
+        return DepthwiseConv2d(in_chs, out_chs, kernel_size, padding=padding, **kwargs)
     return nn.Conv2d(in_chs, out_chs, kernel_size, padding=padding, **kwargs)
 
 

@@ -1,0 +1,413 @@
+// Depthwise 2-D convolution for gfx950 (CDNA4), NHWC (torch channels_last).
+//
+// Replaces the MIOpen grouped-conv path for depthwise convs (groups == C):
+// the reference delegates these to cuDNN (SURVEY.md §2.6 item 3,
+// dfd/timm/models/layers/create_conv2d.py:24-25); on ROCm the MIOpen/CK
+// grouped bwd-weight kernel dominated the whole training step (87% of GPU
+// time, profiles/r01_bench_b4_299_bs192_top_kernels.md), so all three
+// passes are hand-written here.
+//
+// Layouts:
+//   x, y, dy, dx : NHWC, C fastest (channels_last), dtype bf16/fp16/fp32
+//   weight       : repacked by the python wrapper to (KH, KW, C), same dtype
+//   dweight      : (KH, KW, C) fp32 (wrapper casts/permutes back)
+//
+// Design: bandwidth-bound op. Each thread owns a VEC-wide contiguous channel
+// slice (VEC chosen so C % VEC == 0; 16 B loads for bf16 at VEC=8) and
+// accumulates in fp32 registers. bwd-weight blocks are (kh, channel-vector)
+// grids of threads looping over a spatial chunk, finishing with one fp32
+// atomicAdd per (kh,kw,c) — K*VEC atomics per thread total.
+
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+
+#include "common.h"
+
+namespace {
+
+template <typename T, int N>
+struct alignas(sizeof(T) * N) TVec {
+  T v[N];
+};
+
+template <typename T, int N>
+DFD_DEV TVec<T, N> vload(const T* p) {
+  return *reinterpret_cast<const TVec<T, N>*>(p);
+}
+
+template <typename T, int N>
+DFD_DEV void vstore(T* p, const TVec<T, N>& x) {
+  *reinterpret_cast<TVec<T, N>*>(p) = x;
+}
+
+// ---------------------------------------------------------------------------
+// forward: y[n,ho,wo,c] = sum_kh,kw x[n, ho*sh-ph+kh, wo*sw-pw+kw, c] * w[kh,kw,c]
+// ---------------------------------------------------------------------------
+template <typename T, int K, int VEC>
+__global__ void dw_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                              T* __restrict__ y, int N, int C, int H, int W,
+                              int Ho, int Wo, int sh, int sw, int ph, int pw) {
+  const int cv = C / VEC;
+  const long long total = (long long)N * Ho * Wo * cv;
+  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long long)gridDim.x * blockDim.x) {
+    const int c = (int)(idx % cv) * VEC;
+    long long p = idx / cv;
+    const int wo = (int)(p % Wo);
+    p /= Wo;
+    const int ho = (int)(p % Ho);
+    const int n = (int)(p / Ho);
+
+    float acc[VEC];
+#pragma unroll
+    for (int i = 0; i < VEC; ++i) acc[i] = 0.f;
+
+    const int hi0 = ho * sh - ph;
+    const int wi0 = wo * sw - pw;
+#pragma unroll
+    for (int kh = 0; kh < K; ++kh) {
+      const int hi = hi0 + kh;
+      if (hi < 0 || hi >= H) continue;
+#pragma unroll
+      for (int kw = 0; kw < K; ++kw) {
+        const int wi = wi0 + kw;
+        if (wi < 0 || wi >= W) continue;
+        const TVec<T, VEC> xv = vload<T, VEC>(x + (((long long)n * H + hi) * W + wi) * C + c);
+        const TVec<T, VEC> wv = vload<T, VEC>(w + ((long long)kh * K + kw) * C + c);
+#pragma unroll
+        for (int i = 0; i < VEC; ++i)
+          acc[i] += DfdCvt<T>::to_f32(xv.v[i]) * DfdCvt<T>::to_f32(wv.v[i]);
+      }
+    }
+    TVec<T, VEC> yv;
+#pragma unroll
+    for (int i = 0; i < VEC; ++i) yv.v[i] = DfdCvt<T>::from_f32(acc[i]);
+    vstore<T, VEC>(y + (((long long)n * Ho + ho) * Wo + wo) * C + c, yv);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// backward data:
+// dx[n,hi,wi,c] = sum over (kh,kw) with hi = ho*sh-ph+kh solvable:
+//                 dy[n,ho,wo,c] * w[kh,kw,c]
+// ---------------------------------------------------------------------------
+template <typename T, int K, int VEC>
+__global__ void dw_bwd_data_kernel(const T* __restrict__ dy, const T* __restrict__ w,
+                                   T* __restrict__ dx, int N, int C, int H, int W,
+                                   int Ho, int Wo, int sh, int sw, int ph, int pw) {
+  const int cv = C / VEC;
+  const long long total = (long long)N * H * W * cv;
+  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long long)gridDim.x * blockDim.x) {
+    const int c = (int)(idx % cv) * VEC;
+    long long p = idx / cv;
+    const int wi = (int)(p % W);
+    p /= W;
+    const int hi = (int)(p % H);
+    const int n = (int)(p / H);
+
+    float acc[VEC];
+#pragma unroll
+    for (int i = 0; i < VEC; ++i) acc[i] = 0.f;
+
+#pragma unroll
+    for (int kh = 0; kh < K; ++kh) {
+      const int th = hi + ph - kh;
+      if (th < 0 || th % sh) continue;
+      const int ho = th / sh;
+      if (ho >= Ho) continue;
+#pragma unroll
+      for (int kw = 0; kw < K; ++kw) {
+        const int tw = wi + pw - kw;
+        if (tw < 0 || tw % sw) continue;
+        const int wo = tw / sw;
+        if (wo >= Wo) continue;
+        const TVec<T, VEC> gv = vload<T, VEC>(dy + (((long long)n * Ho + ho) * Wo + wo) * C + c);
+        const TVec<T, VEC> wv = vload<T, VEC>(w + ((long long)kh * K + kw) * C + c);
+#pragma unroll
+        for (int i = 0; i < VEC; ++i)
+          acc[i] += DfdCvt<T>::to_f32(gv.v[i]) * DfdCvt<T>::to_f32(wv.v[i]);
+      }
+    }
+    TVec<T, VEC> ov;
+#pragma unroll
+    for (int i = 0; i < VEC; ++i) ov.v[i] = DfdCvt<T>::from_f32(acc[i]);
+    vstore<T, VEC>(dx + (((long long)n * H + hi) * W + wi) * C + c, ov);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// backward weight:
+// dw[kh,kw,c] = sum_n,ho,wo dy[n,ho,wo,c] * x[n, ho*sh-ph+kh, wo*sw-pw+kw, c]
+//
+// Block layout: blockDim.x = K * CVT threads; thread t handles
+//   kh        = t / CVT         (its own kernel row)
+//   cvec slot = t % CVT  →  c = (blockIdx.x * CVT + slot) * VEC
+// so consecutive threads touch consecutive channel vectors (coalesced).
+// Each thread loops over a chunk of (n,ho) rows × all wo, keeping K*VEC fp32
+// partial sums in registers, then one atomicAdd per partial.
+// ---------------------------------------------------------------------------
+template <typename T, int K, int VEC, int CVT>
+__global__ void dw_bwd_weight_kernel(const T* __restrict__ dy, const T* __restrict__ x,
+                                     float* __restrict__ dw, int N, int C, int H, int W,
+                                     int Ho, int Wo, int sh, int sw, int ph, int pw,
+                                     int rows_per_block) {
+  const int t = threadIdx.x;
+  const int kh = t / CVT;
+  const int slot = t % CVT;
+  const int cvec = blockIdx.x * CVT + slot;
+  const int cv = C / VEC;
+  if (cvec >= cv) return;
+  const int c = cvec * VEC;
+
+  float acc[K * VEC];
+#pragma unroll
+  for (int i = 0; i < K * VEC; ++i) acc[i] = 0.f;
+
+  const long long rows_total = (long long)N * Ho;
+  const long long row0 = (long long)blockIdx.y * rows_per_block;
+  const long long row1 = min(row0 + rows_per_block, rows_total);
+
+  for (long long r = row0; r < row1; ++r) {
+    const int ho = (int)(r % Ho);
+    const int n = (int)(r / Ho);
+    const int hi = ho * sh - ph + kh;
+    if (hi < 0 || hi >= H) continue;
+    const T* dy_row = dy + (((long long)n * Ho + ho) * Wo) * C + c;
+    const T* x_row = x + (((long long)n * H + hi) * W) * C + c;
+    for (int wo = 0; wo < Wo; ++wo) {
+      const TVec<T, VEC> gv = vload<T, VEC>(dy_row + (long long)wo * C);
+      float gf[VEC];
+#pragma unroll
+      for (int i = 0; i < VEC; ++i) gf[i] = DfdCvt<T>::to_f32(gv.v[i]);
+      const int wi0 = wo * sw - pw;
+#pragma unroll
+      for (int kw = 0; kw < K; ++kw) {
+        const int wi = wi0 + kw;
+        if (wi < 0 || wi >= W) continue;
+        const TVec<T, VEC> xv = vload<T, VEC>(x_row + (long long)wi * C);
+#pragma unroll
+        for (int i = 0; i < VEC; ++i)
+          acc[kw * VEC + i] += gf[i] * DfdCvt<T>::to_f32(xv.v[i]);
+      }
+    }
+  }
+
+#pragma unroll
+  for (int kw = 0; kw < K; ++kw)
+#pragma unroll
+    for (int i = 0; i < VEC; ++i)
+      atomicAdd(dw + ((long long)kh * K + kw) * C + c + i, acc[kw * VEC + i]);
+}
+
+// ---------------------------------------------------------------------------
+// host-side dispatch
+// ---------------------------------------------------------------------------
+
+int pick_vec(long long c, int elem_size) {
+  const int max_vec = elem_size == 4 ? 4 : 8;  // 16 B per lane load
+  for (int v = max_vec; v > 1; v >>= 1)
+    if (c % v == 0) return v;
+  return 1;
+}
+
+struct Geom {
+  int N, C, H, W, Ho, Wo, sh, sw, ph, pw;
+};
+
+template <typename T, int K, int VEC>
+void launch_fwd(const at::Tensor& x, const at::Tensor& w, at::Tensor& y, const Geom& g,
+                hipStream_t stream) {
+  const long long total = (long long)g.N * g.Ho * g.Wo * (g.C / VEC);
+  const int block = 256;
+  dw_fwd_kernel<T, K, VEC><<<dfd_grid(total, block), block, 0, stream>>>(
+      (const T*)x.data_ptr(), (const T*)w.data_ptr(), (T*)y.data_ptr(), g.N, g.C, g.H,
+      g.W, g.Ho, g.Wo, g.sh, g.sw, g.ph, g.pw);
+}
+
+template <typename T>
+void fwd_ktype(const at::Tensor& x, const at::Tensor& w, at::Tensor& y, const Geom& g,
+               int K, int vec, hipStream_t stream) {
+  switch (K) {
+#define DFD_CASE_K(KK)                                                      \
+  case KK:                                                                  \
+    switch (vec) {                                                          \
+      case 8: launch_fwd<T, KK, 8>(x, w, y, g, stream); break;              \
+      case 4: launch_fwd<T, KK, 4>(x, w, y, g, stream); break;              \
+      case 2: launch_fwd<T, KK, 2>(x, w, y, g, stream); break;              \
+      default: launch_fwd<T, KK, 1>(x, w, y, g, stream); break;             \
+    }                                                                       \
+    break;
+    DFD_CASE_K(3)
+    DFD_CASE_K(5)
+    DFD_CASE_K(7)
+    DFD_CASE_K(9)
+    DFD_CASE_K(11)
+#undef DFD_CASE_K
+    default:
+      TORCH_CHECK(false, "dwconv fwd: unsupported kernel size ", K);
+  }
+}
+
+template <typename T, int K, int VEC>
+void launch_bwd_data(const at::Tensor& dy, const at::Tensor& w, at::Tensor& dx,
+                     const Geom& g, hipStream_t stream) {
+  const long long total = (long long)g.N * g.H * g.W * (g.C / VEC);
+  const int block = 256;
+  dw_bwd_data_kernel<T, K, VEC><<<dfd_grid(total, block), block, 0, stream>>>(
+      (const T*)dy.data_ptr(), (const T*)w.data_ptr(), (T*)dx.data_ptr(), g.N, g.C, g.H,
+      g.W, g.Ho, g.Wo, g.sh, g.sw, g.ph, g.pw);
+}
+
+template <typename T>
+void bwd_data_ktype(const at::Tensor& dy, const at::Tensor& w, at::Tensor& dx,
+                    const Geom& g, int K, int vec, hipStream_t stream) {
+  switch (K) {
+#define DFD_CASE_K(KK)                                                      \
+  case KK:                                                                  \
+    switch (vec) {                                                          \
+      case 8: launch_bwd_data<T, KK, 8>(dy, w, dx, g, stream); break;       \
+      case 4: launch_bwd_data<T, KK, 4>(dy, w, dx, g, stream); break;       \
+      case 2: launch_bwd_data<T, KK, 2>(dy, w, dx, g, stream); break;       \
+      default: launch_bwd_data<T, KK, 1>(dy, w, dx, g, stream); break;      \
+    }                                                                       \
+    break;
+    DFD_CASE_K(3)
+    DFD_CASE_K(5)
+    DFD_CASE_K(7)
+    DFD_CASE_K(9)
+    DFD_CASE_K(11)
+#undef DFD_CASE_K
+    default:
+      TORCH_CHECK(false, "dwconv bwd_data: unsupported kernel size ", K);
+  }
+}
+
+constexpr int kCVT = 64;  // channel-vectors per block in bwd-weight
+
+template <typename T, int K, int VEC>
+void launch_bwd_weight(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw,
+                       const Geom& g, hipStream_t stream) {
+  const int cv = g.C / VEC;
+  const int grid_x = (cv + kCVT - 1) / kCVT;
+  const long long rows_total = (long long)g.N * g.Ho;
+  // aim for ~2048 blocks total to fill 256 CUs
+  int grid_y = (int)std::min<long long>(rows_total, std::max(1, kMaxGrid / grid_x));
+  const int rows_per_block = (int)((rows_total + grid_y - 1) / grid_y);
+  grid_y = (int)((rows_total + rows_per_block - 1) / rows_per_block);
+  dim3 grid(grid_x, grid_y);
+  dw_bwd_weight_kernel<T, K, VEC, kCVT><<<grid, K * kCVT, 0, stream>>>(
+      (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (float*)dw.data_ptr(), g.N, g.C,
+      g.H, g.W, g.Ho, g.Wo, g.sh, g.sw, g.ph, g.pw, rows_per_block);
+}
+
+template <typename T>
+void bwd_weight_ktype(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw,
+                      const Geom& g, int K, int vec, hipStream_t stream) {
+  switch (K) {
+#define DFD_CASE_K(KK)                                                      \
+  case KK:                                                                  \
+    switch (vec) {                                                          \
+      case 8: launch_bwd_weight<T, KK, 8>(dy, x, dw, g, stream); break;     \
+      case 4: launch_bwd_weight<T, KK, 4>(dy, x, dw, g, stream); break;     \
+      case 2: launch_bwd_weight<T, KK, 2>(dy, x, dw, g, stream); break;     \
+      default: launch_bwd_weight<T, KK, 1>(dy, x, dw, g, stream); break;    \
+    }                                                                       \
+    break;
+    DFD_CASE_K(3)
+    DFD_CASE_K(5)
+    DFD_CASE_K(7)
+    DFD_CASE_K(9)
+    DFD_CASE_K(11)
+#undef DFD_CASE_K
+    default:
+      TORCH_CHECK(false, "dwconv bwd_weight: unsupported kernel size ", K);
+  }
+}
+
+Geom make_geom(int64_t N, int64_t C, int64_t H, int64_t W, int64_t K, int64_t sh,
+               int64_t sw, int64_t ph, int64_t pw) {
+  Geom g;
+  g.N = (int)N;
+  g.C = (int)C;
+  g.H = (int)H;
+  g.W = (int)W;
+  g.sh = (int)sh;
+  g.sw = (int)sw;
+  g.ph = (int)ph;
+  g.pw = (int)pw;
+  g.Ho = (int)((H + 2 * ph - K) / sh + 1);
+  g.Wo = (int)((W + 2 * pw - K) / sw + 1);
+  return g;
+}
+
+}  // namespace
+
+// x: NCHW logical, channels_last physical. w_packed: (K, K, C) same dtype.
+at::Tensor dw_conv2d_fwd(at::Tensor x, at::Tensor w_packed, int64_t sh, int64_t sw,
+                         int64_t ph, int64_t pw) {
+  TORCH_CHECK(x.is_cuda() && w_packed.is_cuda(), "dwconv: CUDA tensors required");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "dwconv: x must be channels_last");
+  TORCH_CHECK(w_packed.is_contiguous(), "dwconv: packed weight must be contiguous");
+  const auto K = w_packed.size(0);
+  TORCH_CHECK(w_packed.size(1) == K && w_packed.size(2) == x.size(1),
+              "dwconv: packed weight shape mismatch");
+  auto g = make_geom(x.size(0), x.size(1), x.size(2), x.size(3), K, sh, sw, ph, pw);
+  auto y = at::empty({g.N, g.C, g.Ho, g.Wo},
+                     x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  const int vec = pick_vec(g.C, (int)x.element_size());
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  const auto stype = x.scalar_type();
+  const int Ki = (int)K;
+  switch (stype) {
+    case at::kBFloat16: fwd_ktype<__hip_bfloat16>(x, w_packed, y, g, Ki, vec, stream); break;
+    case at::kHalf: fwd_ktype<__half>(x, w_packed, y, g, Ki, vec, stream); break;
+    case at::kFloat: fwd_ktype<float>(x, w_packed, y, g, Ki, vec, stream); break;
+    default: TORCH_CHECK(false, "dwconv: unsupported dtype");
+  }
+  return y;
+}
+
+at::Tensor dw_conv2d_bwd_data(at::Tensor dy, at::Tensor w_packed, int64_t H, int64_t W,
+                              int64_t sh, int64_t sw, int64_t ph, int64_t pw) {
+  TORCH_CHECK(dy.is_cuda() && w_packed.is_cuda(), "dwconv: CUDA tensors required");
+  TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast), "dwconv: dy must be channels_last");
+  const auto K = w_packed.size(0);
+  Geom g = make_geom(dy.size(0), dy.size(1), H, W, K, sh, sw, ph, pw);
+  TORCH_CHECK(g.Ho == dy.size(2) && g.Wo == dy.size(3), "dwconv bwd_data: geometry mismatch");
+  auto dx = at::empty({g.N, g.C, g.H, g.W},
+                      dy.options().memory_format(at::MemoryFormat::ChannelsLast));
+  const int vec = pick_vec(g.C, (int)dy.element_size());
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  const auto stype = dy.scalar_type();
+  const int Ki = (int)K;
+  switch (stype) {
+    case at::kBFloat16: bwd_data_ktype<__hip_bfloat16>(dy, w_packed, dx, g, Ki, vec, stream); break;
+    case at::kHalf: bwd_data_ktype<__half>(dy, w_packed, dx, g, Ki, vec, stream); break;
+    case at::kFloat: bwd_data_ktype<float>(dy, w_packed, dx, g, Ki, vec, stream); break;
+    default: TORCH_CHECK(false, "dwconv: unsupported dtype");
+  }
+  return dx;
+}
+
+// returns (K, K, C) fp32
+at::Tensor dw_conv2d_bwd_weight(at::Tensor dy, at::Tensor x, int64_t K, int64_t sh,
+                                int64_t sw, int64_t ph, int64_t pw) {
+  TORCH_CHECK(dy.is_cuda() && x.is_cuda(), "dwconv: CUDA tensors required");
+  TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "dwconv: x must be channels_last");
+  TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast), "dwconv: dy must be channels_last");
+  Geom g = make_geom(x.size(0), x.size(1), x.size(2), x.size(3), K, sh, sw, ph, pw);
+  TORCH_CHECK(g.Ho == dy.size(2) && g.Wo == dy.size(3), "dwconv bwd_weight: geometry mismatch");
+  auto dw = at::zeros({K, K, (long long)g.C}, x.options().dtype(at::kFloat));
+  const int vec = pick_vec(g.C, (int)x.element_size());
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  const auto stype = x.scalar_type();
+  const int Ki = (int)K;
+  switch (stype) {
+    case at::kBFloat16: bwd_weight_ktype<__hip_bfloat16>(dy, x, dw, g, Ki, vec, stream); break;
+    case at::kHalf: bwd_weight_ktype<__half>(dy, x, dw, g, Ki, vec, stream); break;
+    case at::kFloat: bwd_weight_ktype<float>(dy, x, dw, g, Ki, vec, stream); break;
+    default: TORCH_CHECK(false, "dwconv: unsupported dtype");
+  }
+  return dw;
+}

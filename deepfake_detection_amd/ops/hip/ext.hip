@@ -31,6 +31,12 @@ void adamw_multi_tensor(std::vector<at::Tensor> params, std::vector<at::Tensor> 
                         double weight_decay);
 void ema_multi_tensor(std::vector<at::Tensor> ema_params, std::vector<at::Tensor> model_params,
                       double decay);
+at::Tensor dw_conv2d_fwd(at::Tensor x, at::Tensor w_packed, int64_t sh, int64_t sw,
+                         int64_t ph, int64_t pw);
+at::Tensor dw_conv2d_bwd_data(at::Tensor dy, at::Tensor w_packed, int64_t H, int64_t W,
+                              int64_t sh, int64_t sw, int64_t ph, int64_t pw);
+at::Tensor dw_conv2d_bwd_weight(at::Tensor dy, at::Tensor x, int64_t K, int64_t sh,
+                                int64_t sw, int64_t ph, int64_t pw);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "deepfake_detection_amd gfx950 (MI355X/CDNA4) kernels";
@@ -45,4 +51,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsprop_tf_multi_tensor", &rmsprop_tf_multi_tensor, "fused RMSpropTF step");
   m.def("adamw_multi_tensor", &adamw_multi_tensor, "fused AdamW step");
   m.def("ema_multi_tensor", &ema_multi_tensor, "fused EMA update");
+  m.def("dw_conv2d_fwd", &dw_conv2d_fwd, "depthwise conv2d forward (NHWC)");
+  m.def("dw_conv2d_bwd_data", &dw_conv2d_bwd_data, "depthwise conv2d bwd data (NHWC)");
+  m.def("dw_conv2d_bwd_weight", &dw_conv2d_bwd_weight, "depthwise conv2d bwd weight (NHWC)");
 }
