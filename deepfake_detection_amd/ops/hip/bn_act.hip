@@ -268,34 +268,43 @@ __global__ void bn_act_bwd_reduce_kernel(
 //   train: dx = gamma*invstd * (g - dbeta/M - xhat*dgamma/M)
 //   eval:  dx = gamma*invstd * g
 // ---------------------------------------------------------------------------
-template <typename T, Act ACT, bool TRAIN>
+template <typename T, int N>
+struct alignas(sizeof(T) * N) BVec {
+  T v[N];
+};
+
+template <typename T, Act ACT, bool TRAIN, int VEC>
 __global__ void bn_act_bwd_dx_kernel(
     const T* __restrict__ dy, const T* __restrict__ x, T* __restrict__ dx,
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ weight, const float* __restrict__ bias,
     const float* __restrict__ dgamma, const float* __restrict__ dbeta,
-    long long total, int C, float invM) {
-  const long long idx0 = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
-  const long long stride = (long long)gridDim.x * blockDim.x * 4;
-  for (long long i = idx0; i < total; i += stride) {
+    long long M, int C, float invM) {
+  const int cv = C / VEC;
+  const long long total = M * cv;
+  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long long)gridDim.x * blockDim.x) {
+    const int c = (int)(idx % cv) * VEC;
+    const long long r = idx / cv;
+    const BVec<T, VEC> xv = *reinterpret_cast<const BVec<T, VEC>*>(x + r * C + c);
+    const BVec<T, VEC> dv = *reinterpret_cast<const BVec<T, VEC>*>(dy + r * C + c);
+    BVec<T, VEC> ov;
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      const long long k = i + j;
-      if (k >= total) break;
-      const int c = (int)(k % C);
-      const float xf = DfdCvt<T>::to_f32(x[k]);
-      const float xh = (xf - mean[c]) * invstd[c];
-      const float ga = weight ? weight[c] : 1.f;
-      const float z = fmaf(ga, xh, bias ? bias[c] : 0.f);
-      const float g = DfdCvt<T>::to_f32(dy[k]) * act_bwd(z, ACT);
+    for (int j = 0; j < VEC; ++j) {
+      const float is = invstd[c + j];
+      const float xh = (DfdCvt<T>::to_f32(xv.v[j]) - mean[c + j]) * is;
+      const float ga = weight ? weight[c + j] : 1.f;
+      const float z = fmaf(ga, xh, bias ? bias[c + j] : 0.f);
+      const float g = DfdCvt<T>::to_f32(dv.v[j]) * act_bwd(z, ACT);
       float v;
       if (TRAIN) {
-        v = ga * invstd[c] * (g - dbeta[c] * invM - xh * dgamma[c] * invM);
+        v = ga * is * (g - dbeta[c + j] * invM - xh * dgamma[c + j] * invM);
       } else {
-        v = ga * invstd[c] * g;
+        v = ga * is * g;
       }
-      dx[k] = DfdCvt<T>::from_f32(v);
+      ov.v[j] = DfdCvt<T>::from_f32(v);
     }
+    *reinterpret_cast<BVec<T, VEC>*>(dx + r * C + c) = ov;
   }
 }
 
@@ -435,25 +444,36 @@ std::vector<at::Tensor> bn_act_bwd(
   });
 
   const int block = 256;
-  const int grid_e = dfd_grid(total / 4 + 1, block);
+  int dxvec = (x.element_size() == 4) ? 4 : 8;
+  while (dxvec > 1 && (C % dxvec)) dxvec >>= 1;
+  const int grid_e = dfd_grid(M * (C / dxvec), block);
+#define DFD_BN_DX(TRAIN, V)                                                       \
+  hipLaunchKernelGGL((bn_act_bwd_dx_kernel<T, ACT, TRAIN, V>), dim3(grid_e),      \
+                     dim3(block), 0, stream, (const T*)dy.data_ptr(),             \
+                     (const T*)x.data_ptr(), (T*)dx.data_ptr(),                   \
+                     save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),  \
+                     w_p, b_p, dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), \
+                     M, C, invM)
   DISPATCH_DTYPE(x.scalar_type(), "bn_bwd_dx", [&] {
     DISPATCH_ACT(act, [&] {
+      const float invM = 1.f / (float)M;
       if (training) {
-        hipLaunchKernelGGL((bn_act_bwd_dx_kernel<T, ACT, true>), dim3(grid_e), dim3(block), 0,
-                           stream, (const T*)dy.data_ptr(), (const T*)x.data_ptr(),
-                           (T*)dx.data_ptr(), save_mean.data_ptr<float>(),
-                           save_invstd.data_ptr<float>(), w_p, b_p,
-                           dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
-                           total, C, 1.f / (float)M);
+        switch (dxvec) {
+          case 8: DFD_BN_DX(true, 8); break;
+          case 4: DFD_BN_DX(true, 4); break;
+          case 2: DFD_BN_DX(true, 2); break;
+          default: DFD_BN_DX(true, 1); break;
+        }
       } else {
-        hipLaunchKernelGGL((bn_act_bwd_dx_kernel<T, ACT, false>), dim3(grid_e), dim3(block), 0,
-                           stream, (const T*)dy.data_ptr(), (const T*)x.data_ptr(),
-                           (T*)dx.data_ptr(), save_mean.data_ptr<float>(),
-                           save_invstd.data_ptr<float>(), w_p, b_p,
-                           dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
-                           total, C, 1.f / (float)M);
+        switch (dxvec) {
+          case 8: DFD_BN_DX(false, 8); break;
+          case 4: DFD_BN_DX(false, 4); break;
+          case 2: DFD_BN_DX(false, 2); break;
+          default: DFD_BN_DX(false, 1); break;
+        }
       }
     });
   });
+#undef DFD_BN_DX
   return {dx, dgamma, dbeta};
 }

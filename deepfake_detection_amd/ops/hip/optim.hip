@@ -16,7 +16,10 @@
 
 namespace {
 
-constexpr long long kChunk = 1 << 20;  // elements per chunk
+// Elements per chunk: small enough that a 62M-param model yields thousands
+// of blocks (the first cut used 1M-element chunks -> ~220 blocks -> latency
+// bound at 2.7 ms/step; roofline for the update is ~0.3 ms).
+constexpr long long kChunk = 1 << 15;
 
 // chunk table layout (int64 per chunk): [p, g, m1, m2, numel]
 struct ChunkTable {
@@ -52,9 +55,32 @@ ChunkTable build_chunks(const std::vector<at::Tensor>& a,
   return t;
 }
 
+DFD_DEV void rmsprop_tf_update(float& pv, float gv, float& s, float& b, float lr,
+                               float oma, float eps, float momentum,
+                               float weight_decay, bool decoupled, bool lr_in_mom) {
+  if (weight_decay != 0.f) {
+    if (decoupled) pv -= weight_decay * pv;
+    else gv += weight_decay * pv;
+  }
+  s += oma * (gv * gv - s);          // TF order of ops
+  const float avg = sqrtf(s + eps);  // eps inside sqrt
+  if (momentum > 0.f) {
+    if (lr_in_mom) {
+      b = b * momentum + lr * gv / avg;  // LR inside the buffer
+      pv -= b;
+    } else {
+      b = b * momentum + gv / avg;
+      pv -= lr * b;
+    }
+  } else {
+    pv -= lr * gv / avg;
+  }
+}
+
 __global__ void rmsprop_tf_kernel(const long long* __restrict__ table, int nchunks,
                                   float lr, float alpha, float eps, float momentum,
                                   float weight_decay, bool decoupled, bool lr_in_mom) {
+  const float oma = 1.f - alpha;
   for (int ch = blockIdx.x; ch < nchunks; ch += gridDim.x) {
     const long long* row = table + (long long)ch * 5;
     float* p = (float*)row[0];
@@ -62,32 +88,27 @@ __global__ void rmsprop_tf_kernel(const long long* __restrict__ table, int nchun
     float* sa = (float*)row[2];
     float* buf = (float*)row[3];
     const long long nelem = row[4];
-    const float oma = 1.f - alpha;
-    for (long long i = threadIdx.x; i < nelem; i += blockDim.x) {
-      float pv = p[i];
-      float gv = g[i];
-      if (weight_decay != 0.f) {
-        if (decoupled) pv -= weight_decay * pv;
-        else gv += weight_decay * pv;
-      }
-      float s = sa[i];
-      s += oma * (gv * gv - s);        // TF order of ops
-      sa[i] = s;
-      const float avg = sqrtf(s + eps);  // eps inside sqrt
-      if (momentum > 0.f) {
-        float b = buf[i];
-        if (lr_in_mom) {
-          b = b * momentum + lr * gv / avg;  // LR inside the buffer
-          pv -= b;
-        } else {
-          b = b * momentum + gv / avg;
-          pv -= lr * b;
-        }
-        buf[i] = b;
-      } else {
-        pv -= lr * gv / avg;
-      }
+    const long long nv = nelem / 4;
+    const bool has_mom = momentum > 0.f;
+    for (long long i = threadIdx.x; i < nv; i += blockDim.x) {
+      float4 pv = ((float4*)p)[i];
+      const float4 gv = ((const float4*)g)[i];
+      float4 sv = ((float4*)sa)[i];
+      float4 bv = has_mom ? ((float4*)buf)[i] : float4{0, 0, 0, 0};
+      rmsprop_tf_update(pv.x, gv.x, sv.x, bv.x, lr, oma, eps, momentum, weight_decay, decoupled, lr_in_mom);
+      rmsprop_tf_update(pv.y, gv.y, sv.y, bv.y, lr, oma, eps, momentum, weight_decay, decoupled, lr_in_mom);
+      rmsprop_tf_update(pv.z, gv.z, sv.z, bv.z, lr, oma, eps, momentum, weight_decay, decoupled, lr_in_mom);
+      rmsprop_tf_update(pv.w, gv.w, sv.w, bv.w, lr, oma, eps, momentum, weight_decay, decoupled, lr_in_mom);
+      ((float4*)p)[i] = pv;
+      ((float4*)sa)[i] = sv;
+      if (has_mom) ((float4*)buf)[i] = bv;
+    }
+    for (long long i = nv * 4 + threadIdx.x; i < nelem; i += blockDim.x) {
+      float pv = p[i], s = sa[i], b = has_mom ? buf[i] : 0.f;
+      rmsprop_tf_update(pv, g[i], s, b, lr, oma, eps, momentum, weight_decay, decoupled, lr_in_mom);
       p[i] = pv;
+      sa[i] = s;
+      if (has_mom) buf[i] = b;
     }
   }
 }
@@ -103,7 +124,28 @@ __global__ void adamw_kernel(const long long* __restrict__ table, int nchunks,
     float* m2 = (float*)row[3];
     const long long nelem = row[4];
     const float step_size = lr / bias_c1;
-    for (long long i = threadIdx.x; i < nelem; i += blockDim.x) {
+    const long long nv = nelem / 4;
+    for (long long i = threadIdx.x; i < nv; i += blockDim.x) {
+      float4 pv = ((float4*)p)[i];
+      const float4 gv = ((const float4*)g)[i];
+      float4 av = ((float4*)m1)[i];
+      float4 vv = ((float4*)m2)[i];
+      float* pp = &pv.x;
+      const float* gp = &gv.x;
+      float* ap = &av.x;
+      float* vp = &vv.x;
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const float pd = pp[j] * (1.f - lr * weight_decay);
+        const float a = ap[j] = beta1 * ap[j] + (1.f - beta1) * gp[j];
+        const float v = vp[j] = beta2 * vp[j] + (1.f - beta2) * gp[j] * gp[j];
+        pp[j] = pd - step_size * a / (sqrtf(v) * inv_sqrt_bias_c2 + eps);
+      }
+      ((float4*)p)[i] = pv;
+      ((float4*)m1)[i] = av;
+      ((float4*)m2)[i] = vv;
+    }
+    for (long long i = nv * 4 + threadIdx.x; i < nelem; i += blockDim.x) {
       float pv = p[i] * (1.f - lr * weight_decay);
       const float gv = g[i];
       const float a = m1[i] = beta1 * m1[i] + (1.f - beta1) * gv;
