@@ -6,16 +6,18 @@
 // bf16/fp16/fp32; all statistics and parameters fp32.
 //
 // Layout: channels_last (N,C,H,W) == row-major [M, C] with M = N*H*W and C
-// contiguous. Per-channel reductions: each wave owns a 64*VEC-channel slab,
-// lanes read ushort4 (8 B) vectors; 4 waves per block cover different rows;
-// per-block partials combine in LDS and one atomicAdd per channel publishes
-// to the fp32 accumulator (Guideline 12).
+// contiguous. Every kernel uses the same block shape: a power-of-two number
+// of channel-slots (cpb ≤ 64) × row-groups; each thread owns a VEC-wide
+// contiguous channel slice, loads its per-channel fp32 coefficients into
+// REGISTERS once, and streams rows. This keeps the hot loop at pure
+// vectorized stream traffic — the first cut reloaded 6 fp32 coefficients
+// per element from L1 and ran 8× off roofline
+// (bn_act_bwd_dx 774 us vs se_bwd_reduce 90 us on the same bytes).
 //
-// Training forward is two passes (stats reduce -> finalize -> fused
-// normalize+act elementwise); backward is a reduce pass (dgamma/dbeta with
-// act' recompute) + an elementwise dx pass. SiLU backward recomputes
-// sigma(z) from the saved mean/invstd (the "hard part" flagged in
-// SURVEY.md §7).
+// Training forward is stats-reduce -> finalize -> fused normalize+act;
+// backward is reduce (dgamma/dbeta with act' recompute) + elementwise dx.
+// SiLU backward recomputes sigma(z) from saved mean/invstd (the "hard part"
+// flagged in SURVEY.md §7).
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
@@ -24,69 +26,110 @@
 
 namespace {
 
+template <typename T, int N>
+struct alignas(sizeof(T) * N) BVec {
+  T v[N];
+};
+
+template <typename T, int N>
+DFD_DEV BVec<T, N> bvload(const T* p) {
+  return *reinterpret_cast<const BVec<T, N>*>(p);
+}
+
+template <typename T, int N>
+DFD_DEV void bvstore(T* p, const BVec<T, N>& x) {
+  *reinterpret_cast<BVec<T, N>*>(p) = x;
+}
+
+// ---- block/grid plan shared by all kernels --------------------------------
+int bn_pick_vec(long long c, int elem_size) {
+  const int max_vec = elem_size == 4 ? 4 : 8;
+  for (int v = max_vec; v > 1; v >>= 1)
+    if (c % v == 0) return v;
+  return 1;
+}
+
+int bn_pick_log2_cpb(int cv) {
+  int l = 0;
+  while ((1 << l) < cv && l < 6) ++l;
+  return l;
+}
+
+struct BnPlan {
+  int log2_cpb, ctiles, chunks, rows_per_chunk;
+};
+
+BnPlan bn_plan(int cv, long long rows) {
+  BnPlan p;
+  p.log2_cpb = bn_pick_log2_cpb(cv);
+  const int cpb = 1 << p.log2_cpb;
+  p.ctiles = (cv + cpb - 1) / cpb;
+  long long want = (2048 + p.ctiles - 1) / p.ctiles;
+  const int nrg = 256 >> p.log2_cpb;
+  long long max_chunks = (rows + nrg - 1) / nrg;
+  if (want > max_chunks) want = max_chunks;
+  if (want < 1) want = 1;
+  p.chunks = (int)want;
+  p.rows_per_chunk = (int)((rows + p.chunks - 1) / p.chunks);
+  return p;
+}
+
 // ---------------------------------------------------------------------------
-// stats reduce: sum and sumsq per channel
+// stats reduce: sum and sumsq per channel (grid: ctiles × row-chunks)
 // ---------------------------------------------------------------------------
 template <typename T, int VEC>
 __global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum,
-                                float* __restrict__ sumsq, long long M, int C) {
-  const int lane = threadIdx.x & (kWave - 1);
-  const int wid = threadIdx.x / kWave;
-  const int nw = blockDim.x / kWave;
-
-  const int cb = blockIdx.x;              // channel slab: 64*VEC channels
-  const int c0 = cb * kWave * VEC + lane * VEC;
-  if (c0 >= C) return;
+                                float* __restrict__ sumsq, long long M, int C,
+                                int log2_cpb, int rows_per_chunk) {
+  extern __shared__ float lds[];  // [256 * VEC]
+  const int cpb = 1 << log2_cpb;
+  const int slot = threadIdx.x & (cpb - 1);
+  const int rg = threadIdx.x >> log2_cpb;
+  const int nrg = blockDim.x >> log2_cpb;
+  const int cv = C / VEC;
+  const int cvec = blockIdx.x * cpb + slot;
+  const bool active = cvec < cv;
+  const int c = cvec * VEC;
 
   float s[VEC], q[VEC];
 #pragma unroll
   for (int j = 0; j < VEC; ++j) { s[j] = 0.f; q[j] = 0.f; }
 
-  const bool full = (c0 + VEC) <= C;
-  const long long row0 = (long long)blockIdx.y * nw + wid;
-  const long long rstride = (long long)gridDim.y * nw;
-
-  for (long long r = row0; r < M; r += rstride) {
-    const T* px = x + r * C + c0;
-    if (full && VEC == 4 && (sizeof(T) == 2)) {
-      ushort4 v = *reinterpret_cast<const ushort4*>(px);
-      const T* e = reinterpret_cast<const T*>(&v);
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        float f = DfdCvt<T>::to_f32(e[j]);
-        s[j] += f; q[j] += f * f;
-      }
-    } else {
+  if (active) {
+    const long long r0 = (long long)blockIdx.y * rows_per_chunk;
+    const long long r1 = min(r0 + rows_per_chunk, M);
+    const T* xc = x + c;
+    for (long long r = r0 + rg; r < r1; r += nrg) {
+      const BVec<T, VEC> xv = bvload<T, VEC>(xc + r * C);
 #pragma unroll
       for (int j = 0; j < VEC; ++j) {
-        if (c0 + j < C) {
-          float f = DfdCvt<T>::to_f32(px[j]);
-          s[j] += f; q[j] += f * f;
-        }
+        const float f = DfdCvt<T>::to_f32(xv.v[j]);
+        s[j] += f;
+        q[j] += f * f;
       }
     }
   }
 
-  // combine the block's waves per channel through LDS
-  __shared__ float lds[4][kWave];  // one VEC element at a time
+  float* my = lds + (size_t)threadIdx.x * VEC;
+  // reduce s then q through the same LDS buffer
 #pragma unroll
-  for (int j = 0; j < VEC; ++j) {
+  for (int pass = 0; pass < 2; ++pass) {
     __syncthreads();
-    lds[wid][lane] = s[j];
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) my[j] = pass == 0 ? s[j] : q[j];
     __syncthreads();
-    if (wid == 0) {
-      float acc = 0.f;
-      for (int w = 0; w < nw; ++w) acc += lds[w][lane];
-      if (c0 + j < C && acc != 0.f) atomicAdd(&sum[c0 + j], acc);
-      else if (c0 + j < C) atomicAdd(&sum[c0 + j], acc);
+    for (int step = nrg >> 1; step > 0; step >>= 1) {
+      if (rg < step) {
+        const float* other = lds + ((size_t)((rg + step) << log2_cpb) + slot) * VEC;
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) my[j] += other[j];
+      }
+      __syncthreads();
     }
-    __syncthreads();
-    lds[wid][lane] = q[j];
-    __syncthreads();
-    if (wid == 0) {
-      float acc = 0.f;
-      for (int w = 0; w < nw; ++w) acc += lds[w][lane];
-      if (c0 + j < C) atomicAdd(&sumsq[c0 + j], acc);
+    if (rg == 0 && active) {
+      float* out = pass == 0 ? sum : sumsq;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) atomicAdd(out + c + j, my[j]);
     }
   }
 }
@@ -139,43 +182,40 @@ __global__ void bn_finalize_eval_kernel(
 
 // ---------------------------------------------------------------------------
 // fused normalize + act elementwise: y = act(scale*x + shift)
+// scale/shift live in registers; rows stream.
 // ---------------------------------------------------------------------------
-template <typename T, Act ACT>
+template <typename T, Act ACT, int VEC>
 __global__ void bn_act_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
                                   const float* __restrict__ scale,
-                                  const float* __restrict__ shift,
-                                  long long total, int C) {
-  // vectorized: 4 elements (8 B for 16-bit dtypes) per thread
-  const long long idx0 = ((long long)blockIdx.x * blockDim.x + threadIdx.x) * 4;
-  const long long stride = (long long)gridDim.x * blockDim.x * 4;
-  for (long long i = idx0; i < total; i += stride) {
-    if (i + 4 <= total && (C % 4 == 0)) {
-      const int c = (int)(i % C);
-      if (sizeof(T) == 2) {
-        ushort4 v = *reinterpret_cast<const ushort4*>(x + i);
-        T* e = reinterpret_cast<T*>(&v);
-        ushort4 o;
-        T* oe = reinterpret_cast<T*>(&o);
+                                  const float* __restrict__ shift, long long M, int C,
+                                  int log2_cpb, int rows_per_chunk) {
+  const int cpb = 1 << log2_cpb;
+  const int slot = threadIdx.x & (cpb - 1);
+  const int rg = threadIdx.x >> log2_cpb;
+  const int nrg = blockDim.x >> log2_cpb;
+  const int cv = C / VEC;
+  const int cvec = blockIdx.x * cpb + slot;
+  if (cvec >= cv) return;
+  const int c = cvec * VEC;
+
+  float sc[VEC], sh[VEC];
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          const float z = fmaf(DfdCvt<T>::to_f32(e[j]), scale[c + j], shift[c + j]);
-          oe[j] = DfdCvt<T>::from_f32(act_fwd(z, ACT));
-        }
-        *reinterpret_cast<ushort4*>(y + i) = o;
-      } else {
+  for (int j = 0; j < VEC; ++j) {
+    sc[j] = scale[c + j];
+    sh[j] = shift[c + j];
+  }
+
+  const long long r0 = (long long)blockIdx.y * rows_per_chunk;
+  const long long r1 = min(r0 + rows_per_chunk, M);
+  for (long long r = r0 + rg; r < r1; r += nrg) {
+    const BVec<T, VEC> xv = bvload<T, VEC>(x + r * C + c);
+    BVec<T, VEC> yv;
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          const float z = fmaf(DfdCvt<T>::to_f32(x[i + j]), scale[c + j], shift[c + j]);
-          y[i + j] = DfdCvt<T>::from_f32(act_fwd(z, ACT));
-        }
-      }
-    } else {
-      for (int j = 0; j < 4 && i + j < total; ++j) {
-        const int c = (int)((i + j) % C);
-        const float z = fmaf(DfdCvt<T>::to_f32(x[i + j]), scale[c], shift[c]);
-        y[i + j] = DfdCvt<T>::from_f32(act_fwd(z, ACT));
-      }
+    for (int j = 0; j < VEC; ++j) {
+      const float z = fmaf(DfdCvt<T>::to_f32(xv.v[j]), sc[j], sh[j]);
+      yv.v[j] = DfdCvt<T>::from_f32(act_fwd(z, ACT));
     }
+    bvstore<T, VEC>(y + r * C + c, yv);
   }
 }
 
@@ -187,124 +227,122 @@ __global__ void bn_act_bwd_reduce_kernel(
     const T* __restrict__ dy, const T* __restrict__ x,
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ weight, const float* __restrict__ bias,
-    float* __restrict__ dgamma, float* __restrict__ dbeta,
-    long long M, int C) {
-  const int lane = threadIdx.x & (kWave - 1);
-  const int wid = threadIdx.x / kWave;
-  const int nw = blockDim.x / kWave;
+    float* __restrict__ dgamma, float* __restrict__ dbeta, long long M, int C,
+    int log2_cpb, int rows_per_chunk) {
+  extern __shared__ float lds[];
+  const int cpb = 1 << log2_cpb;
+  const int slot = threadIdx.x & (cpb - 1);
+  const int rg = threadIdx.x >> log2_cpb;
+  const int nrg = blockDim.x >> log2_cpb;
+  const int cv = C / VEC;
+  const int cvec = blockIdx.x * cpb + slot;
+  const bool active = cvec < cv;
+  const int c = cvec * VEC;
 
-  const int c0 = blockIdx.x * kWave * VEC + lane * VEC;
-  if (c0 >= C) return;
-
-  float sg[VEC], sgx[VEC];
-  float mn[VEC], is[VEC], ga[VEC], be[VEC];
+  float sg[VEC], sgx[VEC], mn[VEC], is[VEC], ga[VEC], be[VEC];
 #pragma unroll
   for (int j = 0; j < VEC; ++j) {
-    sg[j] = 0.f; sgx[j] = 0.f;
-    const int c = min(c0 + j, C - 1);
-    mn[j] = mean[c]; is[j] = invstd[c];
-    ga[j] = weight ? weight[c] : 1.f;
-    be[j] = bias ? bias[c] : 0.f;
+    sg[j] = 0.f;
+    sgx[j] = 0.f;
+    const int cc = active ? c + j : 0;
+    mn[j] = mean[cc];
+    is[j] = invstd[cc];
+    ga[j] = weight ? weight[cc] : 1.f;
+    be[j] = bias ? bias[cc] : 0.f;
   }
 
-  const bool full = (c0 + VEC) <= C;
-  const long long row0 = (long long)blockIdx.y * nw + wid;
-  const long long rstride = (long long)gridDim.y * nw;
-
-  for (long long r = row0; r < M; r += rstride) {
-    const T* px = x + r * C + c0;
-    const T* pd = dy + r * C + c0;
-    if (full && VEC == 4 && sizeof(T) == 2) {
-      ushort4 vx = *reinterpret_cast<const ushort4*>(px);
-      ushort4 vd = *reinterpret_cast<const ushort4*>(pd);
-      const T* ex = reinterpret_cast<const T*>(&vx);
-      const T* ed = reinterpret_cast<const T*>(&vd);
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        const float xf = DfdCvt<T>::to_f32(ex[j]);
-        const float xh = (xf - mn[j]) * is[j];
-        const float z = fmaf(ga[j], xh, be[j]);
-        const float g = DfdCvt<T>::to_f32(ed[j]) * act_bwd(z, ACT);
-        sg[j] += g; sgx[j] += g * xh;
-      }
-    } else {
+  if (active) {
+    const long long r0 = (long long)blockIdx.y * rows_per_chunk;
+    const long long r1 = min(r0 + rows_per_chunk, M);
+    for (long long r = r0 + rg; r < r1; r += nrg) {
+      const BVec<T, VEC> xv = bvload<T, VEC>(x + r * C + c);
+      const BVec<T, VEC> dv = bvload<T, VEC>(dy + r * C + c);
 #pragma unroll
       for (int j = 0; j < VEC; ++j) {
-        if (c0 + j < C) {
-          const float xf = DfdCvt<T>::to_f32(px[j]);
-          const float xh = (xf - mn[j]) * is[j];
-          const float z = fmaf(ga[j], xh, be[j]);
-          const float g = DfdCvt<T>::to_f32(pd[j]) * act_bwd(z, ACT);
-          sg[j] += g; sgx[j] += g * xh;
-        }
+        const float xh = (DfdCvt<T>::to_f32(xv.v[j]) - mn[j]) * is[j];
+        const float z = fmaf(ga[j], xh, be[j]);
+        const float g = DfdCvt<T>::to_f32(dv.v[j]) * act_bwd(z, ACT);
+        sg[j] += g;
+        sgx[j] += g * xh;
       }
     }
   }
 
-  __shared__ float lds[4][kWave];
+  float* my = lds + (size_t)threadIdx.x * VEC;
 #pragma unroll
-  for (int j = 0; j < VEC; ++j) {
+  for (int pass = 0; pass < 2; ++pass) {
     __syncthreads();
-    lds[wid][lane] = sg[j];
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) my[j] = pass == 0 ? sg[j] : sgx[j];
     __syncthreads();
-    if (wid == 0) {
-      float acc = 0.f;
-      for (int w = 0; w < nw; ++w) acc += lds[w][lane];
-      if (c0 + j < C) atomicAdd(&dbeta[c0 + j], acc);
+    for (int step = nrg >> 1; step > 0; step >>= 1) {
+      if (rg < step) {
+        const float* other = lds + ((size_t)((rg + step) << log2_cpb) + slot) * VEC;
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) my[j] += other[j];
+      }
+      __syncthreads();
     }
-    __syncthreads();
-    lds[wid][lane] = sgx[j];
-    __syncthreads();
-    if (wid == 0) {
-      float acc = 0.f;
-      for (int w = 0; w < nw; ++w) acc += lds[w][lane];
-      if (c0 + j < C) atomicAdd(&dgamma[c0 + j], acc);
+    if (rg == 0 && active) {
+      float* out = pass == 0 ? dbeta : dgamma;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) atomicAdd(out + c + j, my[j]);
     }
   }
 }
 
 // ---------------------------------------------------------------------------
 // backward dx elementwise:
-//   train: dx = gamma*invstd * (g - dbeta/M - xhat*dgamma/M)
-//   eval:  dx = gamma*invstd * g
+//   train: dx = k1*g - k2 - k3*(x - mean),  g = dy*act'(scale*x + shift)
+//          k1 = ga*is, k2 = ga*is*dbeta/M, k3 = ga*is^2*dgamma/M
+//   eval:  dx = k1*g
 // ---------------------------------------------------------------------------
-template <typename T, int N>
-struct alignas(sizeof(T) * N) BVec {
-  T v[N];
-};
-
 template <typename T, Act ACT, bool TRAIN, int VEC>
 __global__ void bn_act_bwd_dx_kernel(
     const T* __restrict__ dy, const T* __restrict__ x, T* __restrict__ dx,
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ weight, const float* __restrict__ bias,
     const float* __restrict__ dgamma, const float* __restrict__ dbeta,
-    long long M, int C, float invM) {
+    long long M, int C, float invM, int log2_cpb, int rows_per_chunk) {
+  const int cpb = 1 << log2_cpb;
+  const int slot = threadIdx.x & (cpb - 1);
+  const int rg = threadIdx.x >> log2_cpb;
+  const int nrg = blockDim.x >> log2_cpb;
   const int cv = C / VEC;
-  const long long total = M * cv;
-  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
-       idx += (long long)gridDim.x * blockDim.x) {
-    const int c = (int)(idx % cv) * VEC;
-    const long long r = idx / cv;
-    const BVec<T, VEC> xv = *reinterpret_cast<const BVec<T, VEC>*>(x + r * C + c);
-    const BVec<T, VEC> dv = *reinterpret_cast<const BVec<T, VEC>*>(dy + r * C + c);
+  const int cvec = blockIdx.x * cpb + slot;
+  if (cvec >= cv) return;
+  const int c = cvec * VEC;
+
+  float sc[VEC], sh[VEC], mn[VEC], k1[VEC], k2[VEC], k3[VEC];
+#pragma unroll
+  for (int j = 0; j < VEC; ++j) {
+    const int cc = c + j;
+    const float is = invstd[cc];
+    const float ga = weight ? weight[cc] : 1.f;
+    mn[j] = mean[cc];
+    sc[j] = ga * is;
+    sh[j] = (bias ? bias[cc] : 0.f) - mn[j] * ga * is;
+    k1[j] = ga * is;
+    k2[j] = TRAIN ? ga * is * dbeta[cc] * invM : 0.f;
+    k3[j] = TRAIN ? ga * is * is * dgamma[cc] * invM : 0.f;
+  }
+
+  const long long r0 = (long long)blockIdx.y * rows_per_chunk;
+  const long long r1 = min(r0 + rows_per_chunk, M);
+  for (long long r = r0 + rg; r < r1; r += nrg) {
+    const BVec<T, VEC> xv = bvload<T, VEC>(x + r * C + c);
+    const BVec<T, VEC> dv = bvload<T, VEC>(dy + r * C + c);
     BVec<T, VEC> ov;
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
-      const float is = invstd[c + j];
-      const float xh = (DfdCvt<T>::to_f32(xv.v[j]) - mean[c + j]) * is;
-      const float ga = weight ? weight[c + j] : 1.f;
-      const float z = fmaf(ga, xh, bias ? bias[c + j] : 0.f);
+      const float xf = DfdCvt<T>::to_f32(xv.v[j]);
+      const float z = fmaf(xf, sc[j], sh[j]);
       const float g = DfdCvt<T>::to_f32(dv.v[j]) * act_bwd(z, ACT);
-      float v;
-      if (TRAIN) {
-        v = ga * is * (g - dbeta[c + j] * invM - xh * dgamma[c + j] * invM);
-      } else {
-        v = ga * is * g;
-      }
+      float v = k1[j] * g;
+      if (TRAIN) v = v - k2[j] - k3[j] * (xf - mn[j]);
       ov.v[j] = DfdCvt<T>::from_f32(v);
     }
-    *reinterpret_cast<BVec<T, VEC>*>(dx + r * C + c) = ov;
+    bvstore<T, VEC>(dx + r * C + c, ov);
   }
 }
 
@@ -345,6 +383,23 @@ Act act_from_string(const std::string& s) {
     }                                                                 \
   }()
 
+#define DISPATCH_VEC(vec, ...)                                        \
+  [&] {                                                               \
+    if (vec == 8) {                                                   \
+      constexpr int KVEC = 8;                                         \
+      return __VA_ARGS__();                                           \
+    } else if (vec == 4) {                                            \
+      constexpr int KVEC = 4;                                         \
+      return __VA_ARGS__();                                           \
+    } else if (vec == 2) {                                            \
+      constexpr int KVEC = 2;                                         \
+      return __VA_ARGS__();                                           \
+    } else {                                                          \
+      constexpr int KVEC = 1;                                         \
+      return __VA_ARGS__();                                           \
+    }                                                                 \
+  }()
+
 }  // namespace
 
 // x: (N,C,H,W) channels_last. Returns {y, save_mean, save_invstd}.
@@ -358,7 +413,6 @@ std::vector<at::Tensor> bn_act_fwd(
   const Act act = act_from_string(act_s);
   const int C = (int)x.size(1);
   const long long M = (long long)x.size(0) * x.size(2) * x.size(3);
-  const long long total = M * C;
 
   auto stream = at::cuda::getCurrentHIPStream();
   auto opts_f = x.options().dtype(at::kFloat);
@@ -371,17 +425,21 @@ std::vector<at::Tensor> bn_act_fwd(
   const float* w_p = weight.defined() ? weight.data_ptr<float>() : nullptr;
   const float* b_p = bias.defined() ? bias.data_ptr<float>() : nullptr;
 
+  const int vec = bn_pick_vec(C, (int)x.element_size());
+  const BnPlan plan = bn_plan(C / vec, M);
+  dim3 grid(plan.ctiles, plan.chunks);
+  const int lds = 256 * vec * sizeof(float);
+
   if (training) {
     auto sum = at::zeros({C}, opts_f);
     auto sumsq = at::zeros({C}, opts_f);
-    constexpr int VEC = 4;
-    const int slabs = (C + kWave * VEC - 1) / (kWave * VEC);
-    int gy = dfd_grid(M / 4 + 1, 64, kMaxGrid / slabs);
-    dim3 grid(slabs, gy);
     DISPATCH_DTYPE(x.scalar_type(), "bn_stats", [&] {
-      hipLaunchKernelGGL((bn_stats_kernel<T, VEC>), grid, dim3(256), 0, stream,
-                         (const T*)x.data_ptr(), sum.data_ptr<float>(),
-                         sumsq.data_ptr<float>(), M, C);
+      DISPATCH_VEC(vec, [&] {
+        hipLaunchKernelGGL((bn_stats_kernel<T, KVEC>), grid, dim3(256), lds, stream,
+                           (const T*)x.data_ptr(), sum.data_ptr<float>(),
+                           sumsq.data_ptr<float>(), M, C, plan.log2_cpb,
+                           plan.rows_per_chunk);
+      });
     });
     hipLaunchKernelGGL(bn_finalize_train_kernel, dim3((C + 255) / 256), dim3(256), 0, stream,
                        sum.data_ptr<float>(), sumsq.data_ptr<float>(), w_p, b_p,
@@ -397,13 +455,14 @@ std::vector<at::Tensor> bn_act_fwd(
                        scale.data_ptr<float>(), shift.data_ptr<float>(), C, (float)eps);
   }
 
-  const int block = 256;
-  const int grid_e = dfd_grid(total / 4 + 1, block);
   DISPATCH_DTYPE(x.scalar_type(), "bn_act_fwd", [&] {
     DISPATCH_ACT(act, [&] {
-      hipLaunchKernelGGL((bn_act_fwd_kernel<T, ACT>), dim3(grid_e), dim3(block), 0, stream,
-                         (const T*)x.data_ptr(), (T*)y.data_ptr(),
-                         scale.data_ptr<float>(), shift.data_ptr<float>(), total, C);
+      DISPATCH_VEC(vec, [&] {
+        hipLaunchKernelGGL((bn_act_fwd_kernel<T, ACT, KVEC>), grid, dim3(256), 0, stream,
+                           (const T*)x.data_ptr(), (T*)y.data_ptr(),
+                           scale.data_ptr<float>(), shift.data_ptr<float>(), M, C,
+                           plan.log2_cpb, plan.rows_per_chunk);
+      });
     });
   });
   return {y, save_mean, save_invstd};
@@ -419,7 +478,6 @@ std::vector<at::Tensor> bn_act_bwd(
   TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast));
   const int C = (int)x.size(1);
   const long long M = (long long)x.size(0) * x.size(2) * x.size(3);
-  const long long total = M * C;
 
   auto stream = at::cuda::getCurrentHIPStream();
   auto opts_f = x.options().dtype(at::kFloat);
@@ -430,50 +488,44 @@ std::vector<at::Tensor> bn_act_bwd(
   const float* w_p = weight.defined() ? weight.data_ptr<float>() : nullptr;
   const float* b_p = bias.defined() ? bias.data_ptr<float>() : nullptr;
 
-  constexpr int VEC = 4;
-  const int slabs = (C + kWave * VEC - 1) / (kWave * VEC);
-  int gy = dfd_grid(M / 4 + 1, 64, kMaxGrid / slabs);
-  dim3 grid(slabs, gy);
+  const int vec = bn_pick_vec(C, (int)x.element_size());
+  const BnPlan plan = bn_plan(C / vec, M);
+  dim3 grid(plan.ctiles, plan.chunks);
+  const int lds = 256 * vec * sizeof(float);
+  const float invM = 1.f / (float)M;
+
   DISPATCH_DTYPE(x.scalar_type(), "bn_bwd_reduce", [&] {
     DISPATCH_ACT(act, [&] {
-      hipLaunchKernelGGL((bn_act_bwd_reduce_kernel<T, ACT, VEC>), grid, dim3(256), 0, stream,
-                         (const T*)dy.data_ptr(), (const T*)x.data_ptr(),
-                         save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),
-                         w_p, b_p, dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), M, C);
+      DISPATCH_VEC(vec, [&] {
+        hipLaunchKernelGGL((bn_act_bwd_reduce_kernel<T, ACT, KVEC>), grid, dim3(256), lds,
+                           stream, (const T*)dy.data_ptr(), (const T*)x.data_ptr(),
+                           save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),
+                           w_p, b_p, dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                           M, C, plan.log2_cpb, plan.rows_per_chunk);
+      });
     });
   });
 
-  const int block = 256;
-  int dxvec = (x.element_size() == 4) ? 4 : 8;
-  while (dxvec > 1 && (C % dxvec)) dxvec >>= 1;
-  const int grid_e = dfd_grid(M * (C / dxvec), block);
-#define DFD_BN_DX(TRAIN, V)                                                       \
-  hipLaunchKernelGGL((bn_act_bwd_dx_kernel<T, ACT, TRAIN, V>), dim3(grid_e),      \
-                     dim3(block), 0, stream, (const T*)dy.data_ptr(),             \
-                     (const T*)x.data_ptr(), (T*)dx.data_ptr(),                   \
-                     save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),  \
-                     w_p, b_p, dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), \
-                     M, C, invM)
   DISPATCH_DTYPE(x.scalar_type(), "bn_bwd_dx", [&] {
     DISPATCH_ACT(act, [&] {
-      const float invM = 1.f / (float)M;
-      if (training) {
-        switch (dxvec) {
-          case 8: DFD_BN_DX(true, 8); break;
-          case 4: DFD_BN_DX(true, 4); break;
-          case 2: DFD_BN_DX(true, 2); break;
-          default: DFD_BN_DX(true, 1); break;
+      DISPATCH_VEC(vec, [&] {
+        if (training) {
+          hipLaunchKernelGGL((bn_act_bwd_dx_kernel<T, ACT, true, KVEC>), grid, dim3(256), 0,
+                             stream, (const T*)dy.data_ptr(), (const T*)x.data_ptr(),
+                             (T*)dx.data_ptr(), save_mean.data_ptr<float>(),
+                             save_invstd.data_ptr<float>(), w_p, b_p,
+                             dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                             M, C, invM, plan.log2_cpb, plan.rows_per_chunk);
+        } else {
+          hipLaunchKernelGGL((bn_act_bwd_dx_kernel<T, ACT, false, KVEC>), grid, dim3(256), 0,
+                             stream, (const T*)dy.data_ptr(), (const T*)x.data_ptr(),
+                             (T*)dx.data_ptr(), save_mean.data_ptr<float>(),
+                             save_invstd.data_ptr<float>(), w_p, b_p,
+                             dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                             M, C, invM, plan.log2_cpb, plan.rows_per_chunk);
         }
-      } else {
-        switch (dxvec) {
-          case 8: DFD_BN_DX(false, 8); break;
-          case 4: DFD_BN_DX(false, 4); break;
-          case 2: DFD_BN_DX(false, 2); break;
-          default: DFD_BN_DX(false, 1); break;
-        }
-      }
+      });
     });
   });
-#undef DFD_BN_DX
   return {dx, dgamma, dbeta};
 }

@@ -140,23 +140,30 @@ __global__ void dw_bwd_data_kernel(const T* __restrict__ dy, const T* __restrict
 // backward weight:
 // dw[kh,kw,c] = sum_n,ho,wo dy[n,ho,wo,c] * x[n, ho*sh-ph+kh, wo*sw-pw+kw, c]
 //
-// Block layout: blockDim.x = K * CVT threads; thread t handles
-//   kh        = t / CVT         (its own kernel row)
-//   cvec slot = t % CVT  →  c = (blockIdx.x * CVT + slot) * VEC
-// so consecutive threads touch consecutive channel vectors (coalesced).
-// Each thread loops over a chunk of (n,ho) rows × all wo, keeping K*VEC fp32
-// partial sums in registers, then one atomicAdd per partial.
+// Block layout: 256 threads split as (channel-slot, kh, row-group):
+//   slot = tid & (cpb-1)            cpb = power-of-two slots sized from C/VEC
+//   rest = tid >> log2_cpb;  kh = rest % K;  row-group = rest / K
+// Consecutive threads touch consecutive channel vectors (coalesced); the
+// slot count adapts to C so small-C layers (C=48 → 6 vectors) still fill
+// the block with row parallelism instead of idling 58/64 lanes (the
+// fixed-64-slot first cut ran 1.15 ms on the k3 layers).
+// Each thread owns one kernel row kh, loops its rows × all wo keeping K*VEC
+// fp32 partials in registers, then one atomicAdd per partial.
 // ---------------------------------------------------------------------------
-template <typename T, int K, int VEC, int CVT>
+template <typename T, int K, int VEC>
 __global__ void dw_bwd_weight_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                                      float* __restrict__ dw, int N, int C, int H, int W,
                                      int Ho, int Wo, int sh, int sw, int ph, int pw,
-                                     int rows_per_block) {
-  const int t = threadIdx.x;
-  const int kh = t / CVT;
-  const int slot = t % CVT;
-  const int cvec = blockIdx.x * CVT + slot;
+                                     int log2_cpb, int rows_per_chunk) {
+  const int cpb = 1 << log2_cpb;
+  const int slot = threadIdx.x & (cpb - 1);
+  const int rest = threadIdx.x >> log2_cpb;
+  const int kh = rest % K;
+  const int rg = rest / K;
+  const int nrg = (blockDim.x >> log2_cpb) / K;
+  if (rg >= nrg) return;
   const int cv = C / VEC;
+  const int cvec = blockIdx.x * cpb + slot;
   if (cvec >= cv) return;
   const int c = cvec * VEC;
 
@@ -165,10 +172,10 @@ __global__ void dw_bwd_weight_kernel(const T* __restrict__ dy, const T* __restri
   for (int i = 0; i < K * VEC; ++i) acc[i] = 0.f;
 
   const long long rows_total = (long long)N * Ho;
-  const long long row0 = (long long)blockIdx.y * rows_per_block;
-  const long long row1 = min(row0 + rows_per_block, rows_total);
+  const long long row0 = (long long)blockIdx.y * rows_per_chunk;
+  const long long row1 = min(row0 + rows_per_chunk, rows_total);
 
-  for (long long r = row0; r < row1; ++r) {
+  for (long long r = row0 + rg; r < row1; r += nrg) {
     const int ho = (int)(r % Ho);
     const int n = (int)(r / Ho);
     const int hi = ho * sh - ph + kh;
@@ -283,22 +290,29 @@ void bwd_data_ktype(const at::Tensor& dy, const at::Tensor& w, at::Tensor& dx,
   }
 }
 
-constexpr int kCVT = 64;  // channel-vectors per block in bwd-weight
-
 template <typename T, int K, int VEC>
 void launch_bwd_weight(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw,
                        const Geom& g, hipStream_t stream) {
   const int cv = g.C / VEC;
-  const int grid_x = (cv + kCVT - 1) / kCVT;
+  // power-of-two channel slots per block, ≤64, sized to cover cv when small
+  // and capped so at least one (kh, row-group) pair fits: cpb * K ≤ 256
+  int log2_cpb = 0;
+  while ((1 << log2_cpb) < cv && log2_cpb < 6) ++log2_cpb;
+  while (log2_cpb > 0 && (256 >> log2_cpb) / K < 1) --log2_cpb;
+  const int cpb = 1 << log2_cpb;
+  const int grid_x = (cv + cpb - 1) / cpb;
+  const int nrg = (256 >> log2_cpb) / K;  // row-groups per block
   const long long rows_total = (long long)g.N * g.Ho;
   // aim for ~2048 blocks total to fill 256 CUs
-  int grid_y = (int)std::min<long long>(rows_total, std::max(1, kMaxGrid / grid_x));
-  const int rows_per_block = (int)((rows_total + grid_y - 1) / grid_y);
-  grid_y = (int)((rows_total + rows_per_block - 1) / rows_per_block);
-  dim3 grid(grid_x, grid_y);
-  dw_bwd_weight_kernel<T, K, VEC, kCVT><<<grid, K * kCVT, 0, stream>>>(
+  long long chunks = std::max<long long>(1, kMaxGrid / grid_x);
+  const long long max_chunks = (rows_total + nrg - 1) / std::max(1, nrg);
+  if (chunks > max_chunks) chunks = max_chunks;
+  const int rows_per_chunk = (int)((rows_total + chunks - 1) / chunks);
+  chunks = (rows_total + rows_per_chunk - 1) / rows_per_chunk;
+  dim3 grid(grid_x, (unsigned)chunks);
+  dw_bwd_weight_kernel<T, K, VEC><<<grid, 256, 0, stream>>>(
       (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (float*)dw.data_ptr(), g.N, g.C,
-      g.H, g.W, g.Ho, g.Wo, g.sh, g.sw, g.ph, g.pw, rows_per_block);
+      g.H, g.W, g.Ho, g.Wo, g.sh, g.sw, g.ph, g.pw, log2_cpb, rows_per_chunk);
 }
 
 template <typename T>
