@@ -7,7 +7,10 @@ Run on a GPU box:  python tools/bench_kernels.py [--batch 192] [--ops dw,bn,se]
 """
 
 import argparse
+import os
 import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
