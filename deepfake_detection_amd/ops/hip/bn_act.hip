@@ -66,6 +66,10 @@ BnPlan bn_plan(int cv, long long rows) {
   p.ctiles = (cv + cpb - 1) / cpb;
   long long want = (2048 + p.ctiles - 1) / p.ctiles;
   const int nrg = 256 >> p.log2_cpb;
+  // floor of ~16 row-iterations per thread: small tensors otherwise explode
+  // into 2048 blocks whose per-block atomics serialize on C addresses
+  long long by_iters = rows / ((long long)nrg * 16);
+  if (want > by_iters) want = by_iters;
   long long max_chunks = (rows + nrg - 1) / nrg;
   if (want > max_chunks) want = max_chunks;
   if (want < 1) want = 1;

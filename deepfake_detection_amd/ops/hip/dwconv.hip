@@ -155,56 +155,80 @@ __global__ void dw_bwd_weight_kernel(const T* __restrict__ dy, const T* __restri
                                      float* __restrict__ dw, int N, int C, int H, int W,
                                      int Ho, int Wo, int sh, int sw, int ph, int pw,
                                      int log2_cpb, int rows_per_chunk) {
+  extern __shared__ float lds[];  // [blockDim.x * VEC]
   const int cpb = 1 << log2_cpb;
   const int slot = threadIdx.x & (cpb - 1);
   const int rest = threadIdx.x >> log2_cpb;
   const int kh = rest % K;
   const int rg = rest / K;
   const int nrg = (blockDim.x >> log2_cpb) / K;
-  if (rg >= nrg) return;
   const int cv = C / VEC;
   const int cvec = blockIdx.x * cpb + slot;
-  if (cvec >= cv) return;
+  // no early return: every thread reaches the barriers below
+  const bool active = (rg < nrg) && (cvec < cv);
   const int c = cvec * VEC;
 
   float acc[K * VEC];
 #pragma unroll
   for (int i = 0; i < K * VEC; ++i) acc[i] = 0.f;
 
-  const long long rows_total = (long long)N * Ho;
-  const long long row0 = (long long)blockIdx.y * rows_per_chunk;
-  const long long row1 = min(row0 + rows_per_chunk, rows_total);
+  if (active) {
+    const long long rows_total = (long long)N * Ho;
+    const long long row0 = (long long)blockIdx.y * rows_per_chunk;
+    const long long row1 = min(row0 + rows_per_chunk, rows_total);
 
-  for (long long r = row0 + rg; r < row1; r += nrg) {
-    const int ho = (int)(r % Ho);
-    const int n = (int)(r / Ho);
-    const int hi = ho * sh - ph + kh;
-    if (hi < 0 || hi >= H) continue;
-    const T* dy_row = dy + (((long long)n * Ho + ho) * Wo) * C + c;
-    const T* x_row = x + (((long long)n * H + hi) * W) * C + c;
-    for (int wo = 0; wo < Wo; ++wo) {
-      const TVec<T, VEC> gv = vload<T, VEC>(dy_row + (long long)wo * C);
-      float gf[VEC];
+    for (long long r = row0 + rg; r < row1; r += nrg) {
+      const int ho = (int)(r % Ho);
+      const int n = (int)(r / Ho);
+      const int hi = ho * sh - ph + kh;
+      if (hi < 0 || hi >= H) continue;
+      const T* dy_row = dy + (((long long)n * Ho + ho) * Wo) * C + c;
+      const T* x_row = x + (((long long)n * H + hi) * W) * C + c;
+      for (int wo = 0; wo < Wo; ++wo) {
+        const TVec<T, VEC> gv = vload<T, VEC>(dy_row + (long long)wo * C);
+        float gf[VEC];
 #pragma unroll
-      for (int i = 0; i < VEC; ++i) gf[i] = DfdCvt<T>::to_f32(gv.v[i]);
-      const int wi0 = wo * sw - pw;
+        for (int i = 0; i < VEC; ++i) gf[i] = DfdCvt<T>::to_f32(gv.v[i]);
+        const int wi0 = wo * sw - pw;
 #pragma unroll
-      for (int kw = 0; kw < K; ++kw) {
-        const int wi = wi0 + kw;
-        if (wi < 0 || wi >= W) continue;
-        const TVec<T, VEC> xv = vload<T, VEC>(x_row + (long long)wi * C);
+        for (int kw = 0; kw < K; ++kw) {
+          const int wi = wi0 + kw;
+          if (wi < 0 || wi >= W) continue;
+          const TVec<T, VEC> xv = vload<T, VEC>(x_row + (long long)wi * C);
 #pragma unroll
-        for (int i = 0; i < VEC; ++i)
-          acc[kw * VEC + i] += gf[i] * DfdCvt<T>::to_f32(xv.v[i]);
+          for (int i = 0; i < VEC; ++i)
+            acc[kw * VEC + i] += gf[i] * DfdCvt<T>::to_f32(xv.v[i]);
+        }
       }
     }
   }
 
+  // fold the nrg row-groups in LDS (one kw plane at a time) so each block
+  // issues only K*K*C_covered atomics — the all-threads-atomic first cut
+  // serialized millions of atomicAdds on C*K*K addresses (4.3 ms on the
+  // C=48 k3 layer; see tools/bench_kernels.py)
+  float* my = lds + (size_t)threadIdx.x * VEC;
+  const int pow2 = 1 << (31 - __clz(nrg > 0 ? nrg : 1));
+  for (int kw = 0; kw < K; ++kw) {
+    __syncthreads();
 #pragma unroll
-  for (int kw = 0; kw < K; ++kw)
+    for (int i = 0; i < VEC; ++i) my[i] = acc[kw * VEC + i];
+    __syncthreads();
+    for (int step = pow2; step > 0; step >>= 1) {
+      if (rg < step && rg + step < nrg) {
+        const float* other =
+            lds + ((size_t)(slot + ((kh + K * (rg + step)) << log2_cpb))) * VEC;
 #pragma unroll
-    for (int i = 0; i < VEC; ++i)
-      atomicAdd(dw + ((long long)kh * K + kw) * C + c + i, acc[kw * VEC + i]);
+        for (int i = 0; i < VEC; ++i) my[i] += other[i];
+      }
+      __syncthreads();
+    }
+    if (rg == 0 && active) {
+#pragma unroll
+      for (int i = 0; i < VEC; ++i)
+        atomicAdd(dw + ((long long)kh * K + kw) * C + c + i, my[i]);
+    }
+  }
 }
 
 // ---------------------------------------------------------------------------
@@ -303,14 +327,19 @@ void launch_bwd_weight(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw
   const int grid_x = (cv + cpb - 1) / cpb;
   const int nrg = (256 >> log2_cpb) / K;  // row-groups per block
   const long long rows_total = (long long)g.N * g.Ho;
-  // aim for ~2048 blocks total to fill 256 CUs
+  // aim for ~2048 blocks total to fill 256 CUs, but keep ≥~4 row-iterations
+  // per thread so per-block atomics amortize
   long long chunks = std::max<long long>(1, kMaxGrid / grid_x);
+  const long long by_iters = rows_total / std::max(1, nrg * 4);
+  if (chunks > by_iters) chunks = by_iters;
   const long long max_chunks = (rows_total + nrg - 1) / std::max(1, nrg);
   if (chunks > max_chunks) chunks = max_chunks;
+  if (chunks < 1) chunks = 1;
   const int rows_per_chunk = (int)((rows_total + chunks - 1) / chunks);
   chunks = (rows_total + rows_per_chunk - 1) / rows_per_chunk;
   dim3 grid(grid_x, (unsigned)chunks);
-  dw_bwd_weight_kernel<T, K, VEC><<<grid, 256, 0, stream>>>(
+  const int lds = 256 * VEC * sizeof(float);
+  dw_bwd_weight_kernel<T, K, VEC><<<grid, 256, lds, stream>>>(
       (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (float*)dw.data_ptr(), g.N, g.C,
       g.H, g.W, g.Ho, g.Wo, g.sh, g.sw, g.ph, g.pw, log2_cpb, rows_per_chunk);
 }

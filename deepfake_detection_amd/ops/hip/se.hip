@@ -270,6 +270,9 @@ ChunkPlan plan_chunks(long long N, int cv, long long HW) {
   const long long base = N * p.ctiles;
   long long want = (2048 + base - 1) / base;  // chunks to reach ~2048 blocks
   const int nrg = 256 >> p.log2_cpb;
+  // keep ≥~16 row-iterations per thread to amortize the per-block atomics
+  long long by_iters = HW / ((long long)nrg * 16);
+  if (want > by_iters) want = by_iters;
   long long max_chunks = (HW + nrg - 1) / nrg;
   if (want > max_chunks) want = max_chunks;
   if (want < 1) want = 1;
