@@ -87,6 +87,72 @@ __global__ void dw_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
 }
 
 // ---------------------------------------------------------------------------
+// forward, stride 1, TW consecutive outputs per thread.
+// Loads per output drop from K*K to ~K*(K+TW-1)/TW (x) and weight loads
+// amortize by TW; bwd-data for stride 1 reuses this kernel with the packed
+// weight flipped in (kh,kw) and padding (K-1-p) — it is the same stride-1
+// correlation.
+// ---------------------------------------------------------------------------
+template <typename T, int K, int VEC, int TW>
+__global__ void dw_fwd_s1_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                                 T* __restrict__ y, int N, int C, int H, int W,
+                                 int Ho, int Wo, int ph, int pw) {
+  const int cv = C / VEC;
+  const int wt = (Wo + TW - 1) / TW;  // wo tiles per row
+  const long long total = (long long)N * Ho * wt * cv;
+  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long long)gridDim.x * blockDim.x) {
+    const int c = (int)(idx % cv) * VEC;
+    long long p = idx / cv;
+    const int wo0 = (int)(p % wt) * TW;
+    p /= wt;
+    const int ho = (int)(p % Ho);
+    const int n = (int)(p / Ho);
+
+    float acc[TW][VEC];
+#pragma unroll
+    for (int t = 0; t < TW; ++t)
+#pragma unroll
+      for (int i = 0; i < VEC; ++i) acc[t][i] = 0.f;
+
+    const int hi0 = ho - ph;
+    const int wi0 = wo0 - pw;
+#pragma unroll
+    for (int kh = 0; kh < K; ++kh) {
+      const int hi = hi0 + kh;
+      if (hi < 0 || hi >= H) continue;
+      const T* xrow = x + (((long long)n * H + hi) * W) * C + c;
+      const T* wrow = w + ((long long)kh * K) * C + c;
+#pragma unroll
+      for (int col = 0; col < K + TW - 1; ++col) {
+        const int wi = wi0 + col;
+        if (wi < 0 || wi >= W) continue;
+        const TVec<T, VEC> xv = vload<T, VEC>(xrow + (long long)wi * C);
+        // this column contributes to outputs t with 0 <= col - t < K
+#pragma unroll
+        for (int t = 0; t < TW; ++t) {
+          const int kw = col - t;
+          if (kw < 0 || kw >= K) continue;
+          const TVec<T, VEC> wv = vload<T, VEC>(wrow + (long long)kw * C);
+#pragma unroll
+          for (int i = 0; i < VEC; ++i)
+            acc[t][i] += DfdCvt<T>::to_f32(xv.v[i]) * DfdCvt<T>::to_f32(wv.v[i]);
+        }
+      }
+    }
+    T* yrow = y + (((long long)n * Ho + ho) * Wo) * C + c;
+#pragma unroll
+    for (int t = 0; t < TW; ++t) {
+      if (wo0 + t >= Wo) break;
+      TVec<T, VEC> yv;
+#pragma unroll
+      for (int i = 0; i < VEC; ++i) yv.v[i] = DfdCvt<T>::from_f32(acc[t][i]);
+      vstore<T, VEC>(yrow + (long long)(wo0 + t) * C, yv);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // backward data:
 // dx[n,hi,wi,c] = sum over (kh,kw) with hi = ho*sh-ph+kh solvable:
 //                 dy[n,ho,wo,c] * w[kh,kw,c]
@@ -203,12 +269,14 @@ __global__ void dw_bwd_weight_kernel(const T* __restrict__ dy, const T* __restri
     }
   }
 
-  // fold the nrg row-groups in LDS (one kw plane at a time) so each block
-  // issues only K*K*C_covered atomics — the all-threads-atomic first cut
-  // serialized millions of atomicAdds on C*K*K addresses (4.3 ms on the
-  // C=48 k3 layer; see tools/bench_kernels.py)
+  // fold the nrg row-groups in LDS (one kw plane at a time), then STORE the
+  // block's partial to dw_part[chunk] — no atomics. A fold-then-atomic cut
+  // still issued K*K*C atomics × 2048 blocks (≈13M) and fp32 global-atomic
+  // throughput capped the kernel (~1.2 ms on C=288 k5); partial stores +
+  // a tiny second-stage reduction run at stream rate.
   float* my = lds + (size_t)threadIdx.x * VEC;
   const int pow2 = 1 << (31 - __clz(nrg > 0 ? nrg : 1));
+  float* part = dw + (long long)blockIdx.y * K * K * C;  // dw is [chunks,K,K,C]
   for (int kw = 0; kw < K; ++kw) {
     __syncthreads();
 #pragma unroll
@@ -226,9 +294,20 @@ __global__ void dw_bwd_weight_kernel(const T* __restrict__ dy, const T* __restri
     if (rg == 0 && active) {
 #pragma unroll
       for (int i = 0; i < VEC; ++i)
-        atomicAdd(dw + ((long long)kh * K + kw) * C + c + i, my[i]);
+        part[((long long)kh * K + kw) * C + c + i] = my[i];
     }
   }
+}
+
+// stage 2: dw[i] = sum_z part[z][i], i in [0, K*K*C)
+__global__ void dw_bwd_weight_reduce_kernel(const float* __restrict__ part,
+                                            float* __restrict__ dw, long long kkc,
+                                            int chunks) {
+  const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= kkc) return;
+  float s = 0.f;
+  for (int z = 0; z < chunks; ++z) s += part[(long long)z * kkc + i];
+  dw[i] = s;
 }
 
 // ---------------------------------------------------------------------------
@@ -249,8 +328,17 @@ struct Geom {
 template <typename T, int K, int VEC>
 void launch_fwd(const at::Tensor& x, const at::Tensor& w, at::Tensor& y, const Geom& g,
                 hipStream_t stream) {
-  const long long total = (long long)g.N * g.Ho * g.Wo * (g.C / VEC);
   const int block = 256;
+  if (g.sh == 1 && g.sw == 1) {
+    constexpr int TW = 4;
+    const int wt = (g.Wo + TW - 1) / TW;
+    const long long total = (long long)g.N * g.Ho * wt * (g.C / VEC);
+    dw_fwd_s1_kernel<T, K, VEC, TW><<<dfd_grid(total, block), block, 0, stream>>>(
+        (const T*)x.data_ptr(), (const T*)w.data_ptr(), (T*)y.data_ptr(), g.N, g.C, g.H,
+        g.W, g.Ho, g.Wo, g.ph, g.pw);
+    return;
+  }
+  const long long total = (long long)g.N * g.Ho * g.Wo * (g.C / VEC);
   dw_fwd_kernel<T, K, VEC><<<dfd_grid(total, block), block, 0, stream>>>(
       (const T*)x.data_ptr(), (const T*)w.data_ptr(), (T*)y.data_ptr(), g.N, g.C, g.H,
       g.W, g.Ho, g.Wo, g.sh, g.sw, g.ph, g.pw);
@@ -339,9 +427,14 @@ void launch_bwd_weight(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw
   chunks = (rows_total + rows_per_chunk - 1) / rows_per_chunk;
   dim3 grid(grid_x, (unsigned)chunks);
   const int lds = 256 * VEC * sizeof(float);
+  const long long kkc = (long long)K * K * g.C;
+  auto part = at::empty({chunks, (long long)K, (long long)K, (long long)g.C},
+                        dw.options());
   dw_bwd_weight_kernel<T, K, VEC><<<grid, 256, lds, stream>>>(
-      (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (float*)dw.data_ptr(), g.N, g.C,
+      (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (float*)part.data_ptr(), g.N, g.C,
       g.H, g.W, g.Ho, g.Wo, g.sh, g.sw, g.ph, g.pw, log2_cpb, rows_per_chunk);
+  dw_bwd_weight_reduce_kernel<<<dfd_grid(kkc, 256, 1 << 20), 256, 0, stream>>>(
+      (const float*)part.data_ptr(), (float*)dw.data_ptr(), kkc, (int)chunks);
 }
 
 template <typename T>
@@ -424,6 +517,22 @@ at::Tensor dw_conv2d_bwd_data(at::Tensor dy, at::Tensor w_packed, int64_t H, int
   auto stream = at::hip::getCurrentHIPStream().stream();
   const auto stype = dy.scalar_type();
   const int Ki = (int)K;
+  if (sh == 1 && sw == 1) {
+    // stride-1 bwd-data is the same correlation with (kh,kw)-flipped weights
+    // and padding K-1-p: run it through the TW-tiled forward kernel
+    auto w_flip = at::flip(w_packed, {0, 1}).contiguous();
+    Geom gf;
+    gf.N = g.N; gf.C = g.C; gf.H = g.Ho; gf.W = g.Wo;
+    gf.Ho = g.H; gf.Wo = g.W; gf.sh = 1; gf.sw = 1;
+    gf.ph = Ki - 1 - g.ph; gf.pw = Ki - 1 - g.pw;
+    switch (stype) {
+      case at::kBFloat16: fwd_ktype<__hip_bfloat16>(dy, w_flip, dx, gf, Ki, vec, stream); break;
+      case at::kHalf: fwd_ktype<__half>(dy, w_flip, dx, gf, Ki, vec, stream); break;
+      case at::kFloat: fwd_ktype<float>(dy, w_flip, dx, gf, Ki, vec, stream); break;
+      default: TORCH_CHECK(false, "dwconv: unsupported dtype");
+    }
+    return dx;
+  }
   switch (stype) {
     case at::kBFloat16: bwd_data_ktype<__hip_bfloat16>(dy, w_packed, dx, g, Ki, vec, stream); break;
     case at::kHalf: bwd_data_ktype<__half>(dy, w_packed, dx, g, Ki, vec, stream); break;
@@ -441,7 +550,7 @@ at::Tensor dw_conv2d_bwd_weight(at::Tensor dy, at::Tensor x, int64_t K, int64_t 
   TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast), "dwconv: dy must be channels_last");
   Geom g = make_geom(x.size(0), x.size(1), x.size(2), x.size(3), K, sh, sw, ph, pw);
   TORCH_CHECK(g.Ho == dy.size(2) && g.Wo == dy.size(3), "dwconv bwd_weight: geometry mismatch");
-  auto dw = at::zeros({K, K, (long long)g.C}, x.options().dtype(at::kFloat));
+  auto dw = at::empty({K, K, (long long)g.C}, x.options().dtype(at::kFloat));
   const int vec = pick_vec(g.C, (int)x.element_size());
   auto stream = at::hip::getCurrentHIPStream().stream();
   const auto stype = x.scalar_type();
