@@ -64,19 +64,35 @@ __global__ void dw_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
 
     const int hi0 = ho * sh - ph;
     const int wi0 = wo * sw - pw;
+    const bool interior = wi0 >= 0 && wi0 + K - 1 < W;
 #pragma unroll
     for (int kh = 0; kh < K; ++kh) {
       const int hi = hi0 + kh;
       if (hi < 0 || hi >= H) continue;
+      const T* xrow = x + (((long long)n * H + hi) * W) * C + c;
+      if (interior) {
+        TVec<T, VEC> xv[K];
 #pragma unroll
-      for (int kw = 0; kw < K; ++kw) {
-        const int wi = wi0 + kw;
-        if (wi < 0 || wi >= W) continue;
-        const TVec<T, VEC> xv = vload<T, VEC>(x + (((long long)n * H + hi) * W + wi) * C + c);
-        const TVec<T, VEC> wv = vload<T, VEC>(w + ((long long)kh * K + kw) * C + c);
+        for (int kw = 0; kw < K; ++kw)
+          xv[kw] = vload<T, VEC>(xrow + (long long)(wi0 + kw) * C);
 #pragma unroll
-        for (int i = 0; i < VEC; ++i)
-          acc[i] += DfdCvt<T>::to_f32(xv.v[i]) * DfdCvt<T>::to_f32(wv.v[i]);
+        for (int kw = 0; kw < K; ++kw) {
+          const TVec<T, VEC> wv = vload<T, VEC>(w + ((long long)kh * K + kw) * C + c);
+#pragma unroll
+          for (int i = 0; i < VEC; ++i)
+            acc[i] += DfdCvt<T>::to_f32(xv[kw].v[i]) * DfdCvt<T>::to_f32(wv.v[i]);
+        }
+      } else {
+#pragma unroll
+        for (int kw = 0; kw < K; ++kw) {
+          const int wi = wi0 + kw;
+          if (wi < 0 || wi >= W) continue;
+          const TVec<T, VEC> xv = vload<T, VEC>(xrow + (long long)wi * C);
+          const TVec<T, VEC> wv = vload<T, VEC>(w + ((long long)kh * K + kw) * C + c);
+#pragma unroll
+          for (int i = 0; i < VEC; ++i)
+            acc[i] += DfdCvt<T>::to_f32(xv.v[i]) * DfdCvt<T>::to_f32(wv.v[i]);
+        }
       }
     }
     TVec<T, VEC> yv;
@@ -117,26 +133,48 @@ __global__ void dw_fwd_s1_kernel(const T* __restrict__ x, const T* __restrict__ 
 
     const int hi0 = ho - ph;
     const int wi0 = wo0 - pw;
+    const bool interior = wi0 >= 0 && wi0 + K + TW - 2 < W;
 #pragma unroll
     for (int kh = 0; kh < K; ++kh) {
       const int hi = hi0 + kh;
       if (hi < 0 || hi >= H) continue;
       const T* xrow = x + (((long long)n * H + hi) * W) * C + c;
       const T* wrow = w + ((long long)kh * K) * C + c;
+      if (interior) {
+        // branchless: issue all K+TW-1 column loads, then consume (per-load
+        // guards would serialize each load behind s_waitcnt vmcnt(0))
+        TVec<T, VEC> xv[K + TW - 1];
 #pragma unroll
-      for (int col = 0; col < K + TW - 1; ++col) {
-        const int wi = wi0 + col;
-        if (wi < 0 || wi >= W) continue;
-        const TVec<T, VEC> xv = vload<T, VEC>(xrow + (long long)wi * C);
-        // this column contributes to outputs t with 0 <= col - t < K
+        for (int col = 0; col < K + TW - 1; ++col)
+          xv[col] = vload<T, VEC>(xrow + (long long)(wi0 + col) * C);
 #pragma unroll
-        for (int t = 0; t < TW; ++t) {
-          const int kw = col - t;
-          if (kw < 0 || kw >= K) continue;
-          const TVec<T, VEC> wv = vload<T, VEC>(wrow + (long long)kw * C);
+        for (int col = 0; col < K + TW - 1; ++col) {
 #pragma unroll
-          for (int i = 0; i < VEC; ++i)
-            acc[t][i] += DfdCvt<T>::to_f32(xv.v[i]) * DfdCvt<T>::to_f32(wv.v[i]);
+          for (int t = 0; t < TW; ++t) {
+            const int kw = col - t;
+            if (kw < 0 || kw >= K) continue;
+            const TVec<T, VEC> wv = vload<T, VEC>(wrow + (long long)kw * C);
+#pragma unroll
+            for (int i = 0; i < VEC; ++i)
+              acc[t][i] += DfdCvt<T>::to_f32(xv[col].v[i]) * DfdCvt<T>::to_f32(wv.v[i]);
+          }
+        }
+      } else {
+#pragma unroll
+        for (int col = 0; col < K + TW - 1; ++col) {
+          const int wi = wi0 + col;
+          if (wi < 0 || wi >= W) continue;
+          const TVec<T, VEC> xv = vload<T, VEC>(xrow + (long long)wi * C);
+          // this column contributes to outputs t with 0 <= col - t < K
+#pragma unroll
+          for (int t = 0; t < TW; ++t) {
+            const int kw = col - t;
+            if (kw < 0 || kw >= K) continue;
+            const TVec<T, VEC> wv = vload<T, VEC>(wrow + (long long)kw * C);
+#pragma unroll
+            for (int i = 0; i < VEC; ++i)
+              acc[t][i] += DfdCvt<T>::to_f32(xv.v[i]) * DfdCvt<T>::to_f32(wv.v[i]);
+          }
         }
       }
     }
@@ -242,6 +280,11 @@ __global__ void dw_bwd_weight_kernel(const T* __restrict__ dy, const T* __restri
     const long long rows_total = (long long)N * Ho;
     const long long row0 = (long long)blockIdx.y * rows_per_chunk;
     const long long row1 = min(row0 + rows_per_chunk, rows_total);
+    // interior range where all K columns are in-bounds: guarded per-kw
+    // branches otherwise force the compiler to serialize every load behind
+    // s_waitcnt vmcnt(0) (measured 10x off the VALU bound)
+    const int wo_lo = min(Wo, max(0, (pw + sw - 1) / sw));
+    const int wo_hi = max(wo_lo, min(Wo, (W - K + pw) / sw + 1));
 
     for (long long r = row0 + rg; r < row1; r += nrg) {
       const int ho = (int)(r % Ho);
@@ -250,7 +293,8 @@ __global__ void dw_bwd_weight_kernel(const T* __restrict__ dy, const T* __restri
       if (hi < 0 || hi >= H) continue;
       const T* dy_row = dy + (((long long)n * Ho + ho) * Wo) * C + c;
       const T* x_row = x + (((long long)n * H + hi) * W) * C + c;
-      for (int wo = 0; wo < Wo; ++wo) {
+
+      auto guarded = [&](int wo) {
         const TVec<T, VEC> gv = vload<T, VEC>(dy_row + (long long)wo * C);
         float gf[VEC];
 #pragma unroll
@@ -265,7 +309,25 @@ __global__ void dw_bwd_weight_kernel(const T* __restrict__ dy, const T* __restri
           for (int i = 0; i < VEC; ++i)
             acc[kw * VEC + i] += gf[i] * DfdCvt<T>::to_f32(xv.v[i]);
         }
+      };
+      for (int wo = 0; wo < wo_lo; ++wo) guarded(wo);
+      for (int wo = wo_lo; wo < wo_hi; ++wo) {
+        // branchless interior: issue dy + all K x loads, then consume
+        const TVec<T, VEC> gv = vload<T, VEC>(dy_row + (long long)wo * C);
+        const T* xp = x_row + (long long)(wo * sw - pw) * C;
+        TVec<T, VEC> xv[K];
+#pragma unroll
+        for (int kw = 0; kw < K; ++kw) xv[kw] = vload<T, VEC>(xp + (long long)kw * C);
+        float gf[VEC];
+#pragma unroll
+        for (int i = 0; i < VEC; ++i) gf[i] = DfdCvt<T>::to_f32(gv.v[i]);
+#pragma unroll
+        for (int kw = 0; kw < K; ++kw)
+#pragma unroll
+          for (int i = 0; i < VEC; ++i)
+            acc[kw * VEC + i] += gf[i] * DfdCvt<T>::to_f32(xv[kw].v[i]);
       }
+      for (int wo = wo_hi; wo < Wo; ++wo) guarded(wo);
     }
   }
 
