@@ -103,7 +103,23 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum
     const long long r0 = (long long)blockIdx.y * rows_per_chunk;
     const long long r1 = min(r0 + rows_per_chunk, M);
     const T* xc = x + c;
-    for (long long r = r0 + rg; r < r1; r += nrg) {
+    long long r = r0 + rg;
+    // 4-row batches: independent loads issue together instead of one
+    // load-wait-consume cycle per row
+    for (; r + 3 * (long long)nrg < r1; r += 4 * (long long)nrg) {
+      BVec<T, VEC> v[4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) v[u] = bvload<T, VEC>(xc + (r + u * (long long)nrg) * C);
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) {
+          const float f = DfdCvt<T>::to_f32(v[u].v[j]);
+          s[j] += f;
+          q[j] += f * f;
+        }
+    }
+    for (; r < r1; r += nrg) {
       const BVec<T, VEC> xv = bvload<T, VEC>(xc + r * C);
 #pragma unroll
       for (int j = 0; j < VEC; ++j) {
@@ -211,7 +227,22 @@ __global__ void bn_act_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
 
   const long long r0 = (long long)blockIdx.y * rows_per_chunk;
   const long long r1 = min(r0 + rows_per_chunk, M);
-  for (long long r = r0 + rg; r < r1; r += nrg) {
+  long long r = r0 + rg;
+  for (; r + nrg < r1; r += 2 * (long long)nrg) {
+    BVec<T, VEC> xv[2], yv[2];
+#pragma unroll
+    for (int u = 0; u < 2; ++u) xv[u] = bvload<T, VEC>(x + (r + u * (long long)nrg) * C + c);
+#pragma unroll
+    for (int u = 0; u < 2; ++u) {
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        const float z = fmaf(DfdCvt<T>::to_f32(xv[u].v[j]), sc[j], sh[j]);
+        yv[u].v[j] = DfdCvt<T>::from_f32(act_fwd(z, ACT));
+      }
+      bvstore<T, VEC>(y + (r + u * (long long)nrg) * C + c, yv[u]);
+    }
+  }
+  for (; r < r1; r += nrg) {
     const BVec<T, VEC> xv = bvload<T, VEC>(x + r * C + c);
     BVec<T, VEC> yv;
 #pragma unroll
@@ -258,7 +289,26 @@ __global__ void bn_act_bwd_reduce_kernel(
   if (active) {
     const long long r0 = (long long)blockIdx.y * rows_per_chunk;
     const long long r1 = min(r0 + rows_per_chunk, M);
-    for (long long r = r0 + rg; r < r1; r += nrg) {
+    long long r = r0 + rg;
+    for (; r + nrg < r1; r += 2 * (long long)nrg) {
+      BVec<T, VEC> xv[2], dv[2];
+#pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        xv[u] = bvload<T, VEC>(x + (r + u * (long long)nrg) * C + c);
+        dv[u] = bvload<T, VEC>(dy + (r + u * (long long)nrg) * C + c);
+      }
+#pragma unroll
+      for (int u = 0; u < 2; ++u)
+#pragma unroll
+        for (int j = 0; j < VEC; ++j) {
+          const float xh = (DfdCvt<T>::to_f32(xv[u].v[j]) - mn[j]) * is[j];
+          const float z = fmaf(ga[j], xh, be[j]);
+          const float g = DfdCvt<T>::to_f32(dv[u].v[j]) * act_bwd(z, ACT);
+          sg[j] += g;
+          sgx[j] += g * xh;
+        }
+    }
+    for (; r < r1; r += nrg) {
       const BVec<T, VEC> xv = bvload<T, VEC>(x + r * C + c);
       const BVec<T, VEC> dv = bvload<T, VEC>(dy + r * C + c);
 #pragma unroll
@@ -333,7 +383,29 @@ __global__ void bn_act_bwd_dx_kernel(
 
   const long long r0 = (long long)blockIdx.y * rows_per_chunk;
   const long long r1 = min(r0 + rows_per_chunk, M);
-  for (long long r = r0 + rg; r < r1; r += nrg) {
+  long long r = r0 + rg;
+  for (; r + nrg < r1; r += 2 * (long long)nrg) {
+    BVec<T, VEC> xv[2], dv[2], ov[2];
+#pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      xv[u] = bvload<T, VEC>(x + (r + u * (long long)nrg) * C + c);
+      dv[u] = bvload<T, VEC>(dy + (r + u * (long long)nrg) * C + c);
+    }
+#pragma unroll
+    for (int u = 0; u < 2; ++u) {
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) {
+        const float xf = DfdCvt<T>::to_f32(xv[u].v[j]);
+        const float z = fmaf(xf, sc[j], sh[j]);
+        const float g = DfdCvt<T>::to_f32(dv[u].v[j]) * act_bwd(z, ACT);
+        float v = k1[j] * g;
+        if (TRAIN) v = v - k2[j] - k3[j] * (xf - mn[j]);
+        ov[u].v[j] = DfdCvt<T>::from_f32(v);
+      }
+      bvstore<T, VEC>(dx + (r + u * (long long)nrg) * C + c, ov[u]);
+    }
+  }
+  for (; r < r1; r += nrg) {
     const BVec<T, VEC> xv = bvload<T, VEC>(x + r * C + c);
     const BVec<T, VEC> dv = bvload<T, VEC>(dy + r * C + c);
     BVec<T, VEC> ov;
