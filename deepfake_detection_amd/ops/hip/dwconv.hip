@@ -361,15 +361,22 @@ __global__ void dw_bwd_weight_kernel(const T* __restrict__ dy, const T* __restri
   }
 }
 
-// stage 2: dw[i] = sum_z part[z][i], i in [0, K*K*C)
+// stage 2: dw[i] = sum_z part[z][i], i in [0, K*K*C). z is split across
+// blockIdx.y (K*K*C alone is only ~7k threads — not enough to fill the chip
+// when chunks is large), partial sums combine with a few atomics per element.
 __global__ void dw_bwd_weight_reduce_kernel(const float* __restrict__ part,
                                             float* __restrict__ dw, long long kkc,
-                                            int chunks) {
+                                            int chunks, int z_per_blk) {
   const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= kkc) return;
+  const int z0 = blockIdx.y * z_per_blk;
+  const int z1 = min(z0 + z_per_blk, chunks);
   float s = 0.f;
-  for (int z = 0; z < chunks; ++z) s += part[(long long)z * kkc + i];
-  dw[i] = s;
+  for (int z = z0; z < z1; ++z) s += part[(long long)z * kkc + i];
+  if (gridDim.y == 1)
+    dw[i] = s;
+  else
+    atomicAdd(dw + i, s);
 }
 
 // ---------------------------------------------------------------------------
@@ -495,8 +502,14 @@ void launch_bwd_weight(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw
   dw_bwd_weight_kernel<T, K, VEC><<<grid, 256, lds, stream>>>(
       (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (float*)part.data_ptr(), g.N, g.C,
       g.H, g.W, g.Ho, g.Wo, g.sh, g.sw, g.ph, g.pw, log2_cpb, rows_per_chunk);
-  dw_bwd_weight_reduce_kernel<<<dfd_grid(kkc, 256, 1 << 20), 256, 0, stream>>>(
-      (const float*)part.data_ptr(), (float*)dw.data_ptr(), kkc, (int)chunks);
+  const int gx = dfd_grid(kkc, 256, 1 << 20);
+  int zsplit = (int)std::min<long long>((2048 + gx - 1) / gx, (chunks + 7) / 8);
+  if (zsplit < 1) zsplit = 1;
+  const int z_per_blk = (int)((chunks + zsplit - 1) / zsplit);
+  zsplit = (int)((chunks + z_per_blk - 1) / z_per_blk);
+  if (zsplit > 1) dw.zero_();
+  dw_bwd_weight_reduce_kernel<<<dim3(gx, zsplit), 256, 0, stream>>>(
+      (const float*)part.data_ptr(), (float*)dw.data_ptr(), kkc, (int)chunks, z_per_blk);
 }
 
 template <typename T>
