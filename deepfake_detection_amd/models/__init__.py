@@ -25,6 +25,14 @@ from . import dpn  # noqa: F401
 from . import res2net  # noqa: F401
 from . import sknet  # noqa: F401
 from . import inception_v4  # noqa: F401
+from . import gluon_resnet  # noqa: F401
+from . import selecsls  # noqa: F401
+from . import inception_resnet_v2  # noqa: F401
+from . import gluon_xception  # noqa: F401
+from . import dla  # noqa: F401
+from . import hrnet  # noqa: F401
+# nasnet imported below when available
+# pnasnet imported below when available
 from .factory import (  # noqa: F401
     create_model,
     create_deepfake_model,

@@ -512,3 +512,179 @@ for _name, _cm, _dm in [
     ("tf_efficientnet_b8", 2.2, 3.6),
 ]:
     register_model(_tf(_name, _cm, _dm))
+
+
+# ---------------------------------------------------------------------------
+# AdvProp (_ap) and NoisyStudent (_ns) tf variants — same arch, different
+# pretrained weights / input res (reference efficientnet.py:1270-1500)
+# ---------------------------------------------------------------------------
+for _name, _cm, _dm, _res, _cp in [
+    ("tf_efficientnet_b0_ap", 1.0, 1.0, 224, 0.875), ("tf_efficientnet_b1_ap", 1.0, 1.1, 240, 0.882),
+    ("tf_efficientnet_b2_ap", 1.1, 1.2, 260, 0.890), ("tf_efficientnet_b3_ap", 1.2, 1.4, 300, 0.904),
+    ("tf_efficientnet_b4_ap", 1.4, 1.8, 380, 0.922), ("tf_efficientnet_b5_ap", 1.6, 2.2, 456, 0.934),
+    ("tf_efficientnet_b6_ap", 1.8, 2.6, 528, 0.942), ("tf_efficientnet_b7_ap", 2.0, 3.1, 600, 0.949),
+    ("tf_efficientnet_b8_ap", 2.2, 3.6, 672, 0.954),
+    ("tf_efficientnet_b0_ns", 1.0, 1.0, 224, 0.875), ("tf_efficientnet_b1_ns", 1.0, 1.1, 240, 0.882),
+    ("tf_efficientnet_b2_ns", 1.1, 1.2, 260, 0.890), ("tf_efficientnet_b3_ns", 1.2, 1.4, 300, 0.904),
+    ("tf_efficientnet_b4_ns", 1.4, 1.8, 380, 0.922), ("tf_efficientnet_b5_ns", 1.6, 2.2, 456, 0.934),
+    ("tf_efficientnet_b6_ns", 1.8, 2.6, 528, 0.942), ("tf_efficientnet_b7_ns", 2.0, 3.1, 600, 0.949),
+    ("tf_efficientnet_l2_ns_475", 4.3, 5.3, 475, 0.936), ("tf_efficientnet_l2_ns", 4.3, 5.3, 800, 0.961),
+]:
+    _p = _res // 32
+    default_cfgs[_name] = _cfg(input_size=(3, _res, _res), pool_size=(_p, _p), crop_pct=_cp)
+    register_model(_tf(_name, _cm, _dm))
+
+
+def _tf_edge(variant, cm, dm):
+    def fn(pretrained=False, **kwargs):
+        kwargs["bn_eps"] = kwargs.get("bn_eps", 1e-3)
+        kwargs["pad_type"] = kwargs.get("pad_type", "same")
+        return _gen_efficientnet_edge(variant, cm, dm, pretrained, **kwargs)
+
+    fn.__name__ = variant
+    return fn
+
+
+def _tf_cc(variant, cm, dm, ei):
+    def fn(pretrained=False, **kwargs):
+        kwargs["bn_eps"] = kwargs.get("bn_eps", 1e-3)
+        kwargs["pad_type"] = kwargs.get("pad_type", "same")
+        return _gen_efficientnet_condconv(variant, cm, dm, ei, pretrained, **kwargs)
+
+    fn.__name__ = variant
+    return fn
+
+
+for _name, _cm, _dm in [("tf_efficientnet_es", 1.0, 1.0), ("tf_efficientnet_em", 1.0, 1.1),
+                        ("tf_efficientnet_el", 1.2, 1.4)]:
+    _res = {"tf_efficientnet_es": 224, "tf_efficientnet_em": 240, "tf_efficientnet_el": 300}[_name]
+    default_cfgs[_name] = _cfg(input_size=(3, _res, _res), pool_size=(_res // 32, _res // 32))
+    register_model(_tf_edge(_name, _cm, _dm))
+
+for _name, _cm, _dm, _ei in [("tf_efficientnet_cc_b0_4e", 1.0, 1.0, 1),
+                             ("tf_efficientnet_cc_b0_8e", 1.0, 1.0, 2),
+                             ("tf_efficientnet_cc_b1_8e", 1.0, 1.1, 2)]:
+    _res = 240 if _name.endswith("b1_8e") else 224
+    default_cfgs[_name] = _cfg(input_size=(3, _res, _res), pool_size=(_res // 32, _res // 32))
+    register_model(_tf_cc(_name, _cm, _dm, _ei))
+
+
+default_cfgs["efficientnet_b2a"] = _cfg(input_size=(3, 288, 288), pool_size=(9, 9), crop_pct=1.0)
+default_cfgs["efficientnet_b3a"] = _cfg(input_size=(3, 320, 320), pool_size=(10, 10), crop_pct=1.0)
+default_cfgs["efficientnet_b7_deepfake"] = _cfg(input_size=(3, 600, 600), pool_size=(19, 19), crop_pct=0.949)
+
+
+@register_model
+def efficientnet_b2a(pretrained=False, **kwargs):
+    """B2 arch at 288x288, crop_pct 1.0 (reference efficientnet.py:1106-1111)."""
+    return _gen_efficientnet("efficientnet_b2a", 1.1, 1.2, pretrained, **kwargs)
+
+
+@register_model
+def efficientnet_b3a(pretrained=False, **kwargs):
+    """B3 arch at 320x320, crop_pct 1.0 (reference efficientnet.py:1124-1129)."""
+    return _gen_efficientnet("efficientnet_b3a", 1.2, 1.4, pretrained, **kwargs)
+
+
+@register_model
+def efficientnet_b7_deepfake(pretrained=False, **kwargs):
+    """Stock B7 (full 2560 head) under the deepfake name
+    (reference efficientnet.py:1166-1174)."""
+    return _gen_efficientnet("efficientnet_b7_deepfake", 2.0, 3.1, pretrained, **kwargs)
+
+
+# ---------------------------------------------------------------------------
+# MixNet (MixedConv kernels; reference efficientnet.py:913-971,1602-1692)
+# ---------------------------------------------------------------------------
+
+def _gen_mixnet_s(variant, channel_multiplier=1.0, pretrained=False, **kwargs):
+    arch_def = [
+        ["ds_r1_k3_s1_e1_c16"],
+        ["ir_r1_k3_a1.1_p1.1_s2_e6_c24", "ir_r1_k3_a1.1_p1.1_s1_e3_c24"],
+        ["ir_r1_k3.5.7_s2_e6_c40_se0.5_nsw", "ir_r3_k3.5_a1.1_p1.1_s1_e6_c40_se0.5_nsw"],
+        ["ir_r1_k3.5.7_p1.1_s2_e6_c80_se0.25_nsw", "ir_r2_k3.5_p1.1_s1_e6_c80_se0.25_nsw"],
+        ["ir_r1_k3.5.7_a1.1_p1.1_s1_e6_c120_se0.5_nsw", "ir_r2_k3.5.7.9_a1.1_p1.1_s1_e3_c120_se0.5_nsw"],
+        ["ir_r1_k3.5.7.9.11_s2_e6_c200_se0.5_nsw", "ir_r2_k3.5.7.9_p1.1_s1_e6_c200_se0.5_nsw"],
+    ]
+    model_kwargs = dict(
+        block_args=decode_arch_def(arch_def),
+        num_features=1536,
+        stem_size=16,
+        channel_multiplier=channel_multiplier,
+        act_layer=nn.ReLU,
+        norm_kwargs=resolve_bn_args(kwargs),
+        **kwargs,
+    )
+    return _create_model(model_kwargs, default_cfgs[variant], pretrained)
+
+
+def _gen_mixnet_m(variant, channel_multiplier=1.0, depth_multiplier=1.0, pretrained=False, **kwargs):
+    arch_def = [
+        ["ds_r1_k3_s1_e1_c24"],
+        ["ir_r1_k3.5.7_a1.1_p1.1_s2_e6_c32", "ir_r1_k3_a1.1_p1.1_s1_e3_c32"],
+        ["ir_r1_k3.5.7.9_s2_e6_c40_se0.5_nsw", "ir_r3_k3.5_a1.1_p1.1_s1_e6_c40_se0.5_nsw"],
+        ["ir_r1_k3.5.7_s2_e6_c80_se0.25_nsw", "ir_r3_k3.5.7.9_a1.1_p1.1_s1_e6_c80_se0.25_nsw"],
+        ["ir_r1_k3_s1_e6_c120_se0.5_nsw", "ir_r3_k3.5.7.9_a1.1_p1.1_s1_e3_c120_se0.5_nsw"],
+        ["ir_r1_k3.5.7.9_s2_e6_c200_se0.5_nsw", "ir_r3_k3.5.7.9_p1.1_s1_e6_c200_se0.5_nsw"],
+    ]
+    model_kwargs = dict(
+        block_args=decode_arch_def(arch_def, depth_multiplier, depth_trunc="round"),
+        num_features=1536,
+        stem_size=24,
+        channel_multiplier=channel_multiplier,
+        act_layer=nn.ReLU,
+        norm_kwargs=resolve_bn_args(kwargs),
+        **kwargs,
+    )
+    return _create_model(model_kwargs, default_cfgs[variant], pretrained)
+
+
+for _n in ["mixnet_s", "mixnet_m", "mixnet_l", "mixnet_xl", "mixnet_xxl",
+           "tf_mixnet_s", "tf_mixnet_m", "tf_mixnet_l"]:
+    default_cfgs[_n] = _cfg()
+
+
+@register_model
+def mixnet_s(pretrained=False, **kwargs):
+    return _gen_mixnet_s("mixnet_s", 1.0, pretrained, **kwargs)
+
+
+@register_model
+def mixnet_m(pretrained=False, **kwargs):
+    return _gen_mixnet_m("mixnet_m", 1.0, 1.0, pretrained, **kwargs)
+
+
+@register_model
+def mixnet_l(pretrained=False, **kwargs):
+    return _gen_mixnet_m("mixnet_l", 1.3, 1.0, pretrained, **kwargs)
+
+
+@register_model
+def mixnet_xl(pretrained=False, **kwargs):
+    return _gen_mixnet_m("mixnet_xl", 1.6, 1.2, pretrained, **kwargs)
+
+
+@register_model
+def mixnet_xxl(pretrained=False, **kwargs):
+    return _gen_mixnet_m("mixnet_xxl", 2.4, 1.3, pretrained, **kwargs)
+
+
+@register_model
+def tf_mixnet_s(pretrained=False, **kwargs):
+    kwargs.setdefault("bn_eps", 1e-3)
+    kwargs.setdefault("pad_type", "same")
+    return _gen_mixnet_s("tf_mixnet_s", 1.0, pretrained, **kwargs)
+
+
+@register_model
+def tf_mixnet_m(pretrained=False, **kwargs):
+    kwargs.setdefault("bn_eps", 1e-3)
+    kwargs.setdefault("pad_type", "same")
+    return _gen_mixnet_m("tf_mixnet_m", 1.0, 1.0, pretrained, **kwargs)
+
+
+@register_model
+def tf_mixnet_l(pretrained=False, **kwargs):
+    kwargs.setdefault("bn_eps", 1e-3)
+    kwargs.setdefault("pad_type", "same")
+    return _gen_mixnet_m("tf_mixnet_l", 1.3, 1.0, pretrained, **kwargs)

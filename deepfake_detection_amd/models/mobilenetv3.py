@@ -119,6 +119,41 @@ def _create_model(model_kwargs, default_cfg, pretrained=False):
 
 
 def _gen_mobilenet_v3(variant, channel_multiplier=1.0, pretrained=False, **kwargs):
+    if "minimal" in variant:
+        # "minimal" variants: ReLU everywhere, k3 only, no SE
+        act_layer = nn.ReLU
+        if "small" in variant:
+            num_features = 1024
+            arch_def = [
+                ["ds_r1_k3_s2_e1_c16"],
+                ["ir_r1_k3_s2_e4.5_c24", "ir_r1_k3_s1_e3.67_c24"],
+                ["ir_r1_k3_s2_e4_c40", "ir_r2_k3_s1_e6_c40"],
+                ["ir_r2_k3_s1_e3_c48"],
+                ["ir_r3_k3_s2_e6_c96"],
+                ["cn_r1_k1_s1_c576"],
+            ]
+        else:
+            num_features = 1280
+            arch_def = [
+                ["ds_r1_k3_s1_e1_c16"],
+                ["ir_r1_k3_s2_e4_c24", "ir_r1_k3_s1_e3_c24"],
+                ["ir_r3_k3_s2_e3_c40"],
+                ["ir_r1_k3_s2_e6_c80", "ir_r1_k3_s1_e2.5_c80", "ir_r2_k3_s1_e2.3_c80"],
+                ["ir_r2_k3_s1_e6_c112"],
+                ["ir_r3_k3_s2_e6_c160"],
+                ["cn_r1_k1_s1_c960"],
+            ]
+        model_kwargs = dict(
+            block_args=decode_arch_def(arch_def),
+            num_features=num_features,
+            stem_size=16,
+            channel_multiplier=channel_multiplier,
+            act_layer=act_layer,
+            norm_kwargs=resolve_bn_args(kwargs),
+            head_bias=True,
+            **kwargs,
+        )
+        return _create_model(model_kwargs, default_cfgs[variant], pretrained)
     if "small" in variant:
         num_features = 1024
         act_layer = HardSwish
@@ -300,3 +335,110 @@ def fbnetc_100(pretrained=False, **kwargs):
 @register_model
 def spnasnet_100(pretrained=False, **kwargs):
     return _gen_spnasnet("spnasnet_100", 1.0, pretrained, **kwargs)
+
+
+# ---------------------------------------------------------------------------
+# tf_ MobileNetV3 variants (SAME padding, TF bn eps) and the remaining
+# MNASNet / MobileNetV2 entrypoints hosted by the reference in
+# efficientnet.py (reference efficientnet.py:487-700, mobilenetv3.py:352-430)
+# ---------------------------------------------------------------------------
+
+for _n in ["tf_mobilenetv3_large_075", "tf_mobilenetv3_large_100",
+           "tf_mobilenetv3_large_minimal_100", "tf_mobilenetv3_small_075",
+           "tf_mobilenetv3_small_100", "tf_mobilenetv3_small_minimal_100",
+           "mnasnet_140", "semnasnet_050", "semnasnet_075", "semnasnet_100",
+           "semnasnet_140", "mnasnet_small", "mobilenetv2_100"]:
+    default_cfgs.setdefault(_n, _cfg())
+
+
+def _tf_mv3(variant, cm):
+    def fn(pretrained=False, **kwargs):
+        kwargs.setdefault("bn_eps", 1e-3)
+        kwargs.setdefault("pad_type", "same")
+        return _gen_mobilenet_v3(variant, cm, pretrained, **kwargs)
+
+    fn.__name__ = variant
+    return fn
+
+
+for _name, _cm in [("tf_mobilenetv3_large_075", 0.75), ("tf_mobilenetv3_large_100", 1.0),
+                   ("tf_mobilenetv3_large_minimal_100", 1.0), ("tf_mobilenetv3_small_075", 0.75),
+                   ("tf_mobilenetv3_small_100", 1.0), ("tf_mobilenetv3_small_minimal_100", 1.0)]:
+    register_model(_tf_mv3(_name, _cm))
+
+
+@register_model
+def mnasnet_140(pretrained=False, **kwargs):
+    return _gen_mnasnet_b1("mnasnet_140", 1.4, pretrained, **kwargs)
+
+
+@register_model
+def semnasnet_050(pretrained=False, **kwargs):
+    return _gen_mnasnet_a1("semnasnet_050", 0.5, pretrained, **kwargs)
+
+
+@register_model
+def semnasnet_075(pretrained=False, **kwargs):
+    return _gen_mnasnet_a1("semnasnet_075", 0.75, pretrained, **kwargs)
+
+
+@register_model
+def semnasnet_100(pretrained=False, **kwargs):
+    return _gen_mnasnet_a1("semnasnet_100", 1.0, pretrained, **kwargs)
+
+
+@register_model
+def semnasnet_140(pretrained=False, **kwargs):
+    return _gen_mnasnet_a1("semnasnet_140", 1.4, pretrained, **kwargs)
+
+
+def _gen_mnasnet_small(variant, channel_multiplier=1.0, pretrained=False, **kwargs):
+    arch_def = [
+        ["ds_r1_k3_s1_c8"],
+        ["ir_r1_k3_s2_e3_c16"],
+        ["ir_r2_k3_s2_e6_c16"],
+        ["ir_r4_k5_s2_e6_c32_se0.25"],
+        ["ir_r3_k3_s1_e6_c32_se0.25"],
+        ["ir_r3_k5_s2_e6_c88_se0.25"],
+        ["ir_r1_k3_s1_e6_c144"],
+    ]
+    model_kwargs = dict(
+        block_args=decode_arch_def(arch_def),
+        stem_size=8,
+        channel_multiplier=channel_multiplier,
+        norm_kwargs=resolve_bn_args(kwargs),
+        **kwargs,
+    )
+    return _create_model(model_kwargs, default_cfgs[variant], pretrained)
+
+
+@register_model
+def mnasnet_small(pretrained=False, **kwargs):
+    return _gen_mnasnet_small("mnasnet_small", 1.0, pretrained, **kwargs)
+
+
+def _gen_mobilenet_v2(variant, channel_multiplier=1.0, pretrained=False, **kwargs):
+    arch_def = [
+        ["ds_r1_k3_s1_c16"],
+        ["ir_r2_k3_s2_e6_c24"],
+        ["ir_r3_k3_s2_e6_c32"],
+        ["ir_r4_k3_s2_e6_c64"],
+        ["ir_r3_k3_s1_e6_c96"],
+        ["ir_r3_k3_s2_e6_c160"],
+        ["ir_r1_k3_s1_e6_c320"],
+    ]
+    model_kwargs = dict(
+        block_args=decode_arch_def(arch_def),
+        num_features=1280,
+        stem_size=32,
+        channel_multiplier=channel_multiplier,
+        norm_kwargs=resolve_bn_args(kwargs),
+        act_layer=nn.ReLU6,
+        **kwargs,
+    )
+    return _create_model(model_kwargs, default_cfgs[variant], pretrained)
+
+
+@register_model
+def mobilenetv2_100(pretrained=False, **kwargs):
+    return _gen_mobilenet_v2("mobilenetv2_100", 1.0, pretrained, **kwargs)

@@ -10,6 +10,7 @@ import torch.nn.functional as F
 
 from ..ops import functional as O
 from .layers import SelectAdaptivePool2d
+from .layers_extra import EcaModule
 from .registry import register_model
 
 __all__ = ["ResNet", "BasicBlock", "Bottleneck"]
@@ -58,7 +59,7 @@ class BasicBlock(nn.Module):
     expansion = 1
 
     def __init__(self, inplanes, planes, stride=1, downsample=None, cardinality=1,
-                 base_width=64, use_se=False, reduce_first=1, dilation=1,
+                 base_width=64, use_se=False, use_eca=False, reduce_first=1, dilation=1,
                  first_dilation=None, act_layer=nn.ReLU, norm_layer=nn.BatchNorm2d):
         super().__init__()
         assert cardinality == 1 and base_width == 64
@@ -75,7 +76,8 @@ class BasicBlock(nn.Module):
             first_planes, outplanes, kernel_size=3, padding=dilation,
             dilation=dilation, bias=False)
         self.bn2 = norm_layer(outplanes)
-        self.se = SEModule(outplanes, planes // 4) if use_se else None
+        self.se = SEModule(outplanes, planes // 4) if use_se else (
+            EcaModule(outplanes) if use_eca else None)
         self.act2 = act_layer(inplace=True)
         self.downsample = downsample
         self.stride = stride
@@ -101,7 +103,7 @@ class Bottleneck(nn.Module):
     expansion = 4
 
     def __init__(self, inplanes, planes, stride=1, downsample=None, cardinality=1,
-                 base_width=64, use_se=False, reduce_first=1, dilation=1,
+                 base_width=64, use_se=False, use_eca=False, reduce_first=1, dilation=1,
                  first_dilation=None, act_layer=nn.ReLU, norm_layer=nn.BatchNorm2d):
         super().__init__()
         width = int(math.floor(planes * (base_width / 64)) * cardinality)
@@ -119,7 +121,8 @@ class Bottleneck(nn.Module):
         self.act2 = act_layer(inplace=True)
         self.conv3 = nn.Conv2d(width, outplanes, kernel_size=1, bias=False)
         self.bn3 = norm_layer(outplanes)
-        self.se = SEModule(outplanes, planes // 4) if use_se else None
+        self.se = SEModule(outplanes, planes // 4) if use_se else (
+            EcaModule(outplanes) if use_eca else None)
         self.act3 = act_layer(inplace=True)
         self.downsample = downsample
         self.stride = stride
@@ -149,11 +152,13 @@ class ResNet(nn.Module):
     stem width, avg-pool downsample)."""
 
     def __init__(self, block, layers, num_classes=1000, in_chans=3, use_se=False,
+                 use_eca=False, stem_type="",
                  cardinality=1, base_width=64, stem_width=64, deep_stem=False,
                  block_reduce_first=1, down_kernel_size=1, avg_down=False,
                  output_stride=32, act_layer=nn.ReLU, norm_layer=nn.BatchNorm2d,
                  drop_rate=0.0, global_pool="avg"):
         super().__init__()
+        deep_stem = deep_stem or stem_type.startswith("deep")
         self.num_classes = num_classes
         self.inplanes = stem_width * 2 if deep_stem else 64
         self.cardinality = cardinality
@@ -162,14 +167,22 @@ class ResNet(nn.Module):
         self.expansion = block.expansion
 
         if deep_stem:
+            # tiered stems ramp the 3x3x3 stem widths (reference resnet.py
+            # seresnext26t/tn variants): tiered (3c/4, c) narrow (c/2, c/2)
+            if stem_type == "deep_tiered":
+                stem_chs = (3 * stem_width // 4, stem_width)
+            elif stem_type == "deep_tiered_narrow":
+                stem_chs = (stem_width // 2, stem_width // 2)
+            else:
+                stem_chs = (stem_width, stem_width)
             self.conv1 = nn.Sequential(
-                nn.Conv2d(in_chans, stem_width, 3, stride=2, padding=1, bias=False),
-                norm_layer(stem_width),
+                nn.Conv2d(in_chans, stem_chs[0], 3, stride=2, padding=1, bias=False),
+                norm_layer(stem_chs[0]),
                 act_layer(inplace=True),
-                nn.Conv2d(stem_width, stem_width, 3, stride=1, padding=1, bias=False),
-                norm_layer(stem_width),
+                nn.Conv2d(stem_chs[0], stem_chs[1], 3, stride=1, padding=1, bias=False),
+                norm_layer(stem_chs[1]),
                 act_layer(inplace=True),
-                nn.Conv2d(stem_width, self.inplanes, 3, stride=1, padding=1, bias=False))
+                nn.Conv2d(stem_chs[1], self.inplanes, 3, stride=1, padding=1, bias=False))
         else:
             self.conv1 = nn.Conv2d(in_chans, self.inplanes, kernel_size=7, stride=2,
                                    padding=3, bias=False)
@@ -184,7 +197,7 @@ class ResNet(nn.Module):
         else:
             strides, dilations = [1, 2, 1, 1], [1, 1, 2, 4]
 
-        la = dict(use_se=use_se, reduce_first=block_reduce_first,
+        la = dict(use_se=use_se, use_eca=use_eca, reduce_first=block_reduce_first,
                   act_layer=act_layer, norm_layer=norm_layer,
                   avg_down=avg_down, down_kernel_size=down_kernel_size)
         self.layer1 = self._make_layer(block, 64, layers[0], strides[0], dilations[0], **la)
@@ -204,6 +217,7 @@ class ResNet(nn.Module):
                 nn.init.constant_(m.bias, 0.0)
 
     def _make_layer(self, block, planes, blocks, stride=1, dilation=1, use_se=False,
+                    use_eca=False,
                     reduce_first=1, act_layer=nn.ReLU, norm_layer=nn.BatchNorm2d,
                     avg_down=False, down_kernel_size=1):
         downsample = None
@@ -228,13 +242,13 @@ class ResNet(nn.Module):
         first_dilation = 1 if dilation in (1, 2) else 2
         layers = [block(
             self.inplanes, planes, stride, downsample, cardinality=self.cardinality,
-            base_width=self.base_width, use_se=use_se, reduce_first=reduce_first,
+            base_width=self.base_width, use_se=use_se, use_eca=use_eca, reduce_first=reduce_first,
             dilation=dilation, first_dilation=first_dilation, act_layer=act_layer,
             norm_layer=norm_layer)]
         self.inplanes = planes * block.expansion
         for _ in range(1, blocks):
             layers.append(block(
-                self.inplanes, planes, use_se=use_se, reduce_first=reduce_first,
+                self.inplanes, planes, use_se=use_se, use_eca=use_eca, reduce_first=reduce_first,
                 cardinality=self.cardinality, base_width=self.base_width,
                 dilation=dilation, act_layer=act_layer, norm_layer=norm_layer))
         return nn.Sequential(*layers)
@@ -372,3 +386,160 @@ def seresnext26d_32x4d(pretrained=False, **kwargs):
     return _make("seresnext26d_32x4d", Bottleneck, [2, 2, 2, 2], pretrained,
                  cardinality=32, base_width=4, use_se=True, stem_width=32,
                  deep_stem=True, avg_down=True, **kwargs)
+
+
+# ---------------------------------------------------------------------------
+# Weight-variant aliases (torchvision / Instagram / semi- and
+# semi-weakly-supervised pretrain sets) and ECA / tiered-stem variants
+# (reference resnet.py:480-1024) — same architectures, distinct names and
+# pretrained cfgs.
+# ---------------------------------------------------------------------------
+
+for _n in ["tv_resnet34", "tv_resnet50", "tv_resnext50_32x4d",
+           "ig_resnext101_32x8d", "ig_resnext101_32x16d", "ig_resnext101_32x32d",
+           "ig_resnext101_32x48d", "ssl_resnet18", "ssl_resnet50",
+           "ssl_resnext50_32x4d", "ssl_resnext101_32x4d", "ssl_resnext101_32x8d",
+           "ssl_resnext101_32x16d", "swsl_resnet18", "swsl_resnet50",
+           "swsl_resnext50_32x4d", "swsl_resnext101_32x4d", "swsl_resnext101_32x8d",
+           "swsl_resnext101_32x16d", "seresnext26t_32x4d", "seresnext26tn_32x4d",
+           "ecaresnext26tn_32x4d", "ecaresnet18", "ecaresnet50"]:
+    default_cfgs.setdefault(_n, _cfg())
+
+
+@register_model
+def tv_resnet34(pretrained=False, **kwargs):
+    return _make("tv_resnet34", BasicBlock, [3, 4, 6, 3], pretrained, **kwargs)
+
+
+@register_model
+def tv_resnet50(pretrained=False, **kwargs):
+    return _make("tv_resnet50", Bottleneck, [3, 4, 6, 3], pretrained, **kwargs)
+
+
+@register_model
+def tv_resnext50_32x4d(pretrained=False, **kwargs):
+    return _make("tv_resnext50_32x4d", Bottleneck, [3, 4, 6, 3], pretrained,
+                 cardinality=32, base_width=4, **kwargs)
+
+
+def _resnext101(variant, width, pretrained=False, **kwargs):
+    return _make(variant, Bottleneck, [3, 4, 23, 3], pretrained,
+                 cardinality=32, base_width=width, **kwargs)
+
+
+@register_model
+def ig_resnext101_32x8d(pretrained=False, **kwargs):
+    return _resnext101("ig_resnext101_32x8d", 8, pretrained, **kwargs)
+
+
+@register_model
+def ig_resnext101_32x16d(pretrained=False, **kwargs):
+    return _resnext101("ig_resnext101_32x16d", 16, pretrained, **kwargs)
+
+
+@register_model
+def ig_resnext101_32x32d(pretrained=False, **kwargs):
+    return _resnext101("ig_resnext101_32x32d", 32, pretrained, **kwargs)
+
+
+@register_model
+def ig_resnext101_32x48d(pretrained=False, **kwargs):
+    return _resnext101("ig_resnext101_32x48d", 48, pretrained, **kwargs)
+
+
+@register_model
+def ssl_resnet18(pretrained=False, **kwargs):
+    return _make("ssl_resnet18", BasicBlock, [2, 2, 2, 2], pretrained, **kwargs)
+
+
+@register_model
+def ssl_resnet50(pretrained=False, **kwargs):
+    return _make("ssl_resnet50", Bottleneck, [3, 4, 6, 3], pretrained, **kwargs)
+
+
+@register_model
+def ssl_resnext50_32x4d(pretrained=False, **kwargs):
+    return _make("ssl_resnext50_32x4d", Bottleneck, [3, 4, 6, 3], pretrained,
+                 cardinality=32, base_width=4, **kwargs)
+
+
+@register_model
+def ssl_resnext101_32x4d(pretrained=False, **kwargs):
+    return _resnext101("ssl_resnext101_32x4d", 4, pretrained, **kwargs)
+
+
+@register_model
+def ssl_resnext101_32x8d(pretrained=False, **kwargs):
+    return _resnext101("ssl_resnext101_32x8d", 8, pretrained, **kwargs)
+
+
+@register_model
+def ssl_resnext101_32x16d(pretrained=False, **kwargs):
+    return _resnext101("ssl_resnext101_32x16d", 16, pretrained, **kwargs)
+
+
+@register_model
+def swsl_resnet18(pretrained=False, **kwargs):
+    return _make("swsl_resnet18", BasicBlock, [2, 2, 2, 2], pretrained, **kwargs)
+
+
+@register_model
+def swsl_resnet50(pretrained=False, **kwargs):
+    return _make("swsl_resnet50", Bottleneck, [3, 4, 6, 3], pretrained, **kwargs)
+
+
+@register_model
+def swsl_resnext50_32x4d(pretrained=False, **kwargs):
+    return _make("swsl_resnext50_32x4d", Bottleneck, [3, 4, 6, 3], pretrained,
+                 cardinality=32, base_width=4, **kwargs)
+
+
+@register_model
+def swsl_resnext101_32x4d(pretrained=False, **kwargs):
+    return _resnext101("swsl_resnext101_32x4d", 4, pretrained, **kwargs)
+
+
+@register_model
+def swsl_resnext101_32x8d(pretrained=False, **kwargs):
+    return _resnext101("swsl_resnext101_32x8d", 8, pretrained, **kwargs)
+
+
+@register_model
+def swsl_resnext101_32x16d(pretrained=False, **kwargs):
+    return _resnext101("swsl_resnext101_32x16d", 16, pretrained, **kwargs)
+
+
+@register_model
+def seresnext26t_32x4d(pretrained=False, **kwargs):
+    """SE-ResNeXt-26 with deep tiered stem (reference resnet.py)."""
+    return _make("seresnext26t_32x4d", Bottleneck, [2, 2, 2, 2], pretrained,
+                 cardinality=32, base_width=4, use_se=True, stem_width=64,
+                 stem_type="deep_tiered", avg_down=True, **kwargs)
+
+
+@register_model
+def seresnext26tn_32x4d(pretrained=False, **kwargs):
+    """SE-ResNeXt-26 with narrow tiered stem."""
+    return _make("seresnext26tn_32x4d", Bottleneck, [2, 2, 2, 2], pretrained,
+                 cardinality=32, base_width=4, use_se=True, stem_width=64,
+                 stem_type="deep_tiered_narrow", avg_down=True, **kwargs)
+
+
+@register_model
+def ecaresnext26tn_32x4d(pretrained=False, **kwargs):
+    """ECA-ResNeXt-26 with narrow tiered stem (ECA instead of SE)."""
+    return _make("ecaresnext26tn_32x4d", Bottleneck, [2, 2, 2, 2], pretrained,
+                 cardinality=32, base_width=4, use_eca=True, stem_width=64,
+                 stem_type="deep_tiered_narrow", avg_down=True, **kwargs)
+
+
+@register_model
+def ecaresnet18(pretrained=False, **kwargs):
+    return _make("ecaresnet18", BasicBlock, [2, 2, 2, 2], pretrained,
+                 use_eca=True, **kwargs)
+
+
+@register_model
+def ecaresnet50(pretrained=False, **kwargs):
+    return _make("ecaresnet50", Bottleneck, [3, 4, 6, 3], pretrained,
+                 use_eca=True, **kwargs)
