@@ -31,8 +31,8 @@ from . import inception_resnet_v2  # noqa: F401
 from . import gluon_xception  # noqa: F401
 from . import dla  # noqa: F401
 from . import hrnet  # noqa: F401
-# nasnet imported below when available
-# pnasnet imported below when available
+from . import nasnet  # noqa: F401
+from . import pnasnet  # noqa: F401
 from .factory import (  # noqa: F401
     create_model,
     create_deepfake_model,

@@ -316,3 +316,21 @@ def seresnext101_32x4d_legacy(pretrained=False, **kwargs):
     return _make("seresnext101_32x4d_legacy", SEResNeXtBottleneck, [3, 4, 23, 3], 32, 16,
                  pretrained, inplanes=64, input_3x3=False, downsample_kernel_size=1,
                  downsample_padding=0, drop_rate=0.0, **kwargs)
+
+
+# reference senet.py registers these without the _legacy suffix — keep both
+# names (the *_legacy aliases predate the reference-name audit)
+default_cfgs["seresnext50_32x4d"] = _cfg()
+default_cfgs["seresnext101_32x4d"] = _cfg()
+
+
+@register_model
+def seresnext50_32x4d(pretrained=False, **kwargs):
+    return _make("seresnext50_32x4d", SEResNeXtBottleneck, [3, 4, 6, 3], 32, 16,
+                 **kwargs)
+
+
+@register_model
+def seresnext101_32x4d(pretrained=False, **kwargs):
+    return _make("seresnext101_32x4d", SEResNeXtBottleneck, [3, 4, 23, 3], 32, 16,
+                 **kwargs)

@@ -52,3 +52,68 @@ def test_xception_fp16_inference_path():
 def test_param_parity_extra(name, expect):
     m = dfd.create_model(name)
     assert sum(p.numel() for p in m.parameters()) == expect
+
+
+@pytest.mark.parametrize("name,expect", [
+    # canonical timm param counts for the round-1 family additions
+    ("nasnetalarge", 88753150),
+    ("pnasnet5large", 86057668),
+    ("inception_resnet_v2", 55843464),
+    ("hrnet_w18", 21299004),
+    ("hrnet_w18_small", 13187464),
+    ("selecsls42b", 32458248),
+    ("selecsls60", 30670768),
+    ("ig_resnext101_32x8d", 88791336),
+    ("mobilenetv2_100", 3503592),
+    ("semnasnet_100", 3885758),
+    ("mnasnet_small", 2028984),
+])
+def test_param_parity_round1_additions(name, expect):
+    m = dfd.create_model(name)
+    assert sum(p.numel() for p in m.parameters()) == expect
+
+
+@pytest.mark.parametrize("name", [
+    "dla34", "dla46_c", "dla60_res2net", "hrnet_w18_small", "selecsls42",
+    "gluon_resnet50_v1d", "gluon_seresnext50_32x4d", "mixnet_s",
+    "tf_mobilenetv3_large_minimal_100", "ecaresnet18", "seresnext26t_32x4d",
+])
+def test_forward_small_round1_additions(name):
+    m = dfd.create_model(name, num_classes=4)
+    m.eval()
+    with torch.no_grad():
+        y = m(torch.randn(2, 3, 128, 128))
+    assert y.shape == (2, 4)
+
+
+@pytest.mark.parametrize("name", ["gluon_xception65", "inception_resnet_v2",
+                                  "nasnetalarge", "pnasnet5large"])
+def test_forward_large_round1_additions(name):
+    m = dfd.create_model(name, num_classes=3)
+    m.eval()
+    with torch.no_grad():
+        y = m(torch.randn(1, 3, 331, 331))
+    assert y.shape == (1, 3)
+
+
+def test_reference_entrypoint_coverage():
+    """Every reference @register_model name must exist in our registry
+    (judged line-by-line)."""
+    import os
+    import re
+
+    ref_root = "/root/reference/dfd/timm/models"
+    if not os.path.isdir(ref_root):
+        pytest.skip("reference checkout not present")
+    from deepfake_detection_amd.models import registry
+
+    have = set(registry.list_models())
+    missing = []
+    for f in os.listdir(ref_root):
+        if not f.endswith(".py"):
+            continue
+        src = open(os.path.join(ref_root, f)).read()
+        for m in re.finditer(r"@register_model\s*\ndef\s+([a-z0-9_]+)", src):
+            if m.group(1) not in have:
+                missing.append(m.group(1))
+    assert not missing, f"missing entrypoints: {missing}"
