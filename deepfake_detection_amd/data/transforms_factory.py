@@ -157,6 +157,11 @@ def create_transform(
     else:
         img_size = input_size
 
+    if tf_preprocessing and not is_training:
+        from .tf_preprocessing import TfPreprocessTransform
+
+        return TfPreprocessTransform(is_training=False, size=img_size,
+                                     interpolation=interpolation)
     if is_training:
         return transforms_imagenet_train(
             img_size, color_jitter=color_jitter, auto_augment=auto_augment,

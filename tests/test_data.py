@@ -179,3 +179,17 @@ def test_resolve_data_config_v2_string():
     cfg = resolve_data_config({"input_size_v2": "12,600,600", "model": "efficientnet_deepfake_v4"})
     assert cfg["input_size"] == (12, 600, 600)
     assert cfg["mean"] == (0.485, 0.456, 0.406)
+
+
+def test_tf_preprocessing_eval_transform():
+    """TF-style eval path (reference tf_preprocessing.py) without a TF dep."""
+    from PIL import Image
+
+    from deepfake_detection_amd.data.transforms_factory import create_transform
+
+    t = create_transform(224, is_training=False, tf_preprocessing=True)
+    img = Image.fromarray((torch.rand(300, 260, 3) * 255).byte().numpy())
+    out = t(img)
+    assert out.shape == (3, 224, 224)
+    assert out.dtype == torch.float32
+    assert 0.0 <= out.min() and out.max() <= 1.0
