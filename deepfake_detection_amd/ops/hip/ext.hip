@@ -37,6 +37,7 @@ at::Tensor dw_conv2d_bwd_data(at::Tensor dy, at::Tensor w_packed, int64_t H, int
                               int64_t sh, int64_t sw, int64_t ph, int64_t pw);
 at::Tensor dw_conv2d_bwd_weight(at::Tensor dy, at::Tensor x, int64_t K, int64_t sh,
                                 int64_t sw, int64_t ph, int64_t pw);
+at::Tensor pw_conv2d_fwd_mfma(at::Tensor x, at::Tensor w);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "deepfake_detection_amd gfx950 (MI355X/CDNA4) kernels";
@@ -54,4 +55,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dw_conv2d_fwd", &dw_conv2d_fwd, "depthwise conv2d forward (NHWC)");
   m.def("dw_conv2d_bwd_data", &dw_conv2d_bwd_data, "depthwise conv2d bwd data (NHWC)");
   m.def("dw_conv2d_bwd_weight", &dw_conv2d_bwd_weight, "depthwise conv2d bwd weight (NHWC)");
+  m.def("pw_conv2d_fwd_mfma", &pw_conv2d_fwd_mfma, "1x1 conv as MFMA GEMM (experimental)");
 }

@@ -338,3 +338,22 @@ def test_depthwise_module_routes_to_hip():
     ref = torch.nn.functional.conv2d(x, m.weight, m.bias, m.stride, m.padding,
                                      m.dilation, m.groups)
     assert torch.allclose(y, ref, atol=1e-4, rtol=1e-4)
+
+
+# ---------------------------------------------------------------------------
+# experimental MFMA pointwise conv vs torch fp32 conv
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("m,k,n", [(4 * 19 * 19, 160, 960), (2 * 75 * 75, 192, 32),
+                                   (3 * 38 * 38, 288, 48), (129, 24, 17)])
+def test_pwconv_mfma_matches_torch(m, k, n):
+    from deepfake_detection_amd.ops.pwconv import pw_conv2d_fwd
+
+    torch.manual_seed(11)
+    # express M as (B, H, W) = (1, 1, m) — the kernel only sees M = B*H*W
+    x = torch.randn(1, k, 1, m, device="cuda", dtype=torch.bfloat16)
+    x = x.contiguous(memory_format=torch.channels_last)
+    w = torch.randn(n, k, 1, 1, device="cuda", dtype=torch.bfloat16)
+    y = pw_conv2d_fwd(x, w)
+    ref = torch.nn.functional.conv2d(x.float(), w.float())
+    assert y.shape == ref.shape
+    assert torch.allclose(y.float(), ref, atol=0.1 + 0.02 * (k ** 0.5), rtol=0.05)
