@@ -164,7 +164,7 @@ def bench_se(B, dtype):
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--batch", type=int, default=192)
-    p.add_argument("--ops", default="dw,bn,se")
+    p.add_argument("--ops", default="dw,bn,se")  # add "pw" for the MFMA 1x1 A/B
     p.add_argument("--dtype", default="bf16")
     args = p.parse_args()
     dtype = {"bf16": torch.bfloat16, "fp16": torch.float16, "fp32": torch.float32}[args.dtype]
@@ -176,7 +176,34 @@ def main():
         bench_bn(args.batch, dtype)
     if "se" in ops:
         bench_se(args.batch, dtype)
+    if "pw" in ops:
+        bench_pw(args.batch)
 
 
 if __name__ == "__main__":
     main()
+
+
+def bench_pw(B=192):
+    """A/B the experimental MFMA 1x1 GEMM vs MIOpen conv and rocBLAS matmul
+    on the B4-299 pointwise shapes."""
+    from deepfake_detection_amd.ops.pwconv import pw_conv2d_fwd
+    shapes = [  # (Cin, Cout, H)
+        (48, 24, 150), (24, 144, 150), (144, 32, 75), (192, 32, 75),
+        (192, 288, 38), (288, 48, 38), (672, 112, 19), (960, 160, 19),
+        (960, 272, 10), (1632, 272, 10), (1632, 448, 10), (448, 2688, 10),
+        (2688, 448, 10),
+    ]
+    print(f"== pointwise conv fwd (B={B}, bf16) | mfma ms | miopen ms | matmul ms")
+    for k, n, h in shapes:
+        x = cl(torch.randn(B, k, h, h, device="cuda", dtype=torch.bfloat16))
+        w = torch.randn(n, k, 1, 1, device="cuda", dtype=torch.bfloat16)
+        t_mfma = timeit(lambda: pw_conv2d_fwd(x, w))
+        t_mi = timeit(lambda: torch.nn.functional.conv2d(x, w))
+        x2 = x.permute(0, 2, 3, 1).reshape(-1, k)
+        w2 = w.reshape(n, k)
+        t_mm = timeit(lambda: x2 @ w2.t())
+        gb = (x.numel() + B * n * h * h) * 2 / 1e9
+        tf = 2 * x2.shape[0] * k * n / 1e12
+        print(f"pw K={k:4d} N={n:4d} H={h:3d}: mfma {t_mfma:7.3f} miopen {t_mi:7.3f} "
+              f"matmul {t_mm:7.3f} ms | mfma {gb/t_mfma*1000:5.0f} GB/s {tf/t_mfma*1000:6.1f} TF")
