@@ -161,29 +161,6 @@ def bench_se(B, dtype):
     print(f"se total per step: {tot:.2f} ms")
 
 
-def main():
-    p = argparse.ArgumentParser()
-    p.add_argument("--batch", type=int, default=192)
-    p.add_argument("--ops", default="dw,bn,se")  # add "pw" for the MFMA 1x1 A/B
-    p.add_argument("--dtype", default="bf16")
-    args = p.parse_args()
-    dtype = {"bf16": torch.bfloat16, "fp16": torch.float16, "fp32": torch.float32}[args.dtype]
-    assert torch.cuda.is_available()
-    ops = args.ops.split(",")
-    if "dw" in ops:
-        bench_dw(args.batch, dtype)
-    if "bn" in ops:
-        bench_bn(args.batch, dtype)
-    if "se" in ops:
-        bench_se(args.batch, dtype)
-    if "pw" in ops:
-        bench_pw(args.batch)
-
-
-if __name__ == "__main__":
-    main()
-
-
 def bench_pw(B=192):
     """A/B the experimental MFMA 1x1 GEMM vs MIOpen conv and rocBLAS matmul
     on the B4-299 pointwise shapes."""
@@ -207,3 +184,26 @@ def bench_pw(B=192):
         tf = 2 * x2.shape[0] * k * n / 1e12
         print(f"pw K={k:4d} N={n:4d} H={h:3d}: mfma {t_mfma:7.3f} miopen {t_mi:7.3f} "
               f"matmul {t_mm:7.3f} ms | mfma {gb/t_mfma*1000:5.0f} GB/s {tf/t_mfma*1000:6.1f} TF")
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument("--batch", type=int, default=192)
+    p.add_argument("--ops", default="dw,bn,se")  # add "pw" for the MFMA 1x1 A/B
+    p.add_argument("--dtype", default="bf16")
+    args = p.parse_args()
+    dtype = {"bf16": torch.bfloat16, "fp16": torch.float16, "fp32": torch.float32}[args.dtype]
+    assert torch.cuda.is_available()
+    ops = args.ops.split(",")
+    if "dw" in ops:
+        bench_dw(args.batch, dtype)
+    if "bn" in ops:
+        bench_bn(args.batch, dtype)
+    if "se" in ops:
+        bench_se(args.batch, dtype)
+    if "pw" in ops:
+        bench_pw(args.batch)
+
+
+if __name__ == "__main__":
+    main()
