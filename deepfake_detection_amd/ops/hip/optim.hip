@@ -46,12 +46,32 @@ ChunkTable build_chunks(const std::vector<at::Tensor>& a,
     }
   }
   const int nchunks = (int)(rows.size() / 5);
+
+  // Tensor addresses are stable across steps in steady state (params always;
+  // grads under gradient_as_bucket_view or the caching allocator), so cache
+  // the device table keyed on its full contents: skips a host-alloc + H2D
+  // per step and keeps optimizer steps hipGraph-capturable.
+  struct Cache {
+    std::vector<long long> key;
+    at::Tensor dev;
+  };
+  static thread_local std::vector<Cache> cache;
+  for (auto& e : cache) {
+    if (e.key == rows && e.dev.device() == opts.device()) {
+      ChunkTable t;
+      t.dev = e.dev;
+      t.nchunks = nchunks;
+      return t;
+    }
+  }
   auto host = at::from_blob(rows.data(), {(long long)nchunks, 5},
                             at::TensorOptions().dtype(at::kLong))
                   .clone();
   ChunkTable t;
   t.dev = host.to(opts.device(), /*non_blocking=*/true);
   t.nchunks = nchunks;
+  if (cache.size() > 8) cache.clear();  // bound: a few optimizers/EMA per process
+  cache.push_back({std::move(rows), t.dev});
   return t;
 }
 
