@@ -14,16 +14,23 @@ from .dataset import (  # noqa: F401
     Dataset,
     DatasetTar,
     DeepFakeDataset_v1,
+    DeepFakeDataset_v1_bak,
+    DeepFakeDataset_v2,
     DeepFakeDataset_v3,
     SyntheticDeepFakeDataset,
 )
 from .distributed_sampler import OrderedDistributedSampler  # noqa: F401
 from .loader import (  # noqa: F401
     PrefetchLoader,
+    PrefetchLoader_v1,
     PrefetchLoader_v3,
+    create_deepfake_loader,
+    create_deepfake_loader_v1,
+    create_deepfake_loader_v2,
     create_deepfake_loader_v3,
     create_loader,
     fast_collate,
+    fast_collate_v1,
 )
 from .mixup import FastCollateMixup, mixup_batch, mixup_target, one_hot  # noqa: F401
 from .random_erasing import RandomErasing  # noqa: F401

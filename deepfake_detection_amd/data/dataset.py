@@ -282,6 +282,115 @@ class DeepFakeDataset_v1(data.Dataset):
             self.transform_rotateds = transform_rotateds
 
 
+class DeepFakeDataset_v1_bak(data.Dataset):
+    """Paired legacy variant (reference dataset.py:126): each item returns
+    (fake_img, real_img, fake_rotated, real_rotated); real list shuffled at
+    construction so pairs are decorrelated."""
+
+    def __init__(self, root, result_file, load_bytes=False, transform=None,
+                 transform_rotateds=None, class_map=""):
+        self.root = root
+        self.load_bytes = load_bytes
+        self.transform = transform
+        self.transform_rotateds = transform_rotateds
+        self.results_fake = []
+        self.results_real = []
+        with open(result_file, "r") as f:
+            for line in f.readlines():
+                line_s = line.strip().split(":")
+                if len(line_s) != 3:
+                    continue
+                self.results_fake.append((line_s[0], int(line_s[2])))
+                self.results_real.append((line_s[1], int(line_s[2])))
+        if not self.results_fake:
+            raise RuntimeError(f"Found 0 entries in {result_file}")
+        random.shuffle(self.results_real)
+
+    def _load(self, path):
+        return open(path, "rb").read() if self.load_bytes else Image.open(path).convert("RGB")
+
+    def __getitem__(self, index):
+        fake_path, fake_rot = self.results_fake[index]
+        real_path, real_rot = self.results_real[index]
+        fake_img, real_img = self._load(fake_path), self._load(real_path)
+        if self.transform_rotateds is not None:
+            fake_img = self.transform_rotateds[fake_rot](fake_img)
+            real_img = self.transform_rotateds[real_rot](real_img)
+        if self.transform is not None:
+            fake_img = self.transform(fake_img)
+            real_img = self.transform(real_img)
+        return fake_img, real_img, fake_rot, real_rot
+
+    def __len__(self):
+        return len(self.results_fake)
+
+
+def get_all_images_list(list_files):
+    """Read `<name>` lines from per-root list files -> [(name, root_idx)]
+    (reference dataset.py get_all_images_list)."""
+    files = []
+    for root_index, list_file in enumerate(list_files):
+        if not os.path.isfile(list_file):
+            continue
+        with open(list_file, "r") as f:
+            files += [(line.strip(), root_index) for line in f.readlines()]
+    return files
+
+
+class DeepFakeDataset_v2(data.Dataset):
+    """Single-image predecessor of v3 (reference dataset.py:284): per-root
+    real_list/fake_list files, fakes split into len(real) groups with random
+    choice per group; even index = fake (0), odd = real (1); rotation id
+    parsed from the file-name suffix `*_<rot>.jpg`."""
+
+    def __init__(self, roots, class_names, load_bytes=False, transform=None,
+                 transform_rotateds=None, frac=1):
+        if isinstance(roots, str):
+            roots = [roots]
+        self.roots = roots
+        self.class_to_idx = load_class_map(class_names)
+        self.load_bytes = load_bytes
+        self.transform = transform
+        self.transform_rotateds = transform_rotateds
+
+        real_lists = [os.path.join(r, "real_list.txt") for r in roots]
+        fake_lists = [os.path.join(r, "fake_list.txt") for r in roots]
+        self.real_images = get_all_images_list(real_lists)
+        fake_images = get_all_images_list(fake_lists)
+        assert len(fake_images) >= len(self.real_images)
+        assert len(self.real_images) > 0
+        if 0 < frac < 1:
+            random.seed(1024)
+            self.real_images = random.sample(self.real_images, int(len(self.real_images) * frac))
+            fake_images = random.sample(fake_images, int(len(fake_images) * frac))
+        self.fake_images = np.array_split(np.array(fake_images, dtype=object), len(self.real_images))
+
+    def __getitem__(self, index):
+        target = index % 2
+        ti = index // 2
+        if target == 0:
+            img_name, root_index = random.choice(list(self.fake_images[ti]))
+            img_path = os.path.join(self.roots[int(root_index)], "fake", img_name)
+        else:
+            img_name, root_index = self.real_images[ti]
+            img_path = os.path.join(self.roots[int(root_index)], "real", img_name)
+        rotated = int(img_path.split("_")[-1].split(".")[0])
+        img = open(img_path, "rb").read() if self.load_bytes else Image.open(img_path).convert("RGB")
+        if self.transform_rotateds is not None:
+            img = self.transform_rotateds[rotated](img)
+        if self.transform is not None:
+            img = self.transform(img)
+        return img, target
+
+    def __len__(self):
+        return 2 * len(self.real_images)
+
+    def set_transform(self, transform, transform_rotateds=None):
+        self.transform = transform
+        if transform_rotateds is not None:
+            self.transform_rotateds = transform_rotateds
+
+
 class ConcatDataset(data.ConcatDataset):
     """Concatenation with transform pass-through (reference dataset.py:229)."""
 

@@ -171,6 +171,10 @@ class PrefetchLoader(PrefetchLoader_v3):
             img_num=1, channels_last=channels_last)
 
 
+# legacy name: the v1 prefetcher had the same side-stream copy/normalize
+PrefetchLoader_v1 = PrefetchLoader
+
+
 def create_deepfake_loader_v3(
         dataset, input_size, batch_size, is_training=False, use_prefetcher=True,
         re_prob=0.0, re_mode="const", re_count=1, re_split=False, re_max=0.02,
@@ -288,3 +292,91 @@ def create_loader(
             re_prob=re_prob if is_training else 0.0, re_mode=re_mode,
             re_count=re_count, re_num_splits=re_num_splits)
     return loader
+
+
+def fast_collate_v1(batch):
+    """Pair-interleave collate for DeepFakeDataset_v1_bak items
+    (fake_img, real_img, fake_rot, real_rot) -> shuffled (2B, ...) uint8
+    batch with alternating labels (reference loader.py:48-99)."""
+    assert isinstance(batch[0], tuple)
+    batch_size = len(batch)
+    first = batch[0][0]
+    if isinstance(first, np.ndarray):
+        targets = torch.tensor([i % 2 for i in range(2 * batch_size)], dtype=torch.int64)
+        tensor = torch.zeros((batch_size * 2, *first.shape), dtype=torch.uint8)
+        for i in range(batch_size):
+            tensor[2 * i] += torch.from_numpy(batch[i][0])
+            tensor[2 * i + 1] += torch.from_numpy(batch[i][1])
+        perm = torch.randperm(tensor.size(0))
+        return tensor[perm], targets[perm]
+    if isinstance(first, torch.Tensor):
+        targets = torch.tensor([b[1] for b in batch], dtype=torch.int64)
+        tensor = torch.zeros((batch_size, *first.shape), dtype=torch.uint8)
+        for i in range(batch_size):
+            tensor[i].copy_(batch[i][0])
+        return tensor, targets
+    raise TypeError(f"fast_collate_v1: unsupported item type {type(first)}")
+
+
+def _create_deepfake_loader_generic(
+        dataset, input_size, batch_size, is_training=False, use_prefetcher=True,
+        re_prob=0.0, re_mode="const", re_count=1, color_jitter=0.4,
+        interpolation="bilinear", mean=IMAGENET_DEFAULT_MEAN,
+        std=IMAGENET_DEFAULT_STD, num_workers=1, distributed=False,
+        crop_pct=None, collate_fn=None, pin_memory=True, fp16=False, dtype=None):
+    """Shared body of the legacy single-image loader factories
+    (reference loader.py:457,543,633): imagenet-style transform +
+    fast_collate + single-image PrefetchLoader."""
+    from .transforms_factory import transforms_imagenet_eval, transforms_imagenet_train
+
+    img_size = input_size[-1] if isinstance(input_size, (tuple, list)) else input_size
+    if is_training:
+        transform = transforms_imagenet_train(
+            img_size, color_jitter=color_jitter, interpolation=interpolation,
+            use_prefetcher=use_prefetcher, mean=mean, std=std)
+    else:
+        transform = transforms_imagenet_eval(
+            img_size, interpolation=interpolation, use_prefetcher=use_prefetcher,
+            mean=mean, std=std, crop_pct=crop_pct)
+    if hasattr(dataset, "set_transform"):
+        dataset.set_transform(transform)
+    else:
+        dataset.transform = transform
+
+    sampler = None
+    if distributed:
+        if is_training:
+            sampler = torch.utils.data.distributed.DistributedSampler(dataset)
+        else:
+            sampler = OrderedDistributedSampler(dataset)
+    if collate_fn is None:
+        collate_fn = fast_collate if use_prefetcher else torch.utils.data.dataloader.default_collate
+
+    loader = torch.utils.data.DataLoader(
+        dataset, batch_size=batch_size, shuffle=sampler is None and is_training,
+        num_workers=num_workers, sampler=sampler, collate_fn=collate_fn,
+        pin_memory=pin_memory, drop_last=is_training)
+    if use_prefetcher:
+        loader = PrefetchLoader(
+            loader, mean=mean, std=std, fp16=fp16, dtype=dtype,
+            re_prob=re_prob if is_training else 0.0, re_mode=re_mode,
+            re_count=re_count, img_num=1)
+    return loader
+
+
+def create_deepfake_loader(dataset, input_size, batch_size, **kwargs):
+    """Legacy v0 factory (reference loader.py:457)."""
+    return _create_deepfake_loader_generic(dataset, input_size, batch_size, **kwargs)
+
+
+def create_deepfake_loader_v1(dataset, input_size, batch_size, **kwargs):
+    """Legacy paired factory (reference loader.py:543): fast_collate_v1
+    deinterleaves (fake, real) pairs into a shuffled 2B batch."""
+    kwargs.setdefault("collate_fn", fast_collate_v1)
+    return _create_deepfake_loader_generic(dataset, input_size, batch_size, **kwargs)
+
+
+def create_deepfake_loader_v2(dataset, input_size, batch_size, **kwargs):
+    """Legacy v2 factory (reference loader.py:633): single-image dataset with
+    per-root list files (DeepFakeDataset_v2)."""
+    return _create_deepfake_loader_generic(dataset, input_size, batch_size, **kwargs)

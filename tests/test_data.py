@@ -193,3 +193,34 @@ def test_tf_preprocessing_eval_transform():
     assert out.shape == (3, 224, 224)
     assert out.dtype == torch.float32
     assert 0.0 <= out.min() and out.max() <= 1.0
+
+
+def test_legacy_dataset_and_loader_variants(tmp_path):
+    """Legacy v1_bak/v2 datasets + v1 collate (reference dataset.py:126,284,
+    loader.py:48-99) exist and roundtrip."""
+    import numpy as np
+
+    import deepfake_detection_amd.data as D
+
+    # v1_bak: pair file fake:real:rotated
+    img = tmp_path / "i.jpg"
+    from PIL import Image as PILImage
+
+    PILImage.fromarray(np.zeros((8, 8, 3), dtype=np.uint8)).save(img)
+    pair_file = tmp_path / "pairs.txt"
+    pair_file.write_text(f"{img}:{img}:0\n" * 3)
+    ds = D.DeepFakeDataset_v1_bak(str(tmp_path), str(pair_file))
+    fake, real, fr, rr = ds[0]
+    assert len(ds) == 3 and fr == 0 and rr == 0
+
+    # v2: per-root list files; names carry the _<rot>.jpg suffix
+    for sub in ("real", "fake"):
+        (tmp_path / sub).mkdir()
+        PILImage.fromarray(np.zeros((8, 8, 3), dtype=np.uint8)).save(tmp_path / sub / "a_0.jpg")
+    (tmp_path / "real_list.txt").write_text("a_0.jpg\n")
+    (tmp_path / "fake_list.txt").write_text("a_0.jpg\n")
+    ds2 = D.DeepFakeDataset_v2(str(tmp_path), "fake,real")
+    assert len(ds2) == 2
+    _, t0 = ds2[0]
+    _, t1 = ds2[1]
+    assert (t0, t1) == (0, 1)
