@@ -156,10 +156,11 @@ class ConvBnAct(nn.Module):
         return self.act1(self.bn1(x))
 
 
-def _bn_act(bn, act_module, act_name, x):
+def _bn_act(bn, act_module, act_name, x, residual=None):
     if act_name != "other" and isinstance(bn, nn.BatchNorm2d):
-        return O.bn_act(x, bn, act_name)
-    return act_module(bn(x))
+        return O.bn_act(x, bn, act_name, residual)
+    y = act_module(bn(x))
+    return y if residual is None else y + residual
 
 
 class DepthwiseSeparableConv(nn.Module):
@@ -208,11 +209,15 @@ class DepthwiseSeparableConv(nn.Module):
         if self.se is not None:
             x = self.se(x)
         x = self.conv_pw(x)
-        x = _bn_act(self.bn2, self.act2, self._act2_name, x)
-        if self.has_residual:
-            if self.drop_path_rate > 0.0:
-                x = drop_path(x, self.drop_path_rate, self.training)
-            x = x + residual
+        if (self.has_residual and self._act2_name != "other"
+                and not (self.drop_path_rate > 0.0 and self.training) and x.is_cuda):
+            x = _bn_act(self.bn2, self.act2, self._act2_name, x, residual)
+        else:
+            x = _bn_act(self.bn2, self.act2, self._act2_name, x)
+            if self.has_residual:
+                if self.drop_path_rate > 0.0:
+                    x = drop_path(x, self.drop_path_rate, self.training)
+                x = x + residual
         return x
 
 
@@ -272,11 +277,16 @@ class InvertedResidual(nn.Module):
         if self.se is not None:
             x = self.se(x)
         x = self.conv_pwl(x)
-        x = _bn_act(self.bn3, nn.Identity(), "none", x)
-        if self.has_residual:
-            if self.drop_path_rate > 0.0:
-                x = drop_path(x, self.drop_path_rate, self.training)
-            x = x + residual
+        if (self.has_residual and not (self.drop_path_rate > 0.0 and self.training)
+                and x.is_cuda):
+            # fused BN + residual add (one pass; drop_path is identity here)
+            x = _bn_act(self.bn3, nn.Identity(), "none", x, residual)
+        else:
+            x = _bn_act(self.bn3, nn.Identity(), "none", x)
+            if self.has_residual:
+                if self.drop_path_rate > 0.0:
+                    x = drop_path(x, self.drop_path_rate, self.training)
+                x = x + residual
         return x
 
 
@@ -311,11 +321,16 @@ class CondConvResidual(InvertedResidual):
         if self.se is not None:
             x = self.se(x)
         x = self.conv_pwl(x, routing_weights)
-        x = _bn_act(self.bn3, nn.Identity(), "none", x)
-        if self.has_residual:
-            if self.drop_path_rate > 0.0:
-                x = drop_path(x, self.drop_path_rate, self.training)
-            x = x + residual
+        if (self.has_residual and not (self.drop_path_rate > 0.0 and self.training)
+                and x.is_cuda):
+            # fused BN + residual add (one pass; drop_path is identity here)
+            x = _bn_act(self.bn3, nn.Identity(), "none", x, residual)
+        else:
+            x = _bn_act(self.bn3, nn.Identity(), "none", x)
+            if self.has_residual:
+                if self.drop_path_rate > 0.0:
+                    x = drop_path(x, self.drop_path_rate, self.training)
+                x = x + residual
         return x
 
 

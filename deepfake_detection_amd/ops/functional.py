@@ -29,24 +29,27 @@ def act_name_of(module) -> str:
     return "other"
 
 
-def bn_act(x, bn, act: str = "silu"):
-    """BatchNorm2d + activation, fused on GPU (HIP kernel, NHWC, bf16 I/O,
-    fp32 stats — SURVEY.md §2.6 item 5), torch ops on CPU."""
+def bn_act(x, bn, act: str = "silu", residual=None):
+    """BatchNorm2d + activation (+ optional fused residual add), fused on GPU
+    (HIP kernel, NHWC, bf16 I/O, fp32 stats — SURVEY.md §2.6 items 5 and 7),
+    torch ops on CPU."""
     if x.is_cuda and gpu_ops_required() and act in _FUSED_ACTS:
         from .bn_act import fused_bn_act
 
         return fused_bn_act(
             x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
-            bn.training, bn.momentum, bn.eps, act,
+            bn.training, bn.momentum, bn.eps, act, residual,
         )
     y = F.batch_norm(
         x, bn.running_mean, bn.running_var, bn.weight, bn.bias,
         bn.training, bn.momentum if bn.momentum is not None else 0.1, bn.eps,
     )
     if act == "silu":
-        return F.silu(y)
-    if act == "relu":
-        return F.relu(y)
+        y = F.silu(y)
+    elif act == "relu":
+        y = F.relu(y)
+    if residual is not None:
+        y = y + residual
     return y
 
 

@@ -204,10 +204,11 @@ __global__ void bn_finalize_eval_kernel(
 // fused normalize + act elementwise: y = act(scale*x + shift)
 // scale/shift live in registers; rows stream.
 // ---------------------------------------------------------------------------
-template <typename T, Act ACT, int VEC>
+template <typename T, Act ACT, int VEC, bool RES>
 __global__ void bn_act_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
                                   const float* __restrict__ scale,
-                                  const float* __restrict__ shift, long long M, int C,
+                                  const float* __restrict__ shift,
+                                  const T* __restrict__ res, long long M, int C,
                                   int log2_cpb, int rows_per_chunk) {
   const int cpb = 1 << log2_cpb;
   const int slot = threadIdx.x & (cpb - 1);
@@ -229,15 +230,20 @@ __global__ void bn_act_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
   const long long r1 = min(r0 + rows_per_chunk, M);
   long long r = r0 + rg;
   for (; r + nrg < r1; r += 2 * (long long)nrg) {
-    BVec<T, VEC> xv[2], yv[2];
+    BVec<T, VEC> xv[2], rv[2], yv[2];
 #pragma unroll
-    for (int u = 0; u < 2; ++u) xv[u] = bvload<T, VEC>(x + (r + u * (long long)nrg) * C + c);
+    for (int u = 0; u < 2; ++u) {
+      xv[u] = bvload<T, VEC>(x + (r + u * (long long)nrg) * C + c);
+      if (RES) rv[u] = bvload<T, VEC>(res + (r + u * (long long)nrg) * C + c);
+    }
 #pragma unroll
     for (int u = 0; u < 2; ++u) {
 #pragma unroll
       for (int j = 0; j < VEC; ++j) {
         const float z = fmaf(DfdCvt<T>::to_f32(xv[u].v[j]), sc[j], sh[j]);
-        yv[u].v[j] = DfdCvt<T>::from_f32(act_fwd(z, ACT));
+        float v = act_fwd(z, ACT);
+        if (RES) v += DfdCvt<T>::to_f32(rv[u].v[j]);
+        yv[u].v[j] = DfdCvt<T>::from_f32(v);
       }
       bvstore<T, VEC>(y + (r + u * (long long)nrg) * C + c, yv[u]);
     }
@@ -248,7 +254,9 @@ __global__ void bn_act_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
       const float z = fmaf(DfdCvt<T>::to_f32(xv.v[j]), sc[j], sh[j]);
-      yv.v[j] = DfdCvt<T>::from_f32(act_fwd(z, ACT));
+      float v = act_fwd(z, ACT);
+      if (RES) v += DfdCvt<T>::to_f32(res[r * C + c + j]);
+      yv.v[j] = DfdCvt<T>::from_f32(v);
     }
     bvstore<T, VEC>(y + r * C + c, yv);
   }
@@ -482,7 +490,15 @@ Act act_from_string(const std::string& s) {
 std::vector<at::Tensor> bn_act_fwd(
     at::Tensor x, at::Tensor weight, at::Tensor bias,
     at::Tensor running_mean, at::Tensor running_var,
-    bool training, double momentum, double eps, std::string act_s) {
+    bool training, double momentum, double eps, std::string act_s,
+    at::Tensor residual) {
+  const bool has_res = residual.defined();
+  if (has_res) {
+    TORCH_CHECK(residual.sizes() == x.sizes() && residual.scalar_type() == x.scalar_type(),
+                "bn_act_fwd: residual must match x");
+    TORCH_CHECK(residual.is_contiguous(at::MemoryFormat::ChannelsLast),
+                "bn_act_fwd: residual must be channels_last");
+  }
   TORCH_CHECK(x.is_cuda() && x.dim() == 4, "bn_act_fwd: 4D CUDA tensor expected");
   TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast),
               "bn_act_fwd: channels_last input required");
@@ -534,10 +550,19 @@ std::vector<at::Tensor> bn_act_fwd(
   DISPATCH_DTYPE(x.scalar_type(), "bn_act_fwd", [&] {
     DISPATCH_ACT(act, [&] {
       DISPATCH_VEC(vec, [&] {
-        hipLaunchKernelGGL((bn_act_fwd_kernel<T, ACT, KVEC>), grid, dim3(256), 0, stream,
-                           (const T*)x.data_ptr(), (T*)y.data_ptr(),
-                           scale.data_ptr<float>(), shift.data_ptr<float>(), M, C,
-                           plan.log2_cpb, plan.rows_per_chunk);
+        if (has_res) {
+          hipLaunchKernelGGL((bn_act_fwd_kernel<T, ACT, KVEC, true>), grid, dim3(256), 0,
+                             stream, (const T*)x.data_ptr(), (T*)y.data_ptr(),
+                             scale.data_ptr<float>(), shift.data_ptr<float>(),
+                             (const T*)residual.data_ptr(), M, C,
+                             plan.log2_cpb, plan.rows_per_chunk);
+        } else {
+          hipLaunchKernelGGL((bn_act_fwd_kernel<T, ACT, KVEC, false>), grid, dim3(256), 0,
+                             stream, (const T*)x.data_ptr(), (T*)y.data_ptr(),
+                             scale.data_ptr<float>(), shift.data_ptr<float>(),
+                             (const T*)nullptr, M, C,
+                             plan.log2_cpb, plan.rows_per_chunk);
+        }
       });
     });
   });

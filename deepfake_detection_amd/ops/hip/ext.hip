@@ -7,7 +7,7 @@
 std::vector<at::Tensor> bn_act_fwd(at::Tensor x, at::Tensor weight, at::Tensor bias,
                                    at::Tensor running_mean, at::Tensor running_var,
                                    bool training, double momentum, double eps,
-                                   std::string act);
+                                   std::string act, at::Tensor residual);
 std::vector<at::Tensor> bn_act_bwd(at::Tensor dy, at::Tensor x, at::Tensor weight,
                                    at::Tensor bias, at::Tensor save_mean,
                                    at::Tensor save_invstd, bool training, std::string act);
@@ -41,7 +41,10 @@ at::Tensor pw_conv2d_fwd_mfma(at::Tensor x, at::Tensor w);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "deepfake_detection_amd gfx950 (MI355X/CDNA4) kernels";
-  m.def("bn_act_fwd", &bn_act_fwd, "fused BatchNorm+act forward (NHWC)");
+  m.def("bn_act_fwd", &bn_act_fwd, "fused BatchNorm+act(+residual) forward (NHWC)",
+        py::arg("x"), py::arg("weight"), py::arg("bias"), py::arg("running_mean"),
+        py::arg("running_var"), py::arg("training"), py::arg("momentum"),
+        py::arg("eps"), py::arg("act"), py::arg("residual") = at::Tensor());
   m.def("bn_act_bwd", &bn_act_bwd, "fused BatchNorm+act backward (NHWC)");
   m.def("normalize_uint8_nhwc", &normalize_uint8_nhwc, "uint8 NCHW -> norm NHWC");
   m.def("global_avg_pool_fwd", &global_avg_pool_fwd, "global avg pool fwd (NHWC)");

@@ -357,3 +357,31 @@ def test_pwconv_mfma_matches_torch(m, k, n):
     ref = torch.nn.functional.conv2d(x.float(), w.float())
     assert y.shape == ref.shape
     assert torch.allclose(y.float(), ref, atol=0.1 + 0.02 * (k ** 0.5), rtol=0.05)
+
+
+@pytest.mark.parametrize("act", ["none", "silu"])
+def test_bn_act_fused_residual(act):
+    """y = act(bn(x)) + res in one kernel; residual grad = upstream grad."""
+    from deepfake_detection_amd.ops.bn_act import fused_bn_act
+
+    torch.manual_seed(3)
+    N, C, H, W = 4, 48, 17, 19
+    x = _cl(torch.randn(N, C, H, W, device="cuda")).requires_grad_(True)
+    res = _cl(torch.randn(N, C, H, W, device="cuda")).requires_grad_(True)
+    bn = torch.nn.BatchNorm2d(C, momentum=0.01, eps=1e-3).cuda()
+
+    y = fused_bn_act(x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
+                     True, 0.01, 1e-3, act, res)
+    rm = bn.running_mean.clone().zero_()
+    rv = bn.running_var.clone().fill_(1)
+    ref = torch.nn.functional.batch_norm(
+        x.detach(), rm, rv, bn.weight.detach(), bn.bias.detach(), True, 0.01, 1e-3)
+    if act == "silu":
+        ref = torch.nn.functional.silu(ref)
+    ref = ref + res.detach()
+    assert torch.allclose(y, ref, atol=2e-4, rtol=2e-4)
+
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    assert torch.allclose(res.grad, dy)
+    assert x.grad is not None and torch.isfinite(x.grad).all()
