@@ -491,7 +491,8 @@ std::vector<at::Tensor> bn_act_fwd(
     at::Tensor x, at::Tensor weight, at::Tensor bias,
     at::Tensor running_mean, at::Tensor running_var,
     bool training, double momentum, double eps, std::string act_s,
-    at::Tensor residual) {
+    c10::optional<at::Tensor> residual_opt) {
+  at::Tensor residual = residual_opt.has_value() ? *residual_opt : at::Tensor();
   const bool has_res = residual.defined();
   if (has_res) {
     TORCH_CHECK(residual.sizes() == x.sizes() && residual.scalar_type() == x.scalar_type(),

@@ -7,7 +7,7 @@
 std::vector<at::Tensor> bn_act_fwd(at::Tensor x, at::Tensor weight, at::Tensor bias,
                                    at::Tensor running_mean, at::Tensor running_var,
                                    bool training, double momentum, double eps,
-                                   std::string act, at::Tensor residual);
+                                   std::string act, c10::optional<at::Tensor> residual);
 std::vector<at::Tensor> bn_act_bwd(at::Tensor dy, at::Tensor x, at::Tensor weight,
                                    at::Tensor bias, at::Tensor save_mean,
                                    at::Tensor save_invstd, bool training, std::string act);
@@ -44,7 +44,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bn_act_fwd", &bn_act_fwd, "fused BatchNorm+act(+residual) forward (NHWC)",
         py::arg("x"), py::arg("weight"), py::arg("bias"), py::arg("running_mean"),
         py::arg("running_var"), py::arg("training"), py::arg("momentum"),
-        py::arg("eps"), py::arg("act"), py::arg("residual") = at::Tensor());
+        py::arg("eps"), py::arg("act"), py::arg("residual") = py::none());
   m.def("bn_act_bwd", &bn_act_bwd, "fused BatchNorm+act backward (NHWC)");
   m.def("normalize_uint8_nhwc", &normalize_uint8_nhwc, "uint8 NCHW -> norm NHWC");
   m.def("global_avg_pool_fwd", &global_avg_pool_fwd, "global avg pool fwd (NHWC)");
