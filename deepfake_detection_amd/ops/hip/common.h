@@ -89,3 +89,58 @@ DFD_DEV float block_sum(float v, float* lds_scratch) {
   __syncthreads();
   return out;
 }
+
+// ---- vectorized channel slices -------------------------------------------
+// Every NHWC kernel moves VEC contiguous channels per thread (16 B loads for
+// 2-byte dtypes at VEC=8).
+template <typename T, int N>
+struct alignas(sizeof(T) * N) DfdVec {
+  T v[N];
+};
+
+template <typename T, int N>
+DFD_DEV DfdVec<T, N> dfd_vload(const T* p) {
+  return *reinterpret_cast<const DfdVec<T, N>*>(p);
+}
+
+template <typename T, int N>
+DFD_DEV void dfd_vstore(T* p, const DfdVec<T, N>& x) {
+  *reinterpret_cast<DfdVec<T, N>*>(p) = x;
+}
+
+// largest power-of-two vector width (≤16 B per lane) dividing C
+static inline int dfd_pick_vec(long long c, int elem_size) {
+  const int max_vec = elem_size == 4 ? 4 : 8;
+  for (int v = max_vec; v > 1; v >>= 1)
+    if (c % v == 0) return v;
+  return 1;
+}
+
+// ---- (channel-slot, row-group) block plan --------------------------------
+// Blocks are cpb power-of-two channel slots × row groups; the grid splits
+// rows into chunks. Chunks aim for ~kMaxGrid blocks but keep ≥min_iters
+// row-iterations per thread so per-block reductions/atomics amortize.
+struct DfdPlan {
+  int log2_cpb, ctiles, chunks, rows_per_chunk;
+};
+
+static inline DfdPlan dfd_plan(int cv, long long rows, long long base_blocks = 1,
+                               int min_iters = 16) {
+  DfdPlan p;
+  int l = 0;
+  while ((1 << l) < cv && l < 6) ++l;
+  p.log2_cpb = l;
+  const int cpb = 1 << l;
+  p.ctiles = (cv + cpb - 1) / cpb;
+  const long long base = base_blocks * p.ctiles;
+  long long want = (kMaxGrid + base - 1) / base;
+  const int nrg = 256 >> l;
+  long long by_iters = rows / ((long long)nrg * min_iters);
+  if (want > by_iters) want = by_iters;
+  long long max_chunks = (rows + nrg - 1) / nrg;
+  if (want > max_chunks) want = max_chunks;
+  if (want < 1) want = 1;
+  p.chunks = (int)want;
+  p.rows_per_chunk = (int)((rows + p.chunks - 1) / p.chunks);
+  return p;
+}
