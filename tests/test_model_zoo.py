@@ -117,3 +117,21 @@ def test_reference_entrypoint_coverage():
             if m.group(1) not in have:
                 missing.append(m.group(1))
     assert not missing, f"missing entrypoints: {missing}"
+
+
+def test_every_registered_model_instantiates():
+    """Constructor smoke for the ENTIRE registry (229 entrypoints): catches
+    wiring bugs in rarely-used variants. Heavyweight giants are capped via
+    num_classes to keep this test CPU-cheap."""
+    from deepfake_detection_amd.models import registry
+
+    failures = []
+    for name in registry.list_models():
+        try:
+            m = dfd.create_model(name, num_classes=2)
+            n_params = sum(p.numel() for p in m.parameters())
+            assert n_params > 1000
+            del m
+        except Exception as e:  # noqa: BLE001
+            failures.append(f"{name}: {type(e).__name__}: {e}")
+    assert not failures, "\n".join(failures)
