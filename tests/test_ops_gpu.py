@@ -385,3 +385,26 @@ def test_bn_act_fused_residual(act):
     y.backward(dy)
     assert torch.allclose(res.grad, dy)
     assert x.grad is not None and torch.isfinite(x.grad).all()
+
+
+@pytest.mark.skipif(os.environ.get("DFD_AMD_PW_MFMA", "0") != "1",
+                    reason="experimental MFMA pointwise path (set DFD_AMD_PW_MFMA=1)")
+def test_pwconv_mfma_autograd():
+    """fwd + bwd-data (same kernel, transposed weight) + bwd-weight (rocBLAS)
+    vs fp32 torch conv."""
+    from deepfake_detection_amd.ops.pwconv import pw_conv2d
+
+    torch.manual_seed(12)
+    B, K, N, H = 3, 96, 160, 13
+    x = _cl(torch.randn(B, K, H, H, device="cuda", dtype=torch.bfloat16)).requires_grad_(True)
+    w = torch.randn(N, K, 1, 1, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    y = pw_conv2d(x, w)
+    rx = x.detach().float().requires_grad_(True)
+    rw = w.detach().float().requires_grad_(True)
+    ref = torch.nn.functional.conv2d(rx, rw)
+    assert torch.allclose(y.float(), ref, atol=0.3, rtol=0.05)
+    dy = torch.randn_like(ref)
+    ref.backward(dy)
+    y.backward(dy.to(torch.bfloat16))
+    assert torch.allclose(x.grad.float(), rx.grad, atol=0.5, rtol=0.05)
+    assert torch.allclose(w.grad.float(), rw.grad, atol=2.0, rtol=0.05)
