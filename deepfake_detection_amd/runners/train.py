@@ -201,6 +201,17 @@ def main(rank, args, args_text, world_size=None, start_rank=0):
 
     data_config = resolve_data_config(vars(args), model=model, verbose=whole_rank == 0)
 
+    # AugMix splits + auxiliary split-BN (reference train.py:330-337)
+    num_aug_splits = 0
+    if args.aug_splits > 0:
+        assert args.aug_splits > 1, "A split of 1 makes no sense"
+        num_aug_splits = args.aug_splits
+    if args.split_bn:
+        assert num_aug_splits > 1 or args.resplit
+        from ..models.layers_extra import convert_splitbn_model
+
+        model = convert_splitbn_model(model, max(num_aug_splits, 2))
+
     model = model.to(device)
     if use_cuda:
         model = model.to(memory_format=torch.channels_last)
@@ -262,9 +273,14 @@ def main(rank, args, args_text, world_size=None, start_rank=0):
         # (reference train.py:442-445)
         collate_fn = None
         if args.prefetcher and args.mixup > 0:
+            assert not num_aug_splits, "mixup collate conflicts with aug splits"
             from ..data import FastCollateMixup
 
             collate_fn = FastCollateMixup(args.mixup, args.smoothing, args.num_classes)
+        if num_aug_splits > 1:
+            from ..data.dataset import AugMixDataset
+
+            dataset_train = AugMixDataset(dataset_train, num_splits=num_aug_splits)
         loader_train = create_deepfake_loader_v3(
             dataset_train, input_size=data_config["input_size"], batch_size=args.batch_size,
             collate_fn=collate_fn,
@@ -285,7 +301,13 @@ def main(rank, args, args_text, world_size=None, start_rank=0):
             fp16=False, dtype="bfloat16" if args.amp else "float32")
 
     # loss selection (reference train.py:506-520)
-    if args.mixup > 0.0:
+    if args.jsd:
+        assert num_aug_splits > 1, "JSD only valid with aug splits set"
+        from ..loss import JsdCrossEntropy
+
+        train_loss_fn = JsdCrossEntropy(num_splits=num_aug_splits,
+                                        smoothing=args.smoothing).to(device)
+    elif args.mixup > 0.0:
         train_loss_fn = SoftTargetCrossEntropy().to(device)
     elif args.smoothing:
         train_loss_fn = LabelSmoothingCrossEntropy(smoothing=args.smoothing).to(device)
