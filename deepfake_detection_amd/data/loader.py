@@ -250,16 +250,20 @@ def create_loader(
         interpolation="bilinear", mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD,
         num_workers=1, distributed=False, crop_pct=None, collate_fn=None,
         pin_memory=True, fp16=False, dtype=None, tf_preprocessing=False,
-        persistent_workers=True):
-    """Generic single-image loader (reference loader.py:372)."""
+        auto_augment=None, num_aug_splits=0, persistent_workers=True):
+    """Generic single-image loader (reference loader.py:372). With
+    num_aug_splits>1 the dataset must be an AugMixDataset: the train
+    transform is built as a (primary, secondary, final) triple."""
     re_num_splits = 0
     if re_split:
-        re_num_splits = 2
+        re_num_splits = num_aug_splits or 2
 
     transform = create_transform(
         input_size, is_training=is_training, use_prefetcher=use_prefetcher,
-        color_jitter=color_jitter, interpolation=interpolation, mean=mean, std=std,
-        crop_pct=crop_pct, tf_preprocessing=tf_preprocessing)
+        color_jitter=color_jitter, auto_augment=auto_augment,
+        interpolation=interpolation, mean=mean, std=std,
+        crop_pct=crop_pct, tf_preprocessing=tf_preprocessing,
+        separate=num_aug_splits > 1 and is_training)
     if hasattr(dataset, "set_transform"):
         dataset.set_transform(transform)
     else:

@@ -72,11 +72,16 @@ def transforms_deepfake_eval_v3(img_size=224, use_prefetcher=True):
 def transforms_imagenet_train(
         img_size=224, scale=(0.08, 1.0), color_jitter=0.4, interpolation="random",
         auto_augment=None, random_erasing=0.0, random_erasing_mode="const",
-        use_prefetcher=False, mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD):
-    tfl = [
+        use_prefetcher=False, mean=IMAGENET_DEFAULT_MEAN, std=IMAGENET_DEFAULT_STD,
+        separate=False):
+    """`separate=True` returns the (primary, secondary, final) transform
+    triple AugMixDataset consumes (reference transforms_factory.py:239-318):
+    primary = crop+flip, secondary = AA/jitter, final = tensorize+normalize."""
+    primary_tfl = [
         RandomResizedCropAndInterpolation(img_size, scale=scale, interpolation=interpolation),
         transforms.RandomHorizontalFlip(),
     ]
+    tfl = [] if separate else primary_tfl
     if auto_augment:
         # AA/RandAugment/AugMix replace color jitter
         # (reference transforms_factory.py:269-287)
@@ -109,16 +114,21 @@ def transforms_imagenet_train(
             color_jitter = (float(color_jitter),) * 3
         tfl += [transforms.ColorJitter(*color_jitter)]
 
+    final_tfl = []
     if use_prefetcher:
-        tfl += [ToNumpy()]
+        final_tfl += [ToNumpy()]
     else:
-        tfl += [
+        final_tfl += [
             transforms.ToTensorNormalized(),
             transforms.Normalize(mean=torch.tensor(mean), std=torch.tensor(std)),
         ]
         if random_erasing > 0.0:
-            tfl.append(RandomErasing(random_erasing, mode=random_erasing_mode, device="cpu"))
-    return transforms.Compose(tfl)
+            final_tfl.append(
+                RandomErasing(random_erasing, mode=random_erasing_mode, device="cpu"))
+    if separate:
+        return (transforms.Compose(primary_tfl), transforms.Compose(tfl),
+                transforms.Compose(final_tfl))
+    return transforms.Compose(tfl + final_tfl)
 
 
 def transforms_imagenet_eval(
@@ -151,7 +161,8 @@ def transforms_imagenet_eval(
 def create_transform(
         input_size, is_training=False, use_prefetcher=False, color_jitter=0.4,
         auto_augment=None, interpolation="bilinear", mean=IMAGENET_DEFAULT_MEAN,
-        std=IMAGENET_DEFAULT_STD, crop_pct=None, tf_preprocessing=False):
+        std=IMAGENET_DEFAULT_STD, crop_pct=None, tf_preprocessing=False,
+        separate=False):
     if isinstance(input_size, tuple):
         img_size = input_size[-2:]
     else:
@@ -166,7 +177,7 @@ def create_transform(
         return transforms_imagenet_train(
             img_size, color_jitter=color_jitter, auto_augment=auto_augment,
             interpolation=interpolation if interpolation != "bilinear" else "random",
-            use_prefetcher=use_prefetcher, mean=mean, std=std)
+            use_prefetcher=use_prefetcher, mean=mean, std=std, separate=separate)
     return transforms_imagenet_eval(
         img_size, interpolation=interpolation, use_prefetcher=use_prefetcher,
         mean=mean, std=std, crop_pct=crop_pct)

@@ -148,3 +148,34 @@ def test_legacy_datasets(tmp_path):
     dt = DatasetTar(str(tar_path))
     img, y = dt[0]
     assert y == 0 and img.size == (32, 32)
+
+
+def test_aug_splits_separate_transform_and_augmix_dataset():
+    """AugMix JSD data path: separate (primary, secondary, final) transform
+    triple consumed by AugMixDataset (reference transforms_factory.py:239-318,
+    dataset.py:633)."""
+    import numpy as np
+    from PIL import Image as PILImage
+
+    from deepfake_detection_amd.data.dataset import AugMixDataset
+    from deepfake_detection_amd.data.transforms_factory import transforms_imagenet_train
+
+    triple = transforms_imagenet_train(64, auto_augment="augmix-m3",
+                                       use_prefetcher=True, separate=True)
+    assert isinstance(triple, tuple) and len(triple) == 3
+
+    class _DS:
+        transform = triple
+
+        def __getitem__(self, i):
+            img = PILImage.fromarray(
+                (np.random.rand(80, 80, 3) * 255).astype(np.uint8))
+            return self.transform(img), i % 2
+
+        def __len__(self):
+            return 4
+
+    ds = AugMixDataset(_DS(), num_splits=3)
+    xs, y = ds[0]
+    assert len(xs) == 3
+    assert all(x.shape == xs[0].shape for x in xs)
