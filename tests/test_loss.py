@@ -44,3 +44,28 @@ def test_jsd_runs():
     jsd = JsdCrossEntropy(num_splits=3, alpha=12, smoothing=0.1)
     loss = jsd(x, t)
     assert loss.ndim == 0 and torch.isfinite(loss)
+
+
+def test_jsd_with_splitbn_end_to_end():
+    """AugMix training step: split-BN model + JSD loss over aug splits
+    (reference train.py:330-337,506-509)."""
+    import deepfake_detection_amd as dfd
+    from deepfake_detection_amd.loss import JsdCrossEntropy
+    from deepfake_detection_amd.models.layers_extra import convert_splitbn_model
+
+    torch.manual_seed(0)
+    splits = 3
+    model = dfd.create_model("resnet18", num_classes=4)
+    model = convert_splitbn_model(model, splits)
+    loss_fn = JsdCrossEntropy(num_splits=splits, smoothing=0.1)
+    # batch layout: clean split first, then augmented splits (AugMixDataset
+    # + fast_collate tuple deinterleave)
+    B = 4
+    x = torch.randn(B * splits, 3, 64, 64)
+    y = torch.randint(0, 4, (B,)).repeat(splits)
+    out = model(x)
+    loss = loss_fn(out, y)
+    assert torch.isfinite(loss)
+    loss.backward()
+    g = [p.grad for p in model.parameters() if p.grad is not None]
+    assert g and all(torch.isfinite(t).all() for t in g)

@@ -117,3 +117,17 @@ def test_nonstrict_load_drops_mismatched(tmp_path):
     m3 = dfd.create_deepfake_model_v4(
         "efficientnet_deepfake_v4", num_classes=5, in_chans=12, checkpoint_path=str(p))
     assert m3.classifier.out_features == 5
+
+
+def test_efficientnet_features_backbone():
+    """EfficientNetFeatures multi-stage extractor (reference
+    efficientnet.py:458-518): features_only returns per-stage pyramids."""
+    import deepfake_detection_amd as dfd
+
+    m = dfd.create_model("efficientnet_b0", features_only=True)
+    m.eval()
+    with torch.no_grad():
+        feats = m(torch.randn(1, 3, 128, 128))
+    assert isinstance(feats, (list, tuple)) and len(feats) >= 4
+    hw = [f.shape[-1] for f in feats]
+    assert hw == sorted(hw, reverse=True)  # decreasing spatial pyramid
