@@ -18,7 +18,7 @@ import torch
 import torch.nn as nn
 
 from ..ops import functional as O
-from .layers import CondConv2d, create_conv2d, drop_path, sigmoid
+from .layers import create_conv2d, drop_path, sigmoid
 
 __all__ = [
     "BN_MOMENTUM_TF_DEFAULT",

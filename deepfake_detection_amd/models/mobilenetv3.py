@@ -6,7 +6,6 @@ same arch-string decoder and EfficientNetBuilder as the EfficientNets.
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops import functional as O
 from .blocks import resolve_bn_args, round_channels
 from .builder import EfficientNetBuilder, decode_arch_def, efficientnet_init_weights
 from .layers import HardSwish, SelectAdaptivePool2d, create_conv2d, hard_sigmoid

@@ -7,7 +7,7 @@ import torch
 import torch.nn as nn
 
 from .registry import register_model
-from .resnet import ResNet, default_cfgs as _resnet_cfgs
+from .resnet import ResNet
 
 
 def _cfg(**kwargs):

@@ -6,7 +6,6 @@ replacing the reference's ~6 eager CUDA kernels per parameter tensor
 (reference rmsprop_tf.py:86-120, timm/utils.py:329-340).
 """
 
-import torch
 
 from .extension import load_extension
 

@@ -1,6 +1,5 @@
 """Running meters + accuracy (reference dfd/timm/utils.py:152-186)."""
 
-import torch
 
 
 class AverageMeter:
