@@ -179,3 +179,24 @@ def test_aug_splits_separate_transform_and_augmix_dataset():
     xs, y = ds[0]
     assert len(xs) == 3
     assert all(x.shape == xs[0].shape for x in xs)
+
+
+def test_every_autoaugment_op_applies():
+    """Apply every registered AA op at several magnitudes — catches latent
+    crashes in rarely-drawn ops (reference auto_augment.py op table)."""
+    import numpy as np
+    from PIL import Image as PILImage
+
+    from deepfake_detection_amd.data.auto_augment import NAME_TO_OP, AugmentOp
+
+    img = PILImage.fromarray((np.random.rand(64, 56, 3) * 255).astype(np.uint8))
+    failures = []
+    for name in NAME_TO_OP:
+        for mag in (0, 5, 10):
+            try:
+                op = AugmentOp(name, prob=1.0, magnitude=mag)
+                out = op(img)
+                assert out.size[0] > 0
+            except Exception as e:  # noqa: BLE001
+                failures.append(f"{name}@{mag}: {e}")
+    assert not failures, failures
