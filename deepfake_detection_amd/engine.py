@@ -147,16 +147,20 @@ def validate(model, loader, loss_fn, args, device, world_size=1, rank=0, log_suf
             if use_cuda:
                 input = input.contiguous(memory_format=torch.channels_last)
 
-            if getattr(args, "tta", 0) > 1:
-                # TTA oversampling: average predictions over the repeat dim
-                # (reference train.py:724-727)
-                input = input.repeat(getattr(args, "tta"), 1, 1, 1)
+            tta = getattr(args, "tta", 0)
+            if tta > 1:
+                # TTA oversampling: run tta stochastic passes (block-repeated
+                # batch) and average predictions (reference train.py:724-727
+                # unfold-mean, adapted to block layout)
+                input = input.repeat(tta, 1, 1, 1)
 
             with _autocast(use_amp):
                 output = model(input)
             if isinstance(output, (tuple, list)):
                 output = output[0]
             output = output.float()
+            if tta > 1:
+                output = output.view(tta, -1, output.shape[-1]).mean(dim=0)
 
             loss = loss_fn(output, target)
             prec1 = accuracy(output, target)[0]

@@ -87,3 +87,20 @@ def test_inference_runner_cpu(tmp_path):
     assert len(results) == 1
     score = results[0][1]
     assert 0.0 <= score <= 1.0
+
+
+def test_validate_tta_reduces_output():
+    """--tta > 1 oversamples the batch and averages predictions back to B
+    rows (the first cut repeated inputs without reducing outputs)."""
+    import types
+
+    import deepfake_detection_amd as dfd
+    from deepfake_detection_amd.engine import validate
+
+    model = dfd.create_model("resnet18", num_classes=3).eval()
+    batches = [(torch.randn(4, 3, 64, 64), torch.randint(0, 3, (4,)))]
+    args = types.SimpleNamespace(log_interval=100, prefetcher=False, tta=3, amp=False)
+    metrics = validate(model, batches, torch.nn.CrossEntropyLoss(), args,
+                       torch.device("cpu"))
+    assert set(metrics) == {"loss", "prec1"}
+    assert torch.isfinite(torch.tensor(metrics["loss"]))
