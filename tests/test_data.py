@@ -224,3 +224,37 @@ def test_legacy_dataset_and_loader_variants(tmp_path):
     _, t0 = ds2[0]
     _, t1 = ds2[1]
     assert (t0, t1) == (0, 1)
+
+
+def test_fast_collate_mixup_math():
+    """Collate-time uint8 mixing: x_i*lam + x_rev_i*(1-lam); soft targets sum
+    to 1 (reference mixup.py:27-51)."""
+    import numpy as np
+
+    from deepfake_detection_amd.data import FastCollateMixup
+
+    c = FastCollateMixup(mixup_alpha=0.8, label_smoothing=0.1, num_classes=4)
+    batch = [(np.full((3, 8, 8), v, dtype=np.uint8), v % 4) for v in (10, 200)]
+    x, y = c(batch)
+    assert x.shape == (2, 3, 8, 8) and x.dtype == torch.uint8
+    assert y.shape == (2, 4)
+    assert torch.allclose(y.sum(1), torch.ones(2), atol=1e-5)
+    # mixed pixel values lie between the two sources
+    assert 10 <= x[0, 0, 0, 0].item() <= 200
+
+    c.mixup_enabled = False
+    x2, y2 = c(batch)
+    assert x2[0, 0, 0, 0].item() == 10 and x2[1, 0, 0, 0].item() == 200
+
+
+def test_random_erasing_per_frame_slices():
+    """img_num-aware GPU RandomErasing erases each 3-channel frame slice
+    independently (reference random_erasing.py:96-100)."""
+    from deepfake_detection_amd.data import RandomErasing
+
+    torch.manual_seed(0)
+    re = RandomErasing(probability=1.0, mode="const", img_num=4, device="cpu")
+    x = torch.ones(2, 12, 32, 32)
+    out = re(x.clone())
+    assert out.shape == x.shape
+    assert not torch.equal(out, x)  # something was erased
