@@ -408,3 +408,20 @@ def test_pwconv_mfma_autograd():
     y.backward(dy.to(torch.bfloat16))
     assert torch.allclose(x.grad.float(), rx.grad, atol=0.5, rtol=0.05)
     assert torch.allclose(w.grad.float(), rw.grad, atol=2.0, rtol=0.05)
+
+
+def test_conv2d_same_depthwise_routes_to_hip():
+    """tf_-model SAME-pad depthwise path: pad_same (torch) + HIP dw kernel
+    with padding 0 must equal F.conv2d over the padded input."""
+    from deepfake_detection_amd.models.layers import Conv2dSame
+
+    torch.manual_seed(5)
+    m = Conv2dSame(32, 32, 5, stride=2, groups=32, bias=False).cuda()
+    x = _cl(torch.randn(2, 32, 27, 31, device="cuda"))
+    y = m(x)
+    from deepfake_detection_amd.models.layers import pad_same
+
+    ref = torch.nn.functional.conv2d(
+        pad_same(x, [5, 5], [2, 2]), m.weight, None, 2, 0, 1, 32)
+    assert y.shape == ref.shape
+    assert torch.allclose(y, ref, atol=1e-4, rtol=1e-4)
