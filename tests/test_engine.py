@@ -104,3 +104,28 @@ def test_validate_tta_reduces_output():
                        torch.device("cpu"))
     assert set(metrics) == {"loss", "prec1"}
     assert torch.isfinite(torch.tensor(metrics["loss"]))
+
+
+def test_train_epoch_recovery_interval(tmp_path):
+    """Mid-epoch recovery checkpoints every --recovery-interval batches
+    (reference train.py:686-689, utils.py:128-149)."""
+    import glob
+    import types
+
+    import deepfake_detection_amd as dfd
+    from deepfake_detection_amd.engine import train_epoch
+    from deepfake_detection_amd.utils import CheckpointSaver
+
+    model = dfd.create_model("resnet18", num_classes=2)
+    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    saver = CheckpointSaver(checkpoint_dir=str(tmp_path), recovery_dir=str(tmp_path))
+    batches = [(torch.randn(2, 3, 32, 32), torch.randint(0, 2, (2,)))
+               for _ in range(8)]
+    args = types.SimpleNamespace(log_interval=100, recovery_interval=2,
+                                 save_images=False, tta=0, prefetcher=False)
+    train_epoch(0, model, batches, opt, torch.nn.CrossEntropyLoss(), args,
+                torch.device("cpu"), saver=saver)
+    recs = glob.glob(str(tmp_path / "recovery-*"))
+    # reference cleanup lags one save (utils.py:128-140): current + previous
+    assert len(recs) == 2
+    assert saver.find_recovery()

@@ -64,3 +64,31 @@ def test_factory_cosine_adds_cooldown():
     opt = _opt(0.5)
     sched, num_epochs = create_scheduler(args, opt)
     assert num_epochs == 13
+
+
+def test_lr_noise_deterministic_and_bounded():
+    """LR-noise injection (reference scheduler.py:87-105): seeded, applied
+    only inside the noise window, and multiplicative around the base LR."""
+    import torch
+
+    from deepfake_detection_amd.scheduler import StepLRScheduler
+
+    def make():
+        model = torch.nn.Linear(2, 2)
+        opt = torch.optim.SGD(model.parameters(), lr=1.0)
+        return StepLRScheduler(opt, decay_t=100, decay_rate=1.0,
+                               noise_range_t=(2, 100), noise_pct=0.1,
+                               noise_seed=42), opt
+
+    s1, o1 = make()
+    s2, o2 = make()
+    lrs1, lrs2 = [], []
+    for t in range(6):
+        s1.step(t)
+        s2.step(t)
+        lrs1.append(o1.param_groups[0]["lr"])
+        lrs2.append(o2.param_groups[0]["lr"])
+    assert lrs1 == lrs2  # seeded determinism
+    assert lrs1[0] == 1.0 and lrs1[1] == 1.0  # before the window: no noise
+    assert any(lr != 1.0 for lr in lrs1[2:])  # inside: noise applied
+    assert all(0.5 < lr < 1.5 for lr in lrs1)  # bounded (pct=0.1 normal)
