@@ -308,10 +308,10 @@ class MedianPool2d(nn.Module):
 
 
 class TestTimePoolHead(nn.Module):
-    __test__ = False  # "Test" prefix is the reference's name, not a pytest test
-
     """Test-time pooling head: run the classifier as a 1x1 conv over the
     unpooled feature map, then average (reference test_time_pool.py:12-33)."""
+
+    __test__ = False  # "Test" prefix is the reference's name, not a pytest test
 
     def __init__(self, base, original_pool=7):
         super().__init__()
