@@ -131,3 +131,23 @@ def test_efficientnet_features_backbone():
     assert isinstance(feats, (list, tuple)) and len(feats) >= 4
     hw = [f.shape[-1] for f in feats]
     assert hw == sorted(hw, reverse=True)  # decreasing spatial pyramid
+
+
+def test_model_shapes_tool():
+    """tools/model_shapes.py inventory matches the known B4 block structure
+    (32 depthwise convs; blocks [4,7,7,10,10,13,4] minus... dw per block)."""
+    import os
+    import sys
+
+    sys.path.insert(0, os.path.join(os.path.dirname(__file__), "..", "tools"))
+    import model_shapes
+
+    import deepfake_detection_amd as dfd
+
+    m = dfd.create_model("efficientnet_b0", num_classes=2)
+    rows = model_shapes.collect(m, img_size=224)
+    totals = {}
+    for (kind, *_), n in rows.items():
+        totals[kind] = totals.get(kind, 0) + n
+    assert totals["dw"] == 16  # B0: one dw per block, 16 blocks
+    assert totals["bn"] > 30

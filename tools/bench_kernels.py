@@ -14,22 +14,23 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
-# (C, H_in, kernel, stride) for every depthwise conv in B4-299 (width 1.4,
-# depth 1.8, input 299 -> stem 150). count = occurrences per fwd pass.
+# (C, H_in, kernel, stride, count) for every depthwise conv in B4-299 —
+# exact inventory from `python tools/model_shapes.py --model efficientnet_b4
+# --img-size 299 --kind dw`.
 DW_SHAPES = [
-    # C, H, k, s, count
+    (24, 150, 3, 1, 1),
     (48, 150, 3, 1, 1),
     (144, 150, 3, 2, 1),
     (192, 75, 3, 1, 3),
     (192, 75, 5, 2, 1),
-    (288, 38, 5, 1, 3),
-    (288, 38, 3, 2, 1),
+    (336, 38, 3, 2, 1),
+    (336, 38, 5, 1, 3),
     (672, 19, 3, 1, 5),
     (672, 19, 5, 1, 1),
     (960, 19, 5, 1, 5),
     (960, 19, 5, 2, 1),
-    (1632, 10, 5, 1, 7),
     (1632, 10, 3, 1, 1),
+    (1632, 10, 5, 1, 7),
     (2688, 10, 3, 1, 1),
 ]
 
