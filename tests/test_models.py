@@ -150,4 +150,6 @@ def test_model_shapes_tool():
     for (kind, *_), n in rows.items():
         totals[kind] = totals.get(kind, 0) + n
     assert totals["dw"] == 16  # B0: one dw per block, 16 blocks
-    assert totals["bn"] > 30
+    # BN modules never run their own forward on this stack (the fused
+    # HIP/CPU path consumes their parameters directly), so no "bn" rows
+    assert "bn" not in totals

@@ -4,7 +4,9 @@
 Hooks every depthwise conv, pointwise (1x1) conv, other conv, and BatchNorm
 in a forward pass and prints the deduplicated (shape, count) table — the
 ground truth for tools/bench_kernels.py shape lists and kernel-tuning
-priorities.
+priorities. NOTE: BatchNorm rows appear only for modules whose own forward
+runs; the fused bn_act path consumes BN parameters directly, so BN shapes
+there equal the producing conv's output shapes.
 
 Usage: python tools/model_shapes.py --model efficientnet_b4 --img-size 299
        python tools/model_shapes.py --model efficientnet_deepfake_v4 --img-size 600 --in-chans 12
