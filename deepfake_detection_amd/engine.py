@@ -16,6 +16,7 @@ MI355X-native changes called out in SURVEY.md §7:
 """
 
 import logging
+import os
 import time
 from collections import OrderedDict
 
@@ -108,6 +109,19 @@ def train_epoch(epoch, model, loader, optimizer, loss_fn, args, device,
                     "{:.1f} img/s  LR: {:.3e}  Data: {:.3f}s".format(
                         epoch, batch_idx, len(loader), loss_val, losses_m.avg,
                         prec1_val, prec1_m.avg, rate, lr, data_time_m.avg))
+
+        if getattr(args, "save_images", False) and output_dir and (
+                last_batch or batch_idx % args.log_interval == 0):
+            # input-batch dumps (reference train.py:679-684)
+            try:
+                import torchvision
+
+                torchvision.utils.save_image(
+                    input[:, :3].float(),
+                    os.path.join(output_dir, f"train-batch-{batch_idx}.jpg"),
+                    padding=0, normalize=True)
+            except Exception as e:  # noqa: BLE001
+                _logger.warning("save_images failed: %s", e)
 
         if saver is not None and args.recovery_interval and (
                 last_batch or (batch_idx + 1) % args.recovery_interval == 0):

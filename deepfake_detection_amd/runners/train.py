@@ -57,8 +57,10 @@ def _build_parser():
     parser.add_argument("--synthetic-data", action="store_true", default=False,
                         help="use the synthetic face-crop dataset (benching)")
     parser.add_argument("--synthetic-len", type=int, default=2048)
-    parser.add_argument("--validation_frac", default=0.1, type=float)
-    parser.add_argument("--train_frac", default=1, type=float)
+    parser.add_argument("--validation_frac", default=0.1, type=float,
+                        help="val split when --train_frac is default")
+    parser.add_argument("--train_frac", default=1, type=float,
+                        help="train split ratio (reference train.py:65,437)")
     parser.add_argument("--label_balance", action="store_true", default=False)
     parser.add_argument("--model", default="efficientnet_deepfake_v4", type=str, metavar="MODEL")
     parser.add_argument("--class_names", default="fake,real", type=str)
@@ -186,6 +188,9 @@ def main(rank, args, args_text, world_size=None, start_rank=0):
     if args.input_size_v2:
         in_chans = int(args.input_size_v2.split(",")[0])
 
+    if getattr(args, "drop_connect", None):
+        # DEPRECATED alias (reference train.py:311): drop_connect -> drop_path
+        args.drop_path = args.drop_connect
     if args.model == "efficientnet_deepfake_v4":
         model = create_deepfake_model_v4(
             args.model, pretrained=args.pretrained, num_classes=args.num_classes,
@@ -261,13 +266,14 @@ def main(rank, args, args_text, world_size=None, start_rank=0):
         loader_train = _synthetic_loader(dataset_train, args, data_config, is_training=True)
         loader_eval = _synthetic_loader(dataset_eval, args, data_config, is_training=False)
     else:
+        train_ratio = args.train_frac if args.train_frac < 1 else 1.0 - args.validation_frac
         dataset_train = DeepFakeDataset_v3(
             args.data, args.class_names, train_split=True,
-            train_ratio=1.0 - args.validation_frac, random_state=args.seed,
+            train_ratio=train_ratio, random_state=args.seed,
             is_training=True, label_balance=args.label_balance)
         dataset_eval = DeepFakeDataset_v3(
             args.data, args.class_names, train_split=True,
-            train_ratio=1.0 - args.validation_frac, random_state=args.seed,
+            train_ratio=train_ratio, random_state=args.seed,
             is_training=False, label_balance=args.label_balance)
         # collate-time mixup when the device prefetcher owns normalization
         # (reference train.py:442-445)

@@ -129,3 +129,22 @@ def test_train_epoch_recovery_interval(tmp_path):
     # reference cleanup lags one save (utils.py:128-140): current + previous
     assert len(recs) == 2
     assert saver.find_recovery()
+
+
+def test_train_epoch_save_images(tmp_path):
+    """--save-images dumps input batches at log points (reference
+    train.py:679-684)."""
+    import glob
+    import types
+
+    import deepfake_detection_amd as dfd
+    from deepfake_detection_amd.engine import train_epoch
+
+    model = dfd.create_model("resnet18", num_classes=2)
+    opt = torch.optim.SGD(model.parameters(), lr=0.01)
+    batches = [(torch.rand(2, 3, 32, 32), torch.randint(0, 2, (2,)))]
+    args = types.SimpleNamespace(log_interval=1, recovery_interval=0,
+                                 save_images=True, tta=0, prefetcher=False)
+    train_epoch(0, model, batches, opt, torch.nn.CrossEntropyLoss(), args,
+                torch.device("cpu"), output_dir=str(tmp_path))
+    assert glob.glob(str(tmp_path / "train-batch-*.jpg"))
