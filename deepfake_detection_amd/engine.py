@@ -112,14 +112,19 @@ def train_epoch(epoch, model, loader, optimizer, loss_fn, args, device,
 
         if getattr(args, "save_images", False) and output_dir and (
                 last_batch or batch_idx % args.log_interval == 0):
-            # input-batch dumps (reference train.py:679-684)
+            # input-batch dumps (reference train.py:679-684; torchvision is
+            # not in this image, so build the grid with PIL)
             try:
-                import torchvision
+                import numpy as np
+                from PIL import Image
 
-                torchvision.utils.save_image(
-                    input[:, :3].float(),
-                    os.path.join(output_dir, f"train-batch-{batch_idx}.jpg"),
-                    padding=0, normalize=True)
+                x = input[:, :3].float().detach().cpu()
+                x = x - x.amin(dim=(1, 2, 3), keepdim=True)
+                x = x / x.amax(dim=(1, 2, 3), keepdim=True).clamp(min=1e-6)
+                grid = torch.cat(list(x), dim=2)  # side-by-side
+                arr = (grid.permute(1, 2, 0).numpy() * 255).astype(np.uint8)
+                Image.fromarray(arr).save(
+                    os.path.join(output_dir, f"train-batch-{batch_idx}.jpg"))
             except Exception as e:  # noqa: BLE001
                 _logger.warning("save_images failed: %s", e)
 
