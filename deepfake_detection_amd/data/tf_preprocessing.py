@@ -10,7 +10,6 @@ MI355X stack).
 """
 
 import numpy as np
-import torch
 from PIL import Image
 
 IMAGE_SIZE = 224
@@ -36,8 +35,12 @@ def preprocess_for_eval(img, img_size=IMAGE_SIZE):
 
 
 class TfPreprocessTransform:
-    """Drop-in for the reference TfPreprocessTransform: returns a fp32 CHW
-    tensor scaled to [0, 1] (caller applies mean/std)."""
+    """Drop-in for the reference TfPreprocessTransform: returns a uint8 CHW
+    numpy array in [0, 255] — the reference contract
+    (dfd/timm/data/tf_preprocessing.py returns `np.rollaxis(np.uint8, 2)`).
+    fast_collate packs it into the batch uint8 buffer unchanged and the
+    PrefetchLoader applies the mean*255/std*255 normalization on device; a
+    fp32 [0,1] return here would truncate to 0/1 in that buffer."""
 
     def __init__(self, is_training=False, size=IMAGE_SIZE, interpolation="bicubic"):
         if is_training:
@@ -47,5 +50,5 @@ class TfPreprocessTransform:
         self.size = size[-1] if isinstance(size, (tuple, list)) else size
 
     def __call__(self, img):
-        x = preprocess_for_eval(img, self.size) / 255.0
-        return torch.from_numpy(x.transpose(2, 0, 1))
+        x = preprocess_for_eval(img, self.size)
+        return np.rollaxis(x.astype(np.uint8), 2)

@@ -168,10 +168,13 @@ def create_transform(
     else:
         img_size = input_size
 
-    if tf_preprocessing and not is_training:
+    if tf_preprocessing and use_prefetcher:
+        # TF preprocessing emits uint8 CHW (reference contract,
+        # transforms_factory.py:507-511): the PrefetchLoader normalizes on
+        # device, so this path requires the prefetcher.
         from .tf_preprocessing import TfPreprocessTransform
 
-        return TfPreprocessTransform(is_training=False, size=img_size,
+        return TfPreprocessTransform(is_training=is_training, size=img_size,
                                      interpolation=interpolation)
     if is_training:
         return transforms_imagenet_train(

@@ -75,8 +75,12 @@ def train_epoch(epoch, model, loader, optimizer, loss_fn, args, device,
             model_ema.update(model)
 
         # local (un-reduced) running meters; cross-rank reduction happens at
-        # log points to avoid 2 all-reduces per step
-        if last_batch or batch_idx % args.log_interval == 0:
+        # log points to avoid 2 all-reduces per step. --per-step-metrics
+        # restores the reference's every-step meter/NaN-guard semantics
+        # (reference train.py:625-645) at the cost of a sync per step.
+        per_step = getattr(args, "per_step_metrics", False)
+        log_point = last_batch or batch_idx % args.log_interval == 0
+        if per_step or log_point:
             if use_cuda:
                 torch.cuda.synchronize()
             with torch.no_grad():
@@ -85,7 +89,7 @@ def train_epoch(epoch, model, loader, optimizer, loss_fn, args, device,
                 else:
                     hard_target = target
                 prec1 = accuracy(output.detach().float(), hard_target)[0]
-                if world_size > 1:
+                if world_size > 1 and log_point:
                     packed = torch.stack([loss.detach().float(), prec1])
                     packed = reduce_tensor(packed, world_size)
                     loss_val, prec1_val = packed[0].item(), packed[1].item()
@@ -100,7 +104,7 @@ def train_epoch(epoch, model, loader, optimizer, loss_fn, args, device,
                 prec1_m.update(prec1_val, input.size(0))
 
             batch_time_m.update(time.time() - end)
-            if rank == 0:
+            if rank == 0 and log_point:
                 lrl = [pg["lr"] for pg in optimizer.param_groups]
                 lr = sum(lrl) / len(lrl)
                 rate = input.size(0) * world_size / max(batch_time_m.val, 1e-9)
@@ -118,10 +122,21 @@ def train_epoch(epoch, model, loader, optimizer, loss_fn, args, device,
                 import numpy as np
                 from PIL import Image
 
-                x = input[:, :3].float().detach().cpu()
+                # cap the dump and tile into a grid of 8 per row: a single
+                # row at production batch sizes exceeds PIL's 65,535-px JPEG
+                # dimension limit (batch 384 @ 299px ~= 115k px wide)
+                x = input[:64, :3].float().detach().cpu()
                 x = x - x.amin(dim=(1, 2, 3), keepdim=True)
                 x = x / x.amax(dim=(1, 2, 3), keepdim=True).clamp(min=1e-6)
-                grid = torch.cat(list(x), dim=2)  # side-by-side
+                per_row = 8
+                n, _, h, w = x.shape
+                rows = []
+                for r0 in range(0, n, per_row):
+                    row = list(x[r0:r0 + per_row])
+                    while len(row) < per_row and n > per_row:
+                        row.append(torch.zeros(3, h, w))
+                    rows.append(torch.cat(row, dim=2))
+                grid = torch.cat(rows, dim=1)
                 arr = (grid.permute(1, 2, 0).numpy() * 255).astype(np.uint8)
                 Image.fromarray(arr).save(
                     os.path.join(output_dir, f"train-batch-{batch_idx}.jpg"))
@@ -143,6 +158,29 @@ def train_epoch(epoch, model, loader, optimizer, loss_fn, args, device,
 
     return OrderedDict([("loss", losses_m.avg), ("prec1", prec1_m.avg),
                         ("learning_rate", optimizer.param_groups[0]["lr"])])
+
+
+def _tta_views(x, tta):
+    """tta distinct deterministic views of a NCHW batch: identity, H/V flips,
+    then ±8px shifts. Identical copies through a deterministic eval model
+    would make the TTA average a no-op; these views are cheap and distinct."""
+    views = []
+    for k in range(tta):
+        if k == 0:
+            v = x
+        elif k == 1:
+            v = torch.flip(x, dims=(3,))
+        elif k == 2:
+            v = torch.flip(x, dims=(2,))
+        elif k == 3:
+            v = torch.flip(x, dims=(2, 3))
+        else:
+            s = 8 * ((k - 4) // 4 + 1)
+            axis = 2 + (k % 2)
+            sign = -1 if (k // 2) % 2 else 1
+            v = torch.roll(x, shifts=sign * s, dims=axis)
+        views.append(v)
+    return views
 
 
 def validate(model, loader, loss_fn, args, device, world_size=1, rank=0, log_suffix=""):
@@ -168,10 +206,11 @@ def validate(model, loader, loss_fn, args, device, world_size=1, rank=0, log_suf
 
             tta = getattr(args, "tta", 0)
             if tta > 1:
-                # TTA oversampling: run tta stochastic passes (block-repeated
-                # batch) and average predictions (reference train.py:724-727
-                # unfold-mean, adapted to block layout)
-                input = input.repeat(tta, 1, 1, 1)
+                # TTA: tta DISTINCT deterministic views per sample (flips,
+                # then small shifts), interleaved sample-major so the
+                # reference's unfold-mean reduction applies unchanged
+                # (reference train.py:724-727).
+                input = torch.stack(_tta_views(input, tta), dim=1).flatten(0, 1)
 
             with _autocast(use_amp):
                 output = model(input)
@@ -179,7 +218,7 @@ def validate(model, loader, loss_fn, args, device, world_size=1, rank=0, log_suf
                 output = output[0]
             output = output.float()
             if tta > 1:
-                output = output.view(tta, -1, output.shape[-1]).mean(dim=0)
+                output = output.unfold(0, tta, tta).mean(dim=2)
 
             loss = loss_fn(output, target)
             prec1 = accuracy(output, target)[0]

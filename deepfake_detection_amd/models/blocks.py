@@ -151,13 +151,13 @@ class ConvBnAct(nn.Module):
 
     def forward(self, x):
         x = self.conv(x)
-        if self._act_name != "other" and isinstance(self.bn1, nn.BatchNorm2d):
+        if self._act_name != "other" and O.fusable_bn(self.bn1):
             return O.bn_act(x, self.bn1, self._act_name)
         return self.act1(self.bn1(x))
 
 
 def _bn_act(bn, act_module, act_name, x, residual=None):
-    if act_name != "other" and isinstance(bn, nn.BatchNorm2d):
+    if act_name != "other" and O.fusable_bn(bn):
         return O.bn_act(x, bn, act_name, residual)
     y = act_module(bn(x))
     return y if residual is None else y + residual

@@ -149,13 +149,13 @@ class EfficientNet(nn.Module):
 
     def forward_features(self, x):
         x = self.conv_stem(x)
-        if self._act_name != "other" and isinstance(self.bn1, nn.BatchNorm2d):
+        if self._act_name != "other" and O.fusable_bn(self.bn1):
             x = O.bn_act(x, self.bn1, self._act_name)
         else:
             x = self.act1(self.bn1(x))
         x = self.blocks(x)
         x = self.conv_head(x)
-        if self._act_name != "other" and isinstance(self.bn2, nn.BatchNorm2d):
+        if self._act_name != "other" and O.fusable_bn(self.bn2):
             x = O.bn_act(x, self.bn2, self._act_name)
         else:
             x = self.act2(self.bn2(x))

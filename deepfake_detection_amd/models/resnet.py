@@ -86,10 +86,10 @@ class BasicBlock(nn.Module):
     def forward(self, x):
         residual = x
         out = self.conv1(x)
-        out = O.bn_act(out, self.bn1, "relu") if isinstance(self.bn1, nn.BatchNorm2d) \
+        out = O.bn_act(out, self.bn1, "relu") if O.fusable_bn(self.bn1) \
             else self.act1(self.bn1(out))
         out = self.conv2(out)
-        out = O.bn_act(out, self.bn2, "none") if isinstance(self.bn2, nn.BatchNorm2d) \
+        out = O.bn_act(out, self.bn2, "none") if O.fusable_bn(self.bn2) \
             else self.bn2(out)
         if self.se is not None:
             out = self.se(out)
@@ -131,13 +131,13 @@ class Bottleneck(nn.Module):
     def forward(self, x):
         residual = x
         out = self.conv1(x)
-        out = O.bn_act(out, self.bn1, "relu") if isinstance(self.bn1, nn.BatchNorm2d) \
+        out = O.bn_act(out, self.bn1, "relu") if O.fusable_bn(self.bn1) \
             else self.act1(self.bn1(out))
         out = self.conv2(out)
-        out = O.bn_act(out, self.bn2, "relu") if isinstance(self.bn2, nn.BatchNorm2d) \
+        out = O.bn_act(out, self.bn2, "relu") if O.fusable_bn(self.bn2) \
             else self.act2(self.bn2(out))
         out = self.conv3(out)
-        out = O.bn_act(out, self.bn3, "none") if isinstance(self.bn3, nn.BatchNorm2d) \
+        out = O.bn_act(out, self.bn3, "none") if O.fusable_bn(self.bn3) \
             else self.bn3(out)
         if self.se is not None:
             out = self.se(out)

@@ -14,6 +14,19 @@ from .extension import gpu_ops_required
 _FUSED_ACTS = ("none", "relu", "silu")
 
 
+def fusable_bn(bn) -> bool:
+    """True only for a plain ``nn.BatchNorm2d`` (exact type).
+
+    Subclasses (SplitBatchNorm2d, SyncBatchNorm, ...) override ``forward``
+    with semantics the fused kernel does not implement — an isinstance
+    check would silently bypass them (aux-BN stats never updated), so the
+    fused path requires the exact base type.
+    """
+    import torch.nn as nn
+
+    return type(bn) is nn.BatchNorm2d
+
+
 def act_name_of(module) -> str:
     """Map an activation module instance to a fused-kernel act name."""
     import torch.nn as nn
@@ -33,6 +46,17 @@ def bn_act(x, bn, act: str = "silu", residual=None):
     """BatchNorm2d + activation (+ optional fused residual add), fused on GPU
     (HIP kernel, NHWC, bf16 I/O, fp32 stats — SURVEY.md §2.6 items 5 and 7),
     torch ops on CPU."""
+    if not fusable_bn(bn):
+        # Subclass with its own semantics (SplitBatchNorm2d, SyncBatchNorm...):
+        # module dispatch, then the named activation.
+        y = bn(x)
+        if act == "silu":
+            y = F.silu(y)
+        elif act == "relu":
+            y = F.relu(y)
+        if residual is not None:
+            y = y + residual
+        return y
     if x.is_cuda and gpu_ops_required() and act in _FUSED_ACTS:
         from .bn_act import fused_bn_act
 

@@ -152,6 +152,8 @@ def _build_parser():
     parser.add_argument("--output", default="", type=str, metavar="PATH")
     parser.add_argument("--eval-metric", default="prec1", type=str)
     parser.add_argument("--tta", type=int, default=0)
+    parser.add_argument("--per-step-metrics", action="store_true", default=False,
+                        help="reference train.py:625-645 every-step meter semantics (syncs per step)")
     parser.add_argument("--local_rank", default=0, type=int)
     return parser
 

@@ -73,6 +73,47 @@ def test_split_batchnorm():
     assert m2(torch.randn(2, 3, 8, 8)).shape == (2, 8, 6, 6)
 
 
+def test_split_batchnorm_aux_stats_update_through_blocks():
+    """A split-BN-converted model must dispatch through SplitBatchNorm2d.forward
+    (not the fused plain-BN path): aux_bn running stats change after a
+    training step (subclass bypass via isinstance guard)."""
+    from deepfake_detection_amd.models.blocks import InvertedResidual
+
+    torch.manual_seed(0)
+    block = InvertedResidual(8, 8, exp_ratio=2.0, se_ratio=0.25, act_layer=torch.nn.SiLU)
+    m = convert_splitbn_model(block, num_splits=2)
+    assert isinstance(m.bn1, SplitBatchNorm2d)
+    before = [a.running_mean.clone() for a in
+              [m.bn1.aux_bn[0], m.bn2.aux_bn[0], m.bn3.aux_bn[0]]]
+    m.train()
+    y = m(torch.randn(4, 8, 9, 9) * 3 + 1)
+    y.sum().backward()
+    after = [a.running_mean for a in
+             [m.bn1.aux_bn[0], m.bn2.aux_bn[0], m.bn3.aux_bn[0]]]
+    for b, a in zip(before, after):
+        assert not torch.allclose(b, a), "aux BN stats did not update (fused-path bypass)"
+    # main stats must also have moved
+    assert not torch.allclose(m.bn1.running_mean, torch.zeros(16))
+
+
+def test_bn_act_subclass_routes_through_module_forward():
+    """O.bn_act on a BN subclass must call the subclass forward."""
+    from deepfake_detection_amd.ops import functional as O
+
+    class CountingBN(torch.nn.BatchNorm2d):
+        calls = 0
+
+        def forward(self, x):
+            CountingBN.calls += 1
+            return super().forward(x)
+
+    bn = CountingBN(4)
+    x = torch.randn(2, 4, 5, 5)
+    y = O.bn_act(x, bn, "relu")
+    assert CountingBN.calls == 1
+    assert torch.allclose(y, torch.relu(torch.nn.BatchNorm2d(4)(x)), atol=1e-5)
+
+
 @pytest.mark.parametrize("mod", [
     lambda: SEModule(32),
     lambda: EcaModule(32),

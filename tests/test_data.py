@@ -187,12 +187,22 @@ def test_tf_preprocessing_eval_transform():
 
     from deepfake_detection_amd.data.transforms_factory import create_transform
 
-    t = create_transform(224, is_training=False, tf_preprocessing=True)
+    import numpy as np
+
+    t = create_transform(224, is_training=False, tf_preprocessing=True,
+                         use_prefetcher=True)
     img = Image.fromarray((torch.rand(300, 260, 3) * 255).byte().numpy())
     out = t(img)
+    # reference contract (tf_preprocessing.py:219-226): uint8 CHW in [0,255],
+    # normalized later on-device by the PrefetchLoader
+    assert isinstance(out, np.ndarray)
     assert out.shape == (3, 224, 224)
-    assert out.dtype == torch.float32
-    assert 0.0 <= out.min() and out.max() <= 1.0
+    assert out.dtype == np.uint8
+    assert out.max() > 1  # not a truncated [0,1] float
+    # without the prefetcher the standard eval pipeline serves (normalizes on CPU)
+    t2 = create_transform(224, is_training=False, tf_preprocessing=True)
+    out2 = t2(img)
+    assert out2.shape == (3, 224, 224) and out2.dtype == torch.float32
 
 
 def test_legacy_dataset_and_loader_variants(tmp_path):
