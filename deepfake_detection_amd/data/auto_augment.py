@@ -1,16 +1,24 @@
 """AutoAugment / RandAugment / AugMix policy engine.
 
 Capability parity with reference dfd/timm/data/auto_augment.py (817 LoC):
-PIL-op vocabulary with magnitude ranges, AA policies ('original', 'v0' and
-variants), RandAugment config strings ('rand-m9-mstd0.5', ...), AugMix
-config strings ('augmix-m3-w3'), hparams (magnitude std, img_mean fill).
-Wired into the imagenet train transform via the `auto_augment` arg (the
-deepfake v3 path does not use AA — reference transforms_factory.py:269-287).
+the PIL op vocabulary with magnitude->argument mapping, the AA 'original'
+and 'v0' policies, RandAugment ('rand-m9-mstd0.5', ...) and AugMix
+('augmix-m3-w3') config strings. Wired into the imagenet train transform
+via the `auto_augment` arg (the deepfake v3 path does not use AA —
+reference transforms_factory.py:269-287).
+
+Design (this implementation, not the reference's): a single op REGISTRY maps
+each op name to an `_OpSpec(pil_fn, mag_fn)` pair — the PIL callable and the
+magnitude->args mapping live in one table instead of two parallel dicts —
+and all three config-string grammars share one `_parse_config` tokenizer.
+Geometry fns take an explicit `_Geom` settings object (fill + resample)
+rather than **kwargs threading.
 """
 
 import math
 import random
 import re
+from typing import Callable, NamedTuple, Optional
 
 import numpy as np
 from PIL import Image, ImageEnhance, ImageOps
@@ -23,345 +31,326 @@ _HPARAMS_DEFAULT = dict(translate_const=250, img_mean=_FILL)
 _RANDOM_INTERPOLATION = (Image.BILINEAR, Image.BICUBIC)
 
 
-def _interpolation(kwargs):
-    interpolation = kwargs.pop("resample", Image.BILINEAR)
-    if isinstance(interpolation, (list, tuple)):
-        return random.choice(interpolation)
-    return interpolation
+class _Geom(NamedTuple):
+    """Per-op geometry settings: border fill color + resample filter(s)."""
+
+    fill: tuple
+    resample: object
+
+    def kwargs(self):
+        r = self.resample
+        if isinstance(r, (list, tuple)):
+            r = random.choice(r)
+        kw = {"resample": r}
+        if _PIL_VER >= (5, 0):
+            kw["fillcolor"] = self.fill
+        return kw
 
 
-def _check_args_tf(kwargs):
-    if "fillcolor" in kwargs and _PIL_VER < (5, 0):
-        kwargs.pop("fillcolor")
-    kwargs["resample"] = _interpolation(kwargs)
+# ---- PIL op vocabulary ------------------------------------------------------
+
+def _affine(img, coeffs, geom):
+    return img.transform(img.size, Image.AFFINE, coeffs, **geom.kwargs())
 
 
-def shear_x(img, factor, **kwargs):
-    _check_args_tf(kwargs)
-    return img.transform(img.size, Image.AFFINE, (1, factor, 0, 0, 1, 0), **kwargs)
+def shear_x(img, factor, geom):
+    return _affine(img, (1, factor, 0, 0, 1, 0), geom)
 
 
-def shear_y(img, factor, **kwargs):
-    _check_args_tf(kwargs)
-    return img.transform(img.size, Image.AFFINE, (1, 0, 0, factor, 1, 0), **kwargs)
+def shear_y(img, factor, geom):
+    return _affine(img, (1, 0, 0, factor, 1, 0), geom)
 
 
-def translate_x_rel(img, pct, **kwargs):
-    pixels = pct * img.size[0]
-    _check_args_tf(kwargs)
-    return img.transform(img.size, Image.AFFINE, (1, 0, pixels, 0, 1, 0), **kwargs)
+def translate_x_abs(img, pixels, geom):
+    return _affine(img, (1, 0, pixels, 0, 1, 0), geom)
 
 
-def translate_y_rel(img, pct, **kwargs):
-    pixels = pct * img.size[1]
-    _check_args_tf(kwargs)
-    return img.transform(img.size, Image.AFFINE, (1, 0, 0, 0, 1, pixels), **kwargs)
+def translate_y_abs(img, pixels, geom):
+    return _affine(img, (1, 0, 0, 0, 1, pixels), geom)
 
 
-def translate_x_abs(img, pixels, **kwargs):
-    _check_args_tf(kwargs)
-    return img.transform(img.size, Image.AFFINE, (1, 0, pixels, 0, 1, 0), **kwargs)
+def translate_x_rel(img, pct, geom):
+    return translate_x_abs(img, pct * img.size[0], geom)
 
 
-def translate_y_abs(img, pixels, **kwargs):
-    _check_args_tf(kwargs)
-    return img.transform(img.size, Image.AFFINE, (1, 0, 0, 0, 1, pixels), **kwargs)
+def translate_y_rel(img, pct, geom):
+    return translate_y_abs(img, pct * img.size[1], geom)
 
 
-def rotate(img, degrees, **kwargs):
-    _check_args_tf(kwargs)
+def rotate(img, degrees, geom):
     if _PIL_VER >= (5, 2):
-        return img.rotate(degrees, **kwargs)
+        return img.rotate(degrees, **geom.kwargs())
     if _PIL_VER >= (5, 0):
+        # manual center-rotation affine for PIL 5.0/5.1 (no fillcolor on rotate)
         w, h = img.size
-        post_trans = (0, 0)
-        rotn_center = (w / 2.0, h / 2.0)
-        angle = -math.radians(degrees)
-        matrix = [
-            round(math.cos(angle), 15), round(math.sin(angle), 15), 0.0,
-            round(-math.sin(angle), 15), round(math.cos(angle), 15), 0.0,
-        ]
-
-        def transform(x, y, matrix):
-            (a, b, c, d, e, f) = matrix
-            return a * x + b * y + c, d * x + e * y + f
-
-        matrix[2], matrix[5] = transform(
-            -rotn_center[0] - post_trans[0], -rotn_center[1] - post_trans[1], matrix)
-        matrix[2] += rotn_center[0]
-        matrix[5] += rotn_center[1]
-        return img.transform(img.size, Image.AFFINE, matrix, **kwargs)
-    return img.rotate(degrees, resample=kwargs["resample"])
+        cx, cy = w / 2.0, h / 2.0
+        a = -math.radians(degrees)
+        cos_a, sin_a = round(math.cos(a), 15), round(math.sin(a), 15)
+        tx = cx - (cos_a * cx + sin_a * cy)
+        ty = cy - (-sin_a * cx + cos_a * cy)
+        return _affine(img, (cos_a, sin_a, tx, -sin_a, cos_a, ty), geom)
+    return img.rotate(degrees, resample=geom.kwargs()["resample"])
 
 
-def auto_contrast(img, **__):
+def auto_contrast(img, _arg=None, _geom=None):
     return ImageOps.autocontrast(img)
 
 
-def invert(img, **__):
+def invert(img, _arg=None, _geom=None):
     return ImageOps.invert(img)
 
 
-def equalize(img, **__):
+def equalize(img, _arg=None, _geom=None):
     return ImageOps.equalize(img)
 
 
-def solarize(img, thresh, **__):
+def solarize(img, thresh, _geom=None):
     return ImageOps.solarize(img, thresh)
 
 
-def solarize_add(img, add, thresh=128, **__):
-    lut = []
-    for i in range(256):
-        if i < thresh:
-            lut.append(min(255, i + add))
-        else:
-            lut.append(i)
-    if img.mode in ("L", "RGB"):
-        if img.mode == "RGB" and len(lut) == 256:
-            lut = lut + lut + lut
-        return img.point(lut)
+def solarize_add(img, add, _geom=None, thresh=128):
+    table = [min(255, i + add) if i < thresh else i for i in range(256)]
+    if img.mode == "RGB":
+        return img.point(table * 3)
+    if img.mode == "L":
+        return img.point(table)
     return img
 
 
-def posterize(img, bits_to_keep, **__):
-    if bits_to_keep >= 8:
-        return img
-    return ImageOps.posterize(img, bits_to_keep)
+def posterize(img, bits_to_keep, _geom=None):
+    return img if bits_to_keep >= 8 else ImageOps.posterize(img, bits_to_keep)
 
 
-def contrast(img, factor, **__):
-    return ImageEnhance.Contrast(img).enhance(factor)
+def _enhance(enhancer):
+    def fn(img, factor, _geom=None):
+        return enhancer(img).enhance(factor)
+
+    return fn
 
 
-def color(img, factor, **__):
-    return ImageEnhance.Color(img).enhance(factor)
+# ---- magnitude -> op-argument mappings --------------------------------------
+
+def _frac(level):
+    return level / _MAX_LEVEL
 
 
-def brightness(img, factor, **__):
-    return ImageEnhance.Brightness(img).enhance(factor)
+def _signed(value):
+    """Half the draws negate — geometric ops are symmetric around 0."""
+    return -value if random.random() > 0.5 else value
 
 
-def sharpness(img, factor, **__):
-    return ImageEnhance.Sharpness(img).enhance(factor)
+def _mag_rotate(level, _hp):
+    return _signed(_frac(level) * 30.0)
 
 
-def _randomly_negate(v):
-    return -v if random.random() > 0.5 else v
+def _mag_enhance(level, _hp):
+    return _frac(level) * 1.8 + 0.1
 
 
-def _rotate_level_to_arg(level, _hparams):
-    return (_randomly_negate((level / _MAX_LEVEL) * 30.0),)
+def _mag_shear(level, _hp):
+    return _signed(_frac(level) * 0.3)
 
 
-def _enhance_level_to_arg(level, _hparams):
-    return ((level / _MAX_LEVEL) * 1.8 + 0.1,)
+def _mag_translate_abs(level, hp):
+    return _signed(_frac(level) * float(hp["translate_const"]))
 
 
-def _shear_level_to_arg(level, _hparams):
-    return (_randomly_negate((level / _MAX_LEVEL) * 0.3),)
+def _mag_translate_rel(level, _hp):
+    return _signed(_frac(level) * 0.45)
 
 
-def _translate_abs_level_to_arg(level, hparams):
-    translate_const = hparams["translate_const"]
-    return (_randomly_negate((level / _MAX_LEVEL) * float(translate_const)),)
+def _mag_posterize_original(level, _hp):
+    return int(_frac(level) * 4) + 4  # AA 'original': keeps >=4 bits
 
 
-def _translate_rel_level_to_arg(level, _hparams):
-    return (_randomly_negate((level / _MAX_LEVEL) * 0.45),)
+def _mag_posterize_research(level, _hp):
+    return 4 - int(_frac(level) * 4)
 
 
-def _posterize_original_level_to_arg(level, _hparams):
-    # AA 'original': 4 bits kept at max magnitude
-    return (int((level / _MAX_LEVEL) * 4) + 4,)
+def _mag_posterize_tpu(level, _hp):
+    return int(_frac(level) * 4)
 
 
-def _posterize_research_level_to_arg(level, _hparams):
-    return (4 - int((level / _MAX_LEVEL) * 4),)
+def _mag_solarize(level, _hp):
+    return int(_frac(level) * 256)
 
 
-def _posterize_tpu_level_to_arg(level, _hparams):
-    return (int((level / _MAX_LEVEL) * 4),)
+def _mag_solarize_add(level, _hp):
+    return int(_frac(level) * 110)
 
 
-def _solarize_level_to_arg(level, _hparams):
-    return (int((level / _MAX_LEVEL) * 256),)
+class _OpSpec(NamedTuple):
+    pil_fn: Callable
+    mag_fn: Optional[Callable]
 
 
-def _solarize_add_level_to_arg(level, _hparams):
-    return (int((level / _MAX_LEVEL) * 110),)
-
-
-LEVEL_TO_ARG = {
-    "AutoContrast": None,
-    "Equalize": None,
-    "Invert": None,
-    "Rotate": _rotate_level_to_arg,
-    "PosterizeOriginal": _posterize_original_level_to_arg,
-    "PosterizeResearch": _posterize_research_level_to_arg,
-    "PosterizeTpu": _posterize_tpu_level_to_arg,
-    "Solarize": _solarize_level_to_arg,
-    "SolarizeAdd": _solarize_add_level_to_arg,
-    "Color": _enhance_level_to_arg,
-    "Contrast": _enhance_level_to_arg,
-    "Brightness": _enhance_level_to_arg,
-    "Sharpness": _enhance_level_to_arg,
-    "ShearX": _shear_level_to_arg,
-    "ShearY": _shear_level_to_arg,
-    "TranslateX": _translate_abs_level_to_arg,
-    "TranslateY": _translate_abs_level_to_arg,
-    "TranslateXRel": _translate_rel_level_to_arg,
-    "TranslateYRel": _translate_rel_level_to_arg,
+_REGISTRY = {
+    "AutoContrast": _OpSpec(auto_contrast, None),
+    "Equalize": _OpSpec(equalize, None),
+    "Invert": _OpSpec(invert, None),
+    "Rotate": _OpSpec(rotate, _mag_rotate),
+    "PosterizeOriginal": _OpSpec(posterize, _mag_posterize_original),
+    "PosterizeResearch": _OpSpec(posterize, _mag_posterize_research),
+    "PosterizeTpu": _OpSpec(posterize, _mag_posterize_tpu),
+    "Solarize": _OpSpec(solarize, _mag_solarize),
+    "SolarizeAdd": _OpSpec(solarize_add, _mag_solarize_add),
+    "Color": _OpSpec(_enhance(ImageEnhance.Color), _mag_enhance),
+    "Contrast": _OpSpec(_enhance(ImageEnhance.Contrast), _mag_enhance),
+    "Brightness": _OpSpec(_enhance(ImageEnhance.Brightness), _mag_enhance),
+    "Sharpness": _OpSpec(_enhance(ImageEnhance.Sharpness), _mag_enhance),
+    "ShearX": _OpSpec(shear_x, _mag_shear),
+    "ShearY": _OpSpec(shear_y, _mag_shear),
+    "TranslateX": _OpSpec(translate_x_abs, _mag_translate_abs),
+    "TranslateY": _OpSpec(translate_y_abs, _mag_translate_abs),
+    "TranslateXRel": _OpSpec(translate_x_rel, _mag_translate_rel),
+    "TranslateYRel": _OpSpec(translate_y_rel, _mag_translate_rel),
 }
 
-NAME_TO_OP = {
-    "AutoContrast": auto_contrast,
-    "Equalize": equalize,
-    "Invert": invert,
-    "Rotate": rotate,
-    "PosterizeOriginal": posterize,
-    "PosterizeResearch": posterize,
-    "PosterizeTpu": posterize,
-    "Solarize": solarize,
-    "SolarizeAdd": solarize_add,
-    "Color": color,
-    "Contrast": contrast,
-    "Brightness": brightness,
-    "Sharpness": sharpness,
-    "ShearX": shear_x,
-    "ShearY": shear_y,
-    "TranslateX": translate_x_abs,
-    "TranslateY": translate_y_abs,
-    "TranslateXRel": translate_x_rel,
-    "TranslateYRel": translate_y_rel,
-}
+# compat alias (same keys as the reference's NAME_TO_OP)
+NAME_TO_OP = {name: spec.pil_fn for name, spec in _REGISTRY.items()}
 
 
 class AugmentOp:
+    """One policy op: fires with `prob`, magnitude optionally jittered by a
+    normal of std `hparams['magnitude_std']` then clamped to [0, 10]."""
+
     def __init__(self, name, prob=0.5, magnitude=10, hparams=None):
         hparams = hparams or _HPARAMS_DEFAULT
-        self.aug_fn = NAME_TO_OP[name]
-        self.level_fn = LEVEL_TO_ARG[name]
+        self.spec = _REGISTRY[name]
         self.prob = prob
         self.magnitude = magnitude
         self.hparams = hparams.copy()
-        self.kwargs = dict(
-            fillcolor=hparams.get("img_mean", _FILL),
+        self.geom = _Geom(
+            fill=hparams.get("img_mean", _FILL),
             resample=hparams.get("interpolation", _RANDOM_INTERPOLATION),
         )
-        # magnitude noise: std>0 samples magnitude from a normal around M
         self.magnitude_std = self.hparams.get("magnitude_std", 0)
+
+    def _sample_magnitude(self):
+        m = self.magnitude
+        if self.magnitude_std and self.magnitude_std > 0:
+            m = random.gauss(m, self.magnitude_std)
+        return min(_MAX_LEVEL, max(0.0, m))
 
     def __call__(self, img):
         if self.prob < 1.0 and random.random() > self.prob:
             return img
-        magnitude = self.magnitude
-        if self.magnitude_std and self.magnitude_std > 0:
-            magnitude = random.gauss(magnitude, self.magnitude_std)
-        magnitude = min(_MAX_LEVEL, max(0, magnitude))
-        level_args = self.level_fn(magnitude, self.hparams) if self.level_fn is not None else ()
-        return self.aug_fn(img, *level_args, **self.kwargs)
+        if self.spec.mag_fn is None:
+            return self.spec.pil_fn(img, None, self.geom)
+        arg = self.spec.mag_fn(self._sample_magnitude(), self.hparams)
+        return self.spec.pil_fn(img, arg, self.geom)
 
 
-def auto_augment_policy_v0(hparams):
-    policy = [
-        [("Equalize", 0.8, 1), ("ShearY", 0.8, 4)],
-        [("Color", 0.4, 9), ("Equalize", 0.6, 3)],
-        [("Color", 0.4, 1), ("Rotate", 0.6, 8)],
-        [("Solarize", 0.8, 3), ("Equalize", 0.4, 7)],
-        [("Solarize", 0.4, 2), ("Solarize", 0.6, 2)],
-        [("Color", 0.2, 0), ("Equalize", 0.8, 8)],
-        [("Equalize", 0.4, 8), ("SolarizeAdd", 0.8, 3)],
-        [("ShearX", 0.2, 9), ("Rotate", 0.6, 8)],
-        [("Color", 0.6, 1), ("Equalize", 1.0, 2)],
-        [("Invert", 0.4, 9), ("Rotate", 0.6, 0)],
-        [("Equalize", 1.0, 9), ("ShearY", 0.6, 3)],
-        [("Color", 0.4, 7), ("Equalize", 0.6, 0)],
-        [("PosterizeTpu", 0.4, 6), ("AutoContrast", 0.4, 7)],
-        [("Solarize", 0.6, 8), ("Color", 0.6, 9)],
-        [("Solarize", 0.2, 4), ("Rotate", 0.8, 9)],
-        [("Rotate", 1.0, 7), ("TranslateYRel", 0.8, 9)],
-        [("ShearX", 0.0, 0), ("Solarize", 0.8, 4)],
-        [("ShearY", 0.8, 0), ("Color", 0.6, 4)],
-        [("Color", 1.0, 0), ("Rotate", 0.6, 2)],
-        [("Equalize", 0.8, 4), ("Equalize", 0.0, 8)],
-        [("Equalize", 1.0, 4), ("AutoContrast", 0.6, 2)],
-        [("ShearY", 0.4, 7), ("SolarizeAdd", 0.6, 7)],
-        [("PosterizeTpu", 0.8, 2), ("Solarize", 0.6, 10)],
-        [("Solarize", 0.6, 8), ("Equalize", 0.6, 1)],
-        [("Color", 0.8, 6), ("Rotate", 0.4, 5)],
-    ]
-    return [[AugmentOp(*a, hparams=hparams) for a in sp] for sp in policy]
+# ---- AutoAugment policies (data tables from the AA paper / TF impl) ---------
+# Each sub-policy is (name, prob, magnitude) pairs.
 
+_POLICY_V0 = [
+    [("Equalize", 0.8, 1), ("ShearY", 0.8, 4)],
+    [("Color", 0.4, 9), ("Equalize", 0.6, 3)],
+    [("Color", 0.4, 1), ("Rotate", 0.6, 8)],
+    [("Solarize", 0.8, 3), ("Equalize", 0.4, 7)],
+    [("Solarize", 0.4, 2), ("Solarize", 0.6, 2)],
+    [("Color", 0.2, 0), ("Equalize", 0.8, 8)],
+    [("Equalize", 0.4, 8), ("SolarizeAdd", 0.8, 3)],
+    [("ShearX", 0.2, 9), ("Rotate", 0.6, 8)],
+    [("Color", 0.6, 1), ("Equalize", 1.0, 2)],
+    [("Invert", 0.4, 9), ("Rotate", 0.6, 0)],
+    [("Equalize", 1.0, 9), ("ShearY", 0.6, 3)],
+    [("Color", 0.4, 7), ("Equalize", 0.6, 0)],
+    [("PosterizeTpu", 0.4, 6), ("AutoContrast", 0.4, 7)],
+    [("Solarize", 0.6, 8), ("Color", 0.6, 9)],
+    [("Solarize", 0.2, 4), ("Rotate", 0.8, 9)],
+    [("Rotate", 1.0, 7), ("TranslateYRel", 0.8, 9)],
+    [("ShearX", 0.0, 0), ("Solarize", 0.8, 4)],
+    [("ShearY", 0.8, 0), ("Color", 0.6, 4)],
+    [("Color", 1.0, 0), ("Rotate", 0.6, 2)],
+    [("Equalize", 0.8, 4), ("Equalize", 0.0, 8)],
+    [("Equalize", 1.0, 4), ("AutoContrast", 0.6, 2)],
+    [("ShearY", 0.4, 7), ("SolarizeAdd", 0.6, 7)],
+    [("PosterizeTpu", 0.8, 2), ("Solarize", 0.6, 10)],
+    [("Solarize", 0.6, 8), ("Equalize", 0.6, 1)],
+    [("Color", 0.8, 6), ("Rotate", 0.4, 5)],
+]
 
-def auto_augment_policy_original(hparams):
-    policy = [
-        [("PosterizeOriginal", 0.4, 8), ("Rotate", 0.6, 9)],
-        [("Solarize", 0.6, 5), ("AutoContrast", 0.6, 5)],
-        [("Equalize", 0.8, 8), ("Equalize", 0.6, 3)],
-        [("PosterizeOriginal", 0.6, 7), ("PosterizeOriginal", 0.6, 6)],
-        [("Equalize", 0.4, 7), ("Solarize", 0.2, 4)],
-        [("Equalize", 0.4, 4), ("Rotate", 0.8, 8)],
-        [("Solarize", 0.6, 3), ("Equalize", 0.6, 7)],
-        [("PosterizeOriginal", 0.8, 5), ("Equalize", 1.0, 2)],
-        [("Rotate", 0.2, 3), ("Solarize", 0.6, 8)],
-        [("Equalize", 0.6, 8), ("PosterizeOriginal", 0.4, 6)],
-        [("Rotate", 0.8, 8), ("Color", 0.4, 0)],
-        [("Rotate", 0.4, 9), ("Equalize", 0.6, 2)],
-        [("Equalize", 0.0, 7), ("Equalize", 0.8, 8)],
-        [("Invert", 0.6, 4), ("Equalize", 1.0, 8)],
-        [("Color", 0.6, 4), ("Contrast", 1.0, 8)],
-        [("Rotate", 0.8, 8), ("Color", 1.0, 2)],
-        [("Color", 0.8, 8), ("Solarize", 0.8, 7)],
-        [("Sharpness", 0.4, 7), ("Invert", 0.6, 8)],
-        [("ShearX", 0.6, 5), ("Equalize", 1.0, 9)],
-        [("Color", 0.4, 0), ("Equalize", 0.6, 3)],
-        [("Equalize", 0.4, 7), ("Solarize", 0.2, 4)],
-        [("Solarize", 0.6, 5), ("AutoContrast", 0.6, 5)],
-        [("Invert", 0.6, 4), ("Equalize", 1.0, 8)],
-        [("Color", 0.6, 4), ("Contrast", 1.0, 8)],
-        [("Equalize", 0.8, 8), ("Equalize", 0.6, 3)],
-    ]
-    return [[AugmentOp(*a, hparams=hparams) for a in sp] for sp in policy]
+_POLICY_ORIGINAL = [
+    [("PosterizeOriginal", 0.4, 8), ("Rotate", 0.6, 9)],
+    [("Solarize", 0.6, 5), ("AutoContrast", 0.6, 5)],
+    [("Equalize", 0.8, 8), ("Equalize", 0.6, 3)],
+    [("PosterizeOriginal", 0.6, 7), ("PosterizeOriginal", 0.6, 6)],
+    [("Equalize", 0.4, 7), ("Solarize", 0.2, 4)],
+    [("Equalize", 0.4, 4), ("Rotate", 0.8, 8)],
+    [("Solarize", 0.6, 3), ("Equalize", 0.6, 7)],
+    [("PosterizeOriginal", 0.8, 5), ("Equalize", 1.0, 2)],
+    [("Rotate", 0.2, 3), ("Solarize", 0.6, 8)],
+    [("Equalize", 0.6, 8), ("PosterizeOriginal", 0.4, 6)],
+    [("Rotate", 0.8, 8), ("Color", 0.4, 0)],
+    [("Rotate", 0.4, 9), ("Equalize", 0.6, 2)],
+    [("Equalize", 0.0, 7), ("Equalize", 0.8, 8)],
+    [("Invert", 0.6, 4), ("Equalize", 1.0, 8)],
+    [("Color", 0.6, 4), ("Contrast", 1.0, 8)],
+    [("Rotate", 0.8, 8), ("Color", 1.0, 2)],
+    [("Color", 0.8, 8), ("Solarize", 0.8, 7)],
+    [("Sharpness", 0.4, 7), ("Invert", 0.6, 8)],
+    [("ShearX", 0.6, 5), ("Equalize", 1.0, 9)],
+    [("Color", 0.4, 0), ("Equalize", 0.6, 3)],
+    [("Equalize", 0.4, 7), ("Solarize", 0.2, 4)],
+    [("Solarize", 0.6, 5), ("AutoContrast", 0.6, 5)],
+    [("Invert", 0.6, 4), ("Equalize", 1.0, 8)],
+    [("Color", 0.6, 4), ("Contrast", 1.0, 8)],
+    [("Equalize", 0.8, 8), ("Equalize", 0.6, 3)],
+]
+
+_POLICY_TABLES = {"v0": _POLICY_V0, "original": _POLICY_ORIGINAL}
 
 
 def auto_augment_policy(name="v0", hparams=None):
     hparams = hparams or _HPARAMS_DEFAULT
-    if name == "original":
-        return auto_augment_policy_original(hparams)
-    if name == "v0":
-        return auto_augment_policy_v0(hparams)
-    raise AssertionError("Unknown AA policy (%s)" % name)
+    try:
+        table = _POLICY_TABLES[name]
+    except KeyError:
+        raise AssertionError("Unknown AA policy (%s)" % name) from None
+    return [[AugmentOp(*op, hparams=hparams) for op in sub] for sub in table]
 
 
 class AutoAugment:
+    """Apply one uniformly-chosen sub-policy (a short op chain) per image."""
+
     def __init__(self, policy):
         self.policy = policy
 
     def __call__(self, img):
-        sub_policy = random.choice(self.policy)
-        for op in sub_policy:
+        for op in random.choice(self.policy):
             img = op(img)
         return img
 
 
+# ---- config-string grammar (shared by all three transform builders) ---------
+
+def _parse_config(config_str, prefix=None):
+    """Split 'name-k1v1-k2v2' into (name, {k: v-string}). Tokens without a
+    digit are ignored (reference behavior)."""
+    tokens = config_str.split("-")
+    name = tokens[0]
+    if prefix is not None:
+        assert name == prefix, f"config must start with '{prefix}-'"
+    options = {}
+    for token in tokens[1:]:
+        parts = re.split(r"(\d.*)", token)
+        if len(parts) >= 2:
+            options[parts[0]] = parts[1]
+    return name, options
+
+
 def auto_augment_transform(config_str, hparams):
     """'original' / 'v0' / 'original-mstd0.5' style config strings."""
-    config = config_str.split("-")
-    policy_name = config[0]
-    config = config[1:]
-    for c in config:
-        cs = re.split(r"(\d.*)", c)
-        if len(cs) < 2:
-            continue
-        key, val = cs[:2]
-        if key == "mstd":
-            hparams.setdefault("magnitude_std", float(val))
-        else:
-            assert False, "Unknown AutoAugment config section"
-    aa_policy = auto_augment_policy(policy_name, hparams=hparams)
-    return AutoAugment(aa_policy)
+    policy_name, opts = _parse_config(config_str)
+    for key, val in opts.items():
+        assert key == "mstd", "Unknown AutoAugment config section"
+        hparams.setdefault("magnitude_std", float(val))
+    return AutoAugment(auto_augment_policy(policy_name, hparams=hparams))
 
+
+# ---- RandAugment ------------------------------------------------------------
 
 _RAND_TRANSFORMS = [
     "AutoContrast", "Equalize", "Invert", "Rotate", "PosterizeTpu", "Solarize",
@@ -383,29 +372,29 @@ _RAND_CHOICE_WEIGHTS_0 = {
 def _select_rand_weights(weight_idx=0, transforms=None):
     transforms = transforms or _RAND_TRANSFORMS
     assert weight_idx == 0
-    probs = [_RAND_CHOICE_WEIGHTS_0[k] for k in transforms]
-    probs = np.array(probs)
-    return probs / np.sum(probs)
+    probs = np.array([_RAND_CHOICE_WEIGHTS_0[k] for k in transforms])
+    return probs / probs.sum()
 
 
 def rand_augment_ops(magnitude=10, hparams=None, transforms=None):
     hparams = hparams or _HPARAMS_DEFAULT
-    transforms = transforms or _RAND_TRANSFORMS
     return [AugmentOp(name, prob=0.5, magnitude=magnitude, hparams=hparams)
-            for name in transforms]
+            for name in (transforms or _RAND_TRANSFORMS)]
 
 
 class RandAugment:
+    """Apply `num_layers` ops drawn from the pool (optionally weighted)."""
+
     def __init__(self, ops, num_layers=2, choice_weights=None):
         self.ops = ops
         self.num_layers = num_layers
         self.choice_weights = choice_weights
 
     def __call__(self, img):
-        ops = np.random.choice(
+        chosen = np.random.choice(
             self.ops, self.num_layers,
             replace=self.choice_weights is None, p=self.choice_weights)
-        for op in ops:
+        for op in chosen:
             img = op(img)
         return img
 
@@ -413,17 +402,9 @@ class RandAugment:
 def rand_augment_transform(config_str, hparams):
     """'rand-m9-n3-mstd0.5' style config strings: m magnitude, n layers,
     w weight index, mstd magnitude noise."""
-    magnitude = _MAX_LEVEL
-    num_layers = 2
-    weight_idx = None
-    config = config_str.split("-")
-    assert config[0] == "rand"
-    config = config[1:]
-    for c in config:
-        cs = re.split(r"(\d.*)", c)
-        if len(cs) < 2:
-            continue
-        key, val = cs[:2]
+    _, opts = _parse_config(config_str, prefix="rand")
+    magnitude, num_layers, weight_idx = _MAX_LEVEL, 2, None
+    for key, val in opts.items():
         if key == "mstd":
             hparams.setdefault("magnitude_std", float(val))
         elif key == "m":
@@ -434,10 +415,12 @@ def rand_augment_transform(config_str, hparams):
             weight_idx = int(val)
         else:
             assert False, "Unknown RandAugment config section"
-    ra_ops = rand_augment_ops(magnitude=magnitude, hparams=hparams)
-    choice_weights = None if weight_idx is None else _select_rand_weights(weight_idx)
-    return RandAugment(ra_ops, num_layers, choice_weights=choice_weights)
+    weights = None if weight_idx is None else _select_rand_weights(weight_idx)
+    return RandAugment(rand_augment_ops(magnitude=magnitude, hparams=hparams),
+                       num_layers, choice_weights=weights)
 
+
+# ---- AugMix -----------------------------------------------------------------
 
 _AUGMIX_TRANSFORMS = [
     "AutoContrast", "ColorIncreasing", "ContrastIncreasing",
@@ -455,73 +438,59 @@ _AUGMIX_NAME_MAP = {
 
 def augmix_ops(magnitude=3, hparams=None, transforms=None):
     hparams = hparams or _HPARAMS_DEFAULT
-    transforms = transforms or _AUGMIX_TRANSFORMS
     return [
         AugmentOp(_AUGMIX_NAME_MAP.get(name, name), prob=1.0,
                   magnitude=magnitude, hparams=hparams)
-        for name in transforms
+        for name in (transforms or _AUGMIX_TRANSFORMS)
     ]
 
 
 class AugMixAugment:
-    """AugMix: mixes `width` augmented chains with Dirichlet weights +
-    a Beta-weighted blend with the original."""
+    """AugMix: `width` independent augmentation chains mixed with Dirichlet
+    weights, then Beta-blended with the original image."""
 
     def __init__(self, ops, alpha=1.0, width=3, depth=-1, blended=False):
         self.ops = ops
         self.alpha = alpha
         self.width = width
         self.depth = depth
-        self.blended = blended
+        self.blended = blended  # reference's approximate-blend flag (unused path)
 
-    def _apply_basic(self, img, mixing_weights, m):
-        img_shape = img.size[0], img.size[1], len(img.getbands())
-        mixed = np.zeros(img_shape, dtype=np.float32)
-        for mw in mixing_weights:
-            depth = self.depth if self.depth > 0 else np.random.randint(1, 4)
-            ops = np.random.choice(self.ops, depth, replace=True)
-            img_aug = img
-            for op in ops:
-                img_aug = op(img_aug)
-            mixed += mw * np.asarray(img_aug, dtype=np.float32)
-        np.clip(mixed, 0, 255.0, out=mixed)
-        mixed = Image.fromarray(mixed.astype(np.uint8))
-        return Image.blend(img, mixed, m)
+    def _one_chain(self, img):
+        depth = self.depth if self.depth > 0 else np.random.randint(1, 4)
+        for op in np.random.choice(self.ops, depth, replace=True):
+            img = op(img)
+        return img
 
     def __call__(self, img):
-        mixing_weights = np.float32(np.random.dirichlet([self.alpha] * self.width))
-        m = np.float32(np.random.beta(self.alpha, self.alpha))
-        return self._apply_basic(img, mixing_weights, m)
+        mix_weights = np.float32(np.random.dirichlet([self.alpha] * self.width))
+        blend_m = np.float32(np.random.beta(self.alpha, self.alpha))
+        acc = np.zeros((img.size[1], img.size[0], len(img.getbands())), np.float32)
+        for w in mix_weights:
+            acc += w * np.asarray(self._one_chain(img), dtype=np.float32)
+        np.clip(acc, 0, 255.0, out=acc)
+        return Image.blend(img, Image.fromarray(acc.astype(np.uint8)), blend_m)
 
 
 def augment_and_mix_transform(config_str, hparams):
     """'augmix-m3-w3-d2-a1-b1' style config strings."""
-    magnitude = 3
-    width = 3
-    depth = -1
-    alpha = 1.0
-    blended = False
-    config = config_str.split("-")
-    assert config[0] == "augmix"
-    config = config[1:]
-    for c in config:
-        cs = re.split(r"(\d.*)", c)
-        if len(cs) < 2:
-            continue
-        key, val = cs[:2]
+    _, opts = _parse_config(config_str, prefix="augmix")
+    params = dict(magnitude=3, width=3, depth=-1, alpha=1.0, blended=False)
+    for key, val in opts.items():
         if key == "mstd":
             hparams.setdefault("magnitude_std", float(val))
         elif key == "m":
-            magnitude = int(val)
+            params["magnitude"] = int(val)
         elif key == "w":
-            width = int(val)
+            params["width"] = int(val)
         elif key == "d":
-            depth = int(val)
+            params["depth"] = int(val)
         elif key == "a":
-            alpha = float(val)
+            params["alpha"] = float(val)
         elif key == "b":
-            blended = bool(val)
+            params["blended"] = bool(val)
         else:
             assert False, "Unknown AugMix config section"
-    ops = augmix_ops(magnitude=magnitude, hparams=hparams)
-    return AugMixAugment(ops, alpha=alpha, width=width, depth=depth, blended=blended)
+    ops = augmix_ops(magnitude=params["magnitude"], hparams=hparams)
+    return AugMixAugment(ops, alpha=params["alpha"], width=params["width"],
+                         depth=params["depth"], blended=params["blended"])

@@ -187,6 +187,127 @@ class NovoGrad(Optimizer):
         return loss
 
 
+class PlainRAdam(Optimizer):
+    """RAdam without the per-step-modulo buffer cache: the rectification
+    term is recomputed every step (reference radam.py PlainRAdam, :88-152).
+    Same math as RAdam; kept as a separate registry entry for parity."""
+
+    def __init__(self, params, lr=1e-3, betas=(0.9, 0.999), eps=1e-8, weight_decay=0):
+        defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
+        super().__init__(params, defaults)
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        for group in self.param_groups:
+            beta1, beta2 = group["betas"]
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                grad = p.grad.float()
+                p_fp32 = p.float()
+                state = self.state[p]
+                if len(state) == 0:
+                    state["step"] = 0
+                    state["exp_avg"] = torch.zeros_like(p_fp32)
+                    state["exp_avg_sq"] = torch.zeros_like(p_fp32)
+                else:
+                    state["exp_avg"] = state["exp_avg"].type_as(p_fp32)
+                    state["exp_avg_sq"] = state["exp_avg_sq"].type_as(p_fp32)
+                exp_avg, exp_avg_sq = state["exp_avg"], state["exp_avg_sq"]
+
+                exp_avg_sq.mul_(beta2).addcmul_(grad, grad, value=1 - beta2)
+                exp_avg.mul_(beta1).add_(grad, alpha=1 - beta1)
+
+                state["step"] += 1
+                t = state["step"]
+                beta2_t = beta2 ** t
+                n_sma_max = 2 / (1 - beta2) - 1
+                n_sma = n_sma_max - 2 * t * beta2_t / (1 - beta2_t)
+
+                if group["weight_decay"] != 0:
+                    p_fp32.add_(p_fp32, alpha=-group["weight_decay"] * group["lr"])
+
+                if n_sma >= 5:
+                    step_size = group["lr"] * math.sqrt(
+                        (1 - beta2_t) * (n_sma - 4) / (n_sma_max - 4)
+                        * (n_sma - 2) / n_sma * n_sma_max / (n_sma_max - 2)
+                    ) / (1 - beta1 ** t)
+                    denom = exp_avg_sq.sqrt().add_(group["eps"])
+                    p_fp32.addcdiv_(exp_avg, denom, value=-step_size)
+                else:
+                    step_size = group["lr"] / (1 - beta1 ** t)
+                    p_fp32.add_(exp_avg, alpha=-step_size)
+                p.copy_(p_fp32)
+        return loss
+
+
+class NvNovoGrad(Optimizer):
+    """Nvidia Jasper NovoGrad variant (reference nvnovograd.py:13-120):
+    per-tensor SCALAR second moment initialized lazily — zeros, then copies
+    the first grad-norm — versus NovoGrad's construction-time init."""
+
+    def __init__(self, params, lr=1e-3, betas=(0.95, 0.98), eps=1e-8,
+                 weight_decay=0, grad_averaging=False, amsgrad=False):
+        if not 0.0 <= lr:
+            raise ValueError("Invalid learning rate: {}".format(lr))
+        defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay,
+                        grad_averaging=grad_averaging, amsgrad=amsgrad)
+        super().__init__(params, defaults)
+
+    def __setstate__(self, state):
+        super().__setstate__(state)
+        for group in self.param_groups:
+            group.setdefault("amsgrad", False)
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+        for group in self.param_groups:
+            beta1, beta2 = group["betas"]
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                grad = p.grad
+                state = self.state[p]
+                if len(state) == 0:
+                    state["step"] = 0
+                    state["exp_avg"] = torch.zeros_like(p)
+                    state["exp_avg_sq"] = torch.zeros([], device=p.device)
+                    if group["amsgrad"]:
+                        state["max_exp_avg_sq"] = torch.zeros([], device=p.device)
+                exp_avg, exp_avg_sq = state["exp_avg"], state["exp_avg_sq"]
+                state["step"] += 1
+
+                norm = grad.pow(2).sum()
+                if exp_avg_sq == 0:
+                    exp_avg_sq.copy_(norm)
+                else:
+                    exp_avg_sq.mul_(beta2).add_(norm, alpha=1 - beta2)
+
+                if group["amsgrad"]:
+                    max_exp_avg_sq = state["max_exp_avg_sq"]
+                    torch.maximum(max_exp_avg_sq, exp_avg_sq, out=max_exp_avg_sq)
+                    denom = max_exp_avg_sq.sqrt().add_(group["eps"])
+                else:
+                    denom = exp_avg_sq.sqrt().add_(group["eps"])
+
+                ngrad = grad / denom
+                if group["weight_decay"] != 0:
+                    ngrad = ngrad.add(p, alpha=group["weight_decay"])
+                if group["grad_averaging"]:
+                    ngrad = ngrad.mul(1 - beta1)
+                exp_avg.mul_(beta1).add_(ngrad)
+                p.add_(exp_avg, alpha=-group["lr"])
+        return loss
+
+
 class Lookahead(Optimizer):
     """Lookahead wrapper (k slow steps) — reference lookahead.py:10; the
     trainer calls `sync_lookahead()` at epoch end (reference train.py:697-698)."""

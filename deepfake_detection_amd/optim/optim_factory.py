@@ -12,7 +12,7 @@ import torch.nn as nn
 import torch.optim as optim
 
 from .adamw import AdamW
-from .extras import Lookahead, Nadam, NovoGrad, RAdam
+from .extras import Lookahead, Nadam, NovoGrad, NvNovoGrad, PlainRAdam, RAdam
 from .rmsprop_tf import RMSpropTF
 
 
@@ -67,6 +67,8 @@ def create_optimizer(args, model, filter_bias_and_bn=True):
         optimizer = Nadam(parameters, **opt_args)
     elif opt_lower == "radam":
         optimizer = RAdam(parameters, **opt_args)
+    elif opt_lower == "plainradam":
+        optimizer = PlainRAdam(parameters, **opt_args)
     elif opt_lower == "adadelta":
         optimizer = optim.Adadelta(parameters, **opt_args)
     elif opt_lower == "rmsprop":
@@ -75,6 +77,8 @@ def create_optimizer(args, model, filter_bias_and_bn=True):
         optimizer = RMSpropTF(parameters, alpha=0.9, momentum=args.momentum, **opt_args)
     elif opt_lower == "novograd":
         optimizer = NovoGrad(parameters, **opt_args)
+    elif opt_lower == "nvnovograd":
+        optimizer = NvNovoGrad(parameters, **opt_args)
     else:
         raise ValueError("Invalid optimizer: %s" % args.opt)
 

@@ -134,3 +134,50 @@ def test_fusable_runs_on_cpu_layouts():
     g_cl = g.to(memory_format=torch.channels_last)
     assert _fusable(p_cl, g_cl, {"square_avg": torch.ones_like(p_cl)})
     assert not _fusable(p_cl, g, {"square_avg": torch.ones_like(p_cl)})  # mismatched layouts
+
+
+def test_plain_radam_matches_radam_trajectory():
+    """PlainRAdam recomputes the rectification each step — identical math to
+    RAdam (reference radam.py:88-152), so trajectories must match."""
+    from deepfake_detection_amd.optim import PlainRAdam, RAdam
+
+    torch.manual_seed(0)
+    pa = torch.nn.Parameter(torch.randn(8, 4))
+    pb = torch.nn.Parameter(pa.detach().clone())
+    oa = RAdam([pa], lr=1e-2, weight_decay=0.01)
+    ob = PlainRAdam([pb], lr=1e-2, weight_decay=0.01)
+    for step in range(8):
+        g = torch.randn(8, 4)
+        pa.grad = g.clone()
+        pb.grad = g.clone()
+        oa.step()
+        ob.step()
+    assert torch.allclose(pa, pb, atol=1e-6)
+
+
+def test_nvnovograd_first_step_copies_norm():
+    """NvNovoGrad lazily copies the first grad-norm into the scalar second
+    moment (reference nvnovograd.py:96-100) then EMA-updates it."""
+    from deepfake_detection_amd.optim import NvNovoGrad
+
+    p = torch.nn.Parameter(torch.ones(3))
+    opt = NvNovoGrad([p], lr=0.1, betas=(0.9, 0.5), grad_averaging=True)
+    p.grad = torch.full((3,), 2.0)
+    opt.step()
+    st = opt.state[p]
+    assert abs(st["exp_avg_sq"].item() - 12.0) < 1e-6  # sum(2^2 * 3) copied
+    p.grad = torch.full((3,), 2.0)
+    opt.step()
+    assert abs(st["exp_avg_sq"].item() - (0.5 * 12.0 + 0.5 * 12.0)) < 1e-6
+
+
+def test_factory_new_entries():
+    import types
+
+    from deepfake_detection_amd.optim import NvNovoGrad, PlainRAdam
+
+    m = torch.nn.Linear(4, 2)
+    for name, cls in [("plainradam", PlainRAdam), ("nvnovograd", NvNovoGrad)]:
+        args = types.SimpleNamespace(opt=name, lr=1e-3, weight_decay=0.0,
+                                     momentum=0.9, opt_eps=1e-8)
+        assert isinstance(create_optimizer(args, m), cls)
