@@ -18,7 +18,7 @@ import torch
 import torch.nn as nn
 
 from ..ops import functional as O
-from .layers import create_conv2d, drop_path, sigmoid
+from .layers import PointwiseConv2d, create_conv2d, drop_path, sigmoid
 
 __all__ = [
     "BN_MOMENTUM_TF_DEFAULT",
@@ -142,6 +142,7 @@ class ConvBnAct(nn.Module):
         self.bn1 = norm_layer(out_chs, **norm_kwargs)
         self.act1 = act_layer(inplace=True)
         self._act_name = O.act_name_of(self.act1)
+        _mark_bn_producer(self.conv)
 
     def feature_module(self, location):
         return "act1"
@@ -156,11 +157,32 @@ class ConvBnAct(nn.Module):
         return self.act1(self.bn1(x))
 
 
-def _bn_act(bn, act_module, act_name, x, residual=None):
+def _mark_bn_producer(conv):
+    """Ask a PointwiseConv2d to emit BN stats from its epilogue: its output
+    feeds a fused BatchNorm, which then skips its own stats pass."""
+    if isinstance(conv, PointwiseConv2d):
+        conv.emit_bn_stats = True
+
+
+def _bn_act(bn, act_module, act_name, x, residual=None, drop_path_mask=None):
     if act_name != "other" and O.fusable_bn(bn):
-        return O.bn_act(x, bn, act_name, residual)
+        return O.bn_act(x, bn, act_name, residual, drop_path_mask)
     y = act_module(bn(x))
+    if drop_path_mask is not None:
+        y = y * drop_path_mask.to(y.dtype).view(-1, 1, 1, 1)
     return y if residual is None else y + residual
+
+
+def _drop_path_mask(x, drop_prob, training):
+    """Per-sample stochastic-depth keep mask (0 or 1/keep_prob), drawn with
+    the reference's floor(keep + U[0,1)) binarization (reference
+    drop.py:84-100); fused into the block-tail BN kernel."""
+    if drop_prob <= 0.0 or not training:
+        return None
+    keep = 1.0 - drop_prob
+    mask = torch.floor(keep + torch.rand(x.shape[0], device=x.device,
+                                         dtype=torch.float32))
+    return mask.div_(keep)
 
 
 class DepthwiseSeparableConv(nn.Module):
@@ -191,6 +213,7 @@ class DepthwiseSeparableConv(nn.Module):
             self.se = None
 
         self.conv_pw = create_conv2d(in_chs, out_chs, pw_kernel_size, padding=pad_type)
+        _mark_bn_producer(self.conv_pw)
         self.bn2 = norm_layer(out_chs, **norm_kwargs)
         self.act2 = act_layer(inplace=True) if self.has_pw_act else nn.Identity()
         self._act_name = O.act_name_of(self.act1)
@@ -209,9 +232,10 @@ class DepthwiseSeparableConv(nn.Module):
         if self.se is not None:
             x = self.se(x)
         x = self.conv_pw(x)
-        if (self.has_residual and self._act2_name != "other"
-                and not (self.drop_path_rate > 0.0 and self.training) and x.is_cuda):
-            x = _bn_act(self.bn2, self.act2, self._act2_name, x, residual)
+        if self.has_residual and self._act2_name != "other" and x.is_cuda:
+            # fused BN(+act) + drop_path scale + residual in one pass
+            dp = _drop_path_mask(x, self.drop_path_rate, self.training)
+            x = _bn_act(self.bn2, self.act2, self._act2_name, x, residual, dp)
         else:
             x = _bn_act(self.bn2, self.act2, self._act2_name, x)
             if self.has_residual:
@@ -239,6 +263,7 @@ class InvertedResidual(nn.Module):
         self.drop_path_rate = drop_path_rate
 
         self.conv_pw = create_conv2d(in_chs, mid_chs, exp_kernel_size, padding=pad_type, **conv_kwargs)
+        _mark_bn_producer(self.conv_pw)
         self.bn1 = norm_layer(mid_chs, **norm_kwargs)
         self.act1 = act_layer(inplace=True)
 
@@ -255,6 +280,7 @@ class InvertedResidual(nn.Module):
             self.se = None
 
         self.conv_pwl = create_conv2d(mid_chs, out_chs, pw_kernel_size, padding=pad_type, **conv_kwargs)
+        _mark_bn_producer(self.conv_pwl)
         self.bn3 = norm_layer(out_chs, **norm_kwargs)
         self._act_name = O.act_name_of(self.act1)
 
@@ -277,10 +303,10 @@ class InvertedResidual(nn.Module):
         if self.se is not None:
             x = self.se(x)
         x = self.conv_pwl(x)
-        if (self.has_residual and not (self.drop_path_rate > 0.0 and self.training)
-                and x.is_cuda):
-            # fused BN + residual add (one pass; drop_path is identity here)
-            x = _bn_act(self.bn3, nn.Identity(), "none", x, residual)
+        if self.has_residual and x.is_cuda:
+            # fused BN + drop_path scale + residual add in one pass
+            dp = _drop_path_mask(x, self.drop_path_rate, self.training)
+            x = _bn_act(self.bn3, nn.Identity(), "none", x, residual, dp)
         else:
             x = _bn_act(self.bn3, nn.Identity(), "none", x)
             if self.has_residual:
@@ -321,10 +347,10 @@ class CondConvResidual(InvertedResidual):
         if self.se is not None:
             x = self.se(x)
         x = self.conv_pwl(x, routing_weights)
-        if (self.has_residual and not (self.drop_path_rate > 0.0 and self.training)
-                and x.is_cuda):
-            # fused BN + residual add (one pass; drop_path is identity here)
-            x = _bn_act(self.bn3, nn.Identity(), "none", x, residual)
+        if self.has_residual and x.is_cuda:
+            # fused BN + drop_path scale + residual add in one pass
+            dp = _drop_path_mask(x, self.drop_path_rate, self.training)
+            x = _bn_act(self.bn3, nn.Identity(), "none", x, residual, dp)
         else:
             x = _bn_act(self.bn3, nn.Identity(), "none", x)
             if self.has_residual:
@@ -364,6 +390,7 @@ class EdgeResidual(nn.Module):
 
         self.conv_pwl = create_conv2d(
             mid_chs, out_chs, pw_kernel_size, stride=stride, dilation=dilation, padding=pad_type)
+        _mark_bn_producer(self.conv_pwl)
         self.bn2 = norm_layer(out_chs, **norm_kwargs)
         self._act_name = O.act_name_of(self.act1)
 

@@ -25,6 +25,7 @@ __all__ = [
     "conv2d_same",
     "Conv2dSame",
     "DepthwiseConv2d",
+    "PointwiseConv2d",
     "create_conv2d_pad",
     "MixedConv2d",
     "CondConv2d",
@@ -127,6 +128,33 @@ class DepthwiseConv2d(nn.Conv2d):
         return super().forward(x)
 
 
+class PointwiseConv2d(nn.Conv2d):
+    """1x1 nn.Conv2d routed to the gfx950 MFMA GEMM kernels (ops/pwconv.py)
+    on ROCm devices — the production path for MBConv pointwise convs
+    (reference efficientnet_blocks.py:277,299). Same state_dict as nn.Conv2d.
+
+    When `emit_bn_stats` is set (by the owning block) and the module is in
+    training mode, the forward kernel also accumulates the per-channel
+    sum/sumsq the following BatchNorm needs, saving BN's full stats pass
+    over the activation."""
+
+    emit_bn_stats = False
+
+    def forward(self, x):
+        if x.is_cuda:
+            from ..ops.extension import gpu_ops_required
+            from ..ops.pwconv import pw_supported
+
+            if gpu_ops_required() and pw_supported(
+                    x, self.weight, self.stride, self.padding, self.dilation,
+                    self.groups):
+                from ..ops.pwconv import pw_conv2d
+
+                return pw_conv2d(x, self.weight, self.bias,
+                                 want_stats=self.emit_bn_stats and self.training)
+        return super().forward(x)
+
+
 def get_padding_value(padding, kernel_size, **kwargs) -> Tuple[object, bool]:
     stride = kwargs.get("stride", 1)
     dilation = kwargs.get("dilation", 1)
@@ -154,6 +182,9 @@ def create_conv2d_pad(in_chs, out_chs, kernel_size, **kwargs):
         return Conv2dSame(in_chs, out_chs, kernel_size, **kwargs)
     if (kwargs.get("groups", 1) == in_chs and in_chs == out_chs and in_chs > 1):
         return DepthwiseConv2d(in_chs, out_chs, kernel_size, padding=padding, **kwargs)
+    ks = kernel_size[0] if isinstance(kernel_size, (tuple, list)) else kernel_size
+    if ks == 1 and kwargs.get("groups", 1) == 1:
+        return PointwiseConv2d(in_chs, out_chs, kernel_size, padding=padding, **kwargs)
     return nn.Conv2d(in_chs, out_chs, kernel_size, padding=padding, **kwargs)
 
 

@@ -42,10 +42,14 @@ def act_name_of(module) -> str:
     return "other"
 
 
-def bn_act(x, bn, act: str = "silu", residual=None):
-    """BatchNorm2d + activation (+ optional fused residual add), fused on GPU
-    (HIP kernel, NHWC, bf16 I/O, fp32 stats — SURVEY.md §2.6 items 5 and 7),
-    torch ops on CPU."""
+def bn_act(x, bn, act: str = "silu", residual=None, drop_path_mask=None):
+    """BatchNorm2d + activation (+ optional fused drop_path scale and
+    residual add), fused on GPU (HIP kernel, NHWC, bf16 I/O, fp32 stats —
+    SURVEY.md §2.6 items 5 and 7), torch ops on CPU.
+
+    drop_path_mask: fp32 [B] per-sample 0-or-1/keep values (stochastic
+    depth, reference drop.py:84-100), applied to the BN+act output before
+    the residual add."""
     if not fusable_bn(bn):
         # Subclass with its own semantics (SplitBatchNorm2d, SyncBatchNorm...):
         # module dispatch, then the named activation.
@@ -54,15 +58,26 @@ def bn_act(x, bn, act: str = "silu", residual=None):
             y = F.silu(y)
         elif act == "relu":
             y = F.relu(y)
+        if drop_path_mask is not None:
+            y = y * drop_path_mask.to(y.dtype).view(-1, 1, 1, 1)
         if residual is not None:
             y = y + residual
         return y
     if x.is_cuda and gpu_ops_required() and act in _FUSED_ACTS:
         from .bn_act import fused_bn_act
 
+        # producer-fused BN stats: the conv kernel that wrote x may have
+        # attached bucketed per-channel (sum, sumsq) partials (ops/pwconv.py)
+        stats = None
+        attached = getattr(x, "_dfd_bn_stats", None)
+        if attached is not None and bn.training:
+            buckets, m, c = attached
+            if c == x.shape[1] and m == x.shape[0] * x.shape[2] * x.shape[3]:
+                stats = buckets
         return fused_bn_act(
             x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
-            bn.training, bn.momentum, bn.eps, act, residual,
+            bn.training, bn.momentum, bn.eps, act, residual, stats,
+            drop_path_mask,
         )
     y = F.batch_norm(
         x, bn.running_mean, bn.running_var, bn.weight, bn.bias,
@@ -72,6 +87,8 @@ def bn_act(x, bn, act: str = "silu", residual=None):
         y = F.silu(y)
     elif act == "relu":
         y = F.relu(y)
+    if drop_path_mask is not None:
+        y = y * drop_path_mask.to(y.dtype).view(-1, 1, 1, 1)
     if residual is not None:
         y = y + residual
     return y

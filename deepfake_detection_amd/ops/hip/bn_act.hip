@@ -155,6 +155,24 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum
 }
 
 // ---------------------------------------------------------------------------
+// bucket fold: sum the 64 bucketed partial buffers a producer kernel's stats
+// epilogue wrote ([buckets, 2, C] -> sum[C], sumsq[C])
+// ---------------------------------------------------------------------------
+__global__ void bn_fold_buckets_kernel(const float* __restrict__ buckets,
+                                       float* __restrict__ sum,
+                                       float* __restrict__ sumsq, int C, int nb) {
+  const int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float s = 0.f, q = 0.f;
+  for (int b = 0; b < nb; ++b) {
+    s += buckets[(size_t)b * 2 * C + c];
+    q += buckets[(size_t)b * 2 * C + C + c];
+  }
+  sum[c] = s;
+  sumsq[c] = q;
+}
+
+// ---------------------------------------------------------------------------
 // finalize: stats -> mean/invstd (+ running update) -> scale/shift
 // ---------------------------------------------------------------------------
 __global__ void bn_finalize_train_kernel(
@@ -208,7 +226,10 @@ template <typename T, Act ACT, int VEC, bool RES>
 __global__ void bn_act_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
                                   const float* __restrict__ scale,
                                   const float* __restrict__ shift,
-                                  const T* __restrict__ res, long long M, int C,
+                                  const T* __restrict__ res,
+                                  const float* __restrict__ dp,  // [B] per-sample
+                                  long long hw,                  // rows per sample
+                                  long long M, int C,
                                   int log2_cpb, int rows_per_chunk) {
   const int cpb = 1 << log2_cpb;
   const int slot = threadIdx.x & (cpb - 1);
@@ -238,10 +259,12 @@ __global__ void bn_act_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
     }
 #pragma unroll
     for (int u = 0; u < 2; ++u) {
+      const float keep = dp ? dp[(r + u * (long long)nrg) / hw] : 1.f;
 #pragma unroll
       for (int j = 0; j < VEC; ++j) {
         const float z = fmaf(DfdCvt<T>::to_f32(xv[u].v[j]), sc[j], sh[j]);
         float v = act_fwd(z, ACT);
+        if (dp) v *= keep;  // fused drop_path (stochastic depth) scale
         if (RES) v += DfdCvt<T>::to_f32(rv[u].v[j]);
         yv[u].v[j] = DfdCvt<T>::from_f32(v);
       }
@@ -250,11 +273,13 @@ __global__ void bn_act_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
   }
   for (; r < r1; r += nrg) {
     const BVec<T, VEC> xv = bvload<T, VEC>(x + r * C + c);
+    const float keep = dp ? dp[r / hw] : 1.f;
     BVec<T, VEC> yv;
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
       const float z = fmaf(DfdCvt<T>::to_f32(xv.v[j]), sc[j], sh[j]);
       float v = act_fwd(z, ACT);
+      if (dp) v *= keep;
       if (RES) v += DfdCvt<T>::to_f32(res[r * C + c + j]);
       yv.v[j] = DfdCvt<T>::from_f32(v);
     }
@@ -270,7 +295,8 @@ __global__ void bn_act_bwd_reduce_kernel(
     const T* __restrict__ dy, const T* __restrict__ x,
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ weight, const float* __restrict__ bias,
-    float* __restrict__ dgamma, float* __restrict__ dbeta, long long M, int C,
+    float* __restrict__ dgamma, float* __restrict__ dbeta,
+    const float* __restrict__ dp, long long hw, long long M, int C,
     int log2_cpb, int rows_per_chunk) {
   extern __shared__ float lds[];
   const int cpb = 1 << log2_cpb;
@@ -306,24 +332,29 @@ __global__ void bn_act_bwd_reduce_kernel(
         dv[u] = bvload<T, VEC>(dy + (r + u * (long long)nrg) * C + c);
       }
 #pragma unroll
-      for (int u = 0; u < 2; ++u)
+      for (int u = 0; u < 2; ++u) {
+        const float keep = dp ? dp[(r + u * (long long)nrg) / hw] : 1.f;
 #pragma unroll
         for (int j = 0; j < VEC; ++j) {
           const float xh = (DfdCvt<T>::to_f32(xv[u].v[j]) - mn[j]) * is[j];
           const float z = fmaf(ga[j], xh, be[j]);
-          const float g = DfdCvt<T>::to_f32(dv[u].v[j]) * act_bwd(z, ACT);
+          float g = DfdCvt<T>::to_f32(dv[u].v[j]) * act_bwd(z, ACT);
+          if (dp) g *= keep;
           sg[j] += g;
           sgx[j] += g * xh;
         }
+      }
     }
     for (; r < r1; r += nrg) {
       const BVec<T, VEC> xv = bvload<T, VEC>(x + r * C + c);
       const BVec<T, VEC> dv = bvload<T, VEC>(dy + r * C + c);
+      const float keep = dp ? dp[r / hw] : 1.f;
 #pragma unroll
       for (int j = 0; j < VEC; ++j) {
         const float xh = (DfdCvt<T>::to_f32(xv.v[j]) - mn[j]) * is[j];
         const float z = fmaf(ga[j], xh, be[j]);
-        const float g = DfdCvt<T>::to_f32(dv.v[j]) * act_bwd(z, ACT);
+        float g = DfdCvt<T>::to_f32(dv.v[j]) * act_bwd(z, ACT);
+        if (dp) g *= keep;
         sg[j] += g;
         sgx[j] += g * xh;
       }
@@ -365,6 +396,7 @@ __global__ void bn_act_bwd_dx_kernel(
     const float* __restrict__ mean, const float* __restrict__ invstd,
     const float* __restrict__ weight, const float* __restrict__ bias,
     const float* __restrict__ dgamma, const float* __restrict__ dbeta,
+    const float* __restrict__ dp, long long hw,
     long long M, int C, float invM, int log2_cpb, int rows_per_chunk) {
   const int cpb = 1 << log2_cpb;
   const int slot = threadIdx.x & (cpb - 1);
@@ -401,11 +433,13 @@ __global__ void bn_act_bwd_dx_kernel(
     }
 #pragma unroll
     for (int u = 0; u < 2; ++u) {
+      const float keep = dp ? dp[(r + u * (long long)nrg) / hw] : 1.f;
 #pragma unroll
       for (int j = 0; j < VEC; ++j) {
         const float xf = DfdCvt<T>::to_f32(xv[u].v[j]);
         const float z = fmaf(xf, sc[j], sh[j]);
-        const float g = DfdCvt<T>::to_f32(dv[u].v[j]) * act_bwd(z, ACT);
+        float g = DfdCvt<T>::to_f32(dv[u].v[j]) * act_bwd(z, ACT);
+        if (dp) g *= keep;
         float v = k1[j] * g;
         if (TRAIN) v = v - k2[j] - k3[j] * (xf - mn[j]);
         ov[u].v[j] = DfdCvt<T>::from_f32(v);
@@ -416,12 +450,14 @@ __global__ void bn_act_bwd_dx_kernel(
   for (; r < r1; r += nrg) {
     const BVec<T, VEC> xv = bvload<T, VEC>(x + r * C + c);
     const BVec<T, VEC> dv = bvload<T, VEC>(dy + r * C + c);
+    const float keep = dp ? dp[r / hw] : 1.f;
     BVec<T, VEC> ov;
 #pragma unroll
     for (int j = 0; j < VEC; ++j) {
       const float xf = DfdCvt<T>::to_f32(xv.v[j]);
       const float z = fmaf(xf, sc[j], sh[j]);
-      const float g = DfdCvt<T>::to_f32(dv.v[j]) * act_bwd(z, ACT);
+      float g = DfdCvt<T>::to_f32(dv.v[j]) * act_bwd(z, ACT);
+      if (dp) g *= keep;
       float v = k1[j] * g;
       if (TRAIN) v = v - k2[j] - k3[j] * (xf - mn[j]);
       ov.v[j] = DfdCvt<T>::from_f32(v);
@@ -491,7 +527,17 @@ std::vector<at::Tensor> bn_act_fwd(
     at::Tensor x, at::Tensor weight, at::Tensor bias,
     at::Tensor running_mean, at::Tensor running_var,
     bool training, double momentum, double eps, std::string act_s,
-    c10::optional<at::Tensor> residual_opt) {
+    c10::optional<at::Tensor> residual_opt,
+    c10::optional<at::Tensor> stats_opt,
+    c10::optional<at::Tensor> drop_path_opt) {
+  at::Tensor dp_t = drop_path_opt.has_value() ? *drop_path_opt : at::Tensor();
+  const float* dp_p = nullptr;
+  if (dp_t.defined()) {
+    TORCH_CHECK(dp_t.scalar_type() == at::kFloat && dp_t.is_contiguous() &&
+                    dp_t.numel() == x.size(0),
+                "bn_act_fwd: drop_path mask must be fp32 [B]");
+    dp_p = dp_t.data_ptr<float>();
+  }
   at::Tensor residual = residual_opt.has_value() ? *residual_opt : at::Tensor();
   const bool has_res = residual.defined();
   if (has_res) {
@@ -524,16 +570,32 @@ std::vector<at::Tensor> bn_act_fwd(
   const int lds = 256 * vec * sizeof(float);
 
   if (training) {
-    auto sum = at::zeros({C}, opts_f);
-    auto sumsq = at::zeros({C}, opts_f);
-    DISPATCH_DTYPE(x.scalar_type(), "bn_stats", [&] {
-      DISPATCH_VEC(vec, [&] {
-        hipLaunchKernelGGL((bn_stats_kernel<T, KVEC>), grid, dim3(256), lds, stream,
-                           (const T*)x.data_ptr(), sum.data_ptr<float>(),
-                           sumsq.data_ptr<float>(), M, C, plan.log2_cpb,
-                           plan.rows_per_chunk);
+    auto sum = at::empty({C}, opts_f);
+    auto sumsq = at::empty({C}, opts_f);
+    if (stats_opt.has_value()) {
+      // producer-fused path: the kernel that WROTE x already accumulated
+      // per-channel sum/sumsq into 64 bucketed fp32 buffers — fold them and
+      // skip the full extra read of x.
+      at::Tensor buckets = *stats_opt;
+      TORCH_CHECK(buckets.scalar_type() == at::kFloat && buckets.is_contiguous() &&
+                      buckets.numel() % (2 * (long long)C) == 0,
+                  "bn_act_fwd: stats buckets must be fp32 [nb, 2, C]");
+      const int nb = (int)(buckets.numel() / (2 * (long long)C));
+      hipLaunchKernelGGL(bn_fold_buckets_kernel, dim3((C + 255) / 256), dim3(256), 0,
+                         stream, buckets.data_ptr<float>(), sum.data_ptr<float>(),
+                         sumsq.data_ptr<float>(), C, nb);
+    } else {
+      sum.zero_();
+      sumsq.zero_();
+      DISPATCH_DTYPE(x.scalar_type(), "bn_stats", [&] {
+        DISPATCH_VEC(vec, [&] {
+          hipLaunchKernelGGL((bn_stats_kernel<T, KVEC>), grid, dim3(256), lds, stream,
+                             (const T*)x.data_ptr(), sum.data_ptr<float>(),
+                             sumsq.data_ptr<float>(), M, C, plan.log2_cpb,
+                             plan.rows_per_chunk);
+        });
       });
-    });
+    }
     hipLaunchKernelGGL(bn_finalize_train_kernel, dim3((C + 255) / 256), dim3(256), 0, stream,
                        sum.data_ptr<float>(), sumsq.data_ptr<float>(), w_p, b_p,
                        running_mean.defined() ? running_mean.data_ptr<float>() : nullptr,
@@ -551,17 +613,18 @@ std::vector<at::Tensor> bn_act_fwd(
   DISPATCH_DTYPE(x.scalar_type(), "bn_act_fwd", [&] {
     DISPATCH_ACT(act, [&] {
       DISPATCH_VEC(vec, [&] {
+        const long long hw = (long long)x.size(2) * x.size(3);
         if (has_res) {
           hipLaunchKernelGGL((bn_act_fwd_kernel<T, ACT, KVEC, true>), grid, dim3(256), 0,
                              stream, (const T*)x.data_ptr(), (T*)y.data_ptr(),
                              scale.data_ptr<float>(), shift.data_ptr<float>(),
-                             (const T*)residual.data_ptr(), M, C,
+                             (const T*)residual.data_ptr(), dp_p, hw, M, C,
                              plan.log2_cpb, plan.rows_per_chunk);
         } else {
           hipLaunchKernelGGL((bn_act_fwd_kernel<T, ACT, KVEC, false>), grid, dim3(256), 0,
                              stream, (const T*)x.data_ptr(), (T*)y.data_ptr(),
                              scale.data_ptr<float>(), shift.data_ptr<float>(),
-                             (const T*)nullptr, M, C,
+                             (const T*)nullptr, dp_p, hw, M, C,
                              plan.log2_cpb, plan.rows_per_chunk);
         }
       });
@@ -573,7 +636,17 @@ std::vector<at::Tensor> bn_act_fwd(
 // Returns {dx, dgamma, dbeta}.
 std::vector<at::Tensor> bn_act_bwd(
     at::Tensor dy, at::Tensor x, at::Tensor weight, at::Tensor bias,
-    at::Tensor save_mean, at::Tensor save_invstd, bool training, std::string act_s) {
+    at::Tensor save_mean, at::Tensor save_invstd, bool training, std::string act_s,
+    c10::optional<at::Tensor> drop_path_opt) {
+  at::Tensor dp_t = drop_path_opt.has_value() ? *drop_path_opt : at::Tensor();
+  const float* dp_p = nullptr;
+  if (dp_t.defined()) {
+    TORCH_CHECK(dp_t.scalar_type() == at::kFloat && dp_t.is_contiguous() &&
+                    dp_t.numel() == x.size(0),
+                "bn_act_bwd: drop_path mask must be fp32 [B]");
+    dp_p = dp_t.data_ptr<float>();
+  }
+  const long long hw_bn = (long long)x.size(2) * x.size(3);
   TORCH_CHECK(dy.is_cuda() && dy.dim() == 4, "bn_act_bwd: 4D CUDA tensor expected");
   const Act act = act_from_string(act_s);
   dy = dy.contiguous(at::MemoryFormat::ChannelsLast);
@@ -603,7 +676,7 @@ std::vector<at::Tensor> bn_act_bwd(
                            stream, (const T*)dy.data_ptr(), (const T*)x.data_ptr(),
                            save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),
                            w_p, b_p, dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
-                           M, C, plan.log2_cpb, plan.rows_per_chunk);
+                           dp_p, hw_bn, M, C, plan.log2_cpb, plan.rows_per_chunk);
       });
     });
   });
@@ -617,14 +690,14 @@ std::vector<at::Tensor> bn_act_bwd(
                              (T*)dx.data_ptr(), save_mean.data_ptr<float>(),
                              save_invstd.data_ptr<float>(), w_p, b_p,
                              dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
-                             M, C, invM, plan.log2_cpb, plan.rows_per_chunk);
+                             dp_p, hw_bn, M, C, invM, plan.log2_cpb, plan.rows_per_chunk);
         } else {
           hipLaunchKernelGGL((bn_act_bwd_dx_kernel<T, ACT, false, KVEC>), grid, dim3(256), 0,
                              stream, (const T*)dy.data_ptr(), (const T*)x.data_ptr(),
                              (T*)dx.data_ptr(), save_mean.data_ptr<float>(),
                              save_invstd.data_ptr<float>(), w_p, b_p,
                              dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
-                             M, C, invM, plan.log2_cpb, plan.rows_per_chunk);
+                             dp_p, hw_bn, M, C, invM, plan.log2_cpb, plan.rows_per_chunk);
         }
       });
     });

@@ -7,10 +7,13 @@
 std::vector<at::Tensor> bn_act_fwd(at::Tensor x, at::Tensor weight, at::Tensor bias,
                                    at::Tensor running_mean, at::Tensor running_var,
                                    bool training, double momentum, double eps,
-                                   std::string act, c10::optional<at::Tensor> residual);
+                                   std::string act, c10::optional<at::Tensor> residual,
+                                   c10::optional<at::Tensor> stats,
+                                   c10::optional<at::Tensor> drop_path);
 std::vector<at::Tensor> bn_act_bwd(at::Tensor dy, at::Tensor x, at::Tensor weight,
                                    at::Tensor bias, at::Tensor save_mean,
-                                   at::Tensor save_invstd, bool training, std::string act);
+                                   at::Tensor save_invstd, bool training, std::string act,
+                                   c10::optional<at::Tensor> drop_path);
 at::Tensor normalize_uint8_nhwc(at::Tensor x, at::Tensor mean, at::Tensor std,
                                 std::string dtype, bool channels_last);
 at::Tensor global_avg_pool_fwd(at::Tensor x);
@@ -37,15 +40,21 @@ at::Tensor dw_conv2d_bwd_data(at::Tensor dy, at::Tensor w_packed, int64_t H, int
                               int64_t sh, int64_t sw, int64_t ph, int64_t pw);
 at::Tensor dw_conv2d_bwd_weight(at::Tensor dy, at::Tensor x, int64_t K, int64_t sh,
                                 int64_t sw, int64_t ph, int64_t pw);
-at::Tensor pw_conv2d_fwd_mfma(at::Tensor x, at::Tensor w);
+at::Tensor pw_conv2d_fwd_mfma(at::Tensor x, at::Tensor w,
+                              c10::optional<at::Tensor> stats);
+at::Tensor pw_conv2d_bwd_weight_mfma(at::Tensor dy, at::Tensor x);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "deepfake_detection_amd gfx950 (MI355X/CDNA4) kernels";
   m.def("bn_act_fwd", &bn_act_fwd, "fused BatchNorm+act(+residual) forward (NHWC)",
         py::arg("x"), py::arg("weight"), py::arg("bias"), py::arg("running_mean"),
         py::arg("running_var"), py::arg("training"), py::arg("momentum"),
-        py::arg("eps"), py::arg("act"), py::arg("residual") = py::none());
-  m.def("bn_act_bwd", &bn_act_bwd, "fused BatchNorm+act backward (NHWC)");
+        py::arg("eps"), py::arg("act"), py::arg("residual") = py::none(),
+        py::arg("stats") = py::none(), py::arg("drop_path") = py::none());
+  m.def("bn_act_bwd", &bn_act_bwd, "fused BatchNorm+act backward (NHWC)",
+        py::arg("dy"), py::arg("x"), py::arg("weight"), py::arg("bias"),
+        py::arg("save_mean"), py::arg("save_invstd"), py::arg("training"),
+        py::arg("act"), py::arg("drop_path") = py::none());
   m.def("normalize_uint8_nhwc", &normalize_uint8_nhwc, "uint8 NCHW -> norm NHWC");
   m.def("global_avg_pool_fwd", &global_avg_pool_fwd, "global avg pool fwd (NHWC)");
   m.def("global_avg_pool_bwd", &global_avg_pool_bwd, "global avg pool bwd (NHWC)");
@@ -58,5 +67,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dw_conv2d_fwd", &dw_conv2d_fwd, "depthwise conv2d forward (NHWC)");
   m.def("dw_conv2d_bwd_data", &dw_conv2d_bwd_data, "depthwise conv2d bwd data (NHWC)");
   m.def("dw_conv2d_bwd_weight", &dw_conv2d_bwd_weight, "depthwise conv2d bwd weight (NHWC)");
-  m.def("pw_conv2d_fwd_mfma", &pw_conv2d_fwd_mfma, "1x1 conv as MFMA GEMM (experimental)");
+  m.def("pw_conv2d_fwd_mfma", &pw_conv2d_fwd_mfma, "1x1 conv as MFMA GEMM (NHWC)",
+        py::arg("x"), py::arg("w"), py::arg("stats") = py::none());
+  m.def("pw_conv2d_bwd_weight_mfma", &pw_conv2d_bwd_weight_mfma,
+        "1x1 conv weight grad: split-M MFMA + fp32 chunk reduce");
 }

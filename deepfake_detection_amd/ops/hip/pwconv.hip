@@ -1,43 +1,58 @@
-// Pointwise (1x1) convolution as an NHWC GEMM on MFMA matrix cores, gfx950.
+// Pointwise (1x1) convolution on MFMA matrix cores, NHWC, gfx950.
 //
-// out[M, N] = x[M, K] @ w[N, K]^T with M = B*H*W, K = C_in, N = C_out —
-// the torch conv weight (C_out, C_in, 1, 1) is ALREADY the [N, K]
-// B-transposed layout the kernel wants, so there is no repacking.
+// Forward:    y[M, N] = x[M, K] @ w[N, K]^T   (M = B*H*W, K = C_in, N = C_out;
+//             the torch conv weight (C_out, C_in, 1, 1) is ALREADY the [N, K]
+//             B-transposed layout the kernel wants — no repacking).
+// Bwd-data:   dx[M, K] = dy[M, N] @ w[N, K] — the SAME forward kernel with w
+//             transposed once on the host (tiny).
+// Bwd-weight: dW[N, K] = dy^T[N, M] @ x[M, K] — a tall reduction over M.
+//             Split-M two-stage: blocks own (n-tile, k-tile, m-chunk), stage
+//             dy/x tiles TRANSPOSED through LDS (contraction dim M is the
+//             row-major stride-C dim, so MFMA fragments need m-contiguous
+//             reads), write fp32 partials per chunk, then a reduce kernel
+//             sums chunks into the bf16 grad. No atomics.
 //
-// Tiling (v1, correctness-first):
-//   block = 256 threads (4 waves), block tile 128(M) x 128(N)
-//   wave  = 64x64 sub-tile = 4x4 fragments of v_mfma_f32_16x16x32_bf16
-//   K loop stages A (128x32) and B (128x32, n-major) through LDS with an
-//   8-element row pad (bank-conflict-free fragment reads, 16 B each).
+// Optional forward stats epilogue (template STATS): the kernel accumulates
+// per-channel sum/sumsq of the OUTPUT tile into 64 bucketed fp32 partial
+// buffers ([64, 2, N], bucket = blockIdx.x & 63 so contention per address is
+// grid.x/64 spread over the kernel) — the following BatchNorm consumes them
+// via bn_act_fwd(..., stats) and skips its own full read of y
+// (the single biggest measured cost in r01 profiles).
 //
 // Fragment mappings (cdna4 16x16x32 bf16):
 //   A/B: row(col) = lane & 15, k = (lane >> 4) * 8 + i   (8 bf16 / lane)
 //   C/D: col = lane & 15, row = (lane >> 4) * 4 + reg    (4 fp32 / lane)
 //
-// Experimental: wired behind DFD_AMD_PW_MFMA=1 (ops/pwconv.py); MIOpen's
-// igemm remains the default 1x1 path until this beats it per-shape.
+// Replaces MIOpen igemm for the reference's create_conv2d 1x1 call sites
+// (reference dfd/timm/models/efficientnet_blocks.py:277,299; dispatch
+// ops/pw_dispatch.py).
 
 #include <torch/extension.h>
 #include <ATen/hip/HIPContext.h>
 
 #include "common.h"
 
-#if defined(__gfx950__) || defined(__gfx942__) || !defined(__HIP_DEVICE_COMPILE__)
-
 namespace {
 
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
+constexpr int kStatsBuckets = 64;
+
+// ---------------------------------------------------------------------------
+// forward GEMM: block 256 = 4 waves; block tile 128(M) x 128(N); wave 64x64
+// ---------------------------------------------------------------------------
 constexpr int BM = 128;  // block tile M
 constexpr int BN = 128;  // block tile N
 constexpr int BK = 32;   // K step (one MFMA K)
-constexpr int LDA = BK + 8;   // padded LDS row (elements)
+constexpr int LDA = BK + 8;  // padded LDS row (elements)
 
+template <bool STATS>
 __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
     const __hip_bfloat16* __restrict__ x,  // [M, K] row-major
     const __hip_bfloat16* __restrict__ w,  // [N, K] row-major
     __hip_bfloat16* __restrict__ y,        // [M, N] row-major
+    float* __restrict__ stats,             // [kStatsBuckets, 2, N] or null
     long long M, int N, int K) {
   __shared__ __hip_bfloat16 a_lds[BM * LDA];
   __shared__ __hip_bfloat16 b_lds[BN * LDA];
@@ -80,8 +95,7 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
                 kk < K ? *reinterpret_cast<const __bf16*>(x + gm * K + kk) : (__bf16)0.f;
           }
         }
-        *reinterpret_cast<bf16x8*>(&a_lds[row * LDA + c]) =
-            *reinterpret_cast<bf16x8*>(&v);
+        *reinterpret_cast<bf16x8*>(&a_lds[row * LDA + c]) = v;
       }
     }
     // stage B: 128 n-rows x 32 k (w is [N, K] so this is a straight copy)
@@ -103,14 +117,11 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
                 kk < K ? *reinterpret_cast<const __bf16*>(w + (long long)gn * K + kk) : (__bf16)0.f;
           }
         }
-        *reinterpret_cast<bf16x8*>(&b_lds[row * LDA + c]) =
-            *reinterpret_cast<bf16x8*>(&v);
+        *reinterpret_cast<bf16x8*>(&b_lds[row * LDA + c]) = v;
       }
     }
     __syncthreads();
 
-    // two MFMA K-halves of 8 (BK=32 total... 16x16x32 consumes all 32 at once
-    // via the 8-elem fragments at lk and lk+... lk spans (lane>>4)*8 = 0..24)
     bf16x8 afrag[4], bfrag[4];
 #pragma unroll
     for (int i = 0; i < 4; ++i)
@@ -130,6 +141,12 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
   // writeback: C/D col = lane&15, row = (lane>>4)*4 + reg
   const int ccol = lane & 15;
   const int crow0 = (lane >> 4) * 4;
+  // per-lane per-column stats of the ROUNDED outputs (so the sums match what
+  // a separate pass over the stored bf16 y would produce)
+  float ssum[4], sq[4];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) { ssum[j] = 0.f; sq[j] = 0.f; }
+
 #pragma unroll
   for (int i = 0; i < 4; ++i) {
 #pragma unroll
@@ -138,19 +155,201 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
       for (int r = 0; r < 4; ++r) {
         const long long gm = m0 + wm + i * 16 + crow0 + r;
         const int gn = n0 + wn + j * 16 + ccol;
-        if (gm < M && gn < N)
-          y[gm * N + gn] = __float2bfloat16(acc[i][j][r]);
+        if (gm < M && gn < N) {
+          const __hip_bfloat16 v = __float2bfloat16(acc[i][j][r]);
+          y[gm * N + gn] = v;
+          if (STATS) {
+            const float f = __bfloat162float(v);
+            ssum[j] += f;
+            sq[j] += f * f;
+          }
+        }
+      }
+    }
+  }
+
+  if (STATS) {
+    // rows 16/32/48 fold onto lanes 0-15 (same output column)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      ssum[j] += __shfl_down(ssum[j], 32);
+      ssum[j] += __shfl_down(ssum[j], 16);
+      sq[j] += __shfl_down(sq[j], 32);
+      sq[j] += __shfl_down(sq[j], 16);
+    }
+    __syncthreads();  // a_lds rewritten as fp32 scratch below
+    float* sscr = reinterpret_cast<float*>(a_lds);        // [4 waves][64]
+    float* qscr = sscr + 4 * 64;
+    if (lane < 16) {
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        sscr[wid * 64 + j * 16 + lane] = ssum[j];
+        qscr[wid * 64 + j * 16 + lane] = sq[j];
+      }
+    }
+    __syncthreads();
+    if (tid < BN) {
+      const int col = tid;               // n-column within the block tile
+      const int half = col >> 6;         // 0: waves {0,1}; 1: waves {2,3}
+      const int sub = col & 63;
+      const int gn = n0 + col;
+      if (gn < N) {
+        const float sv = sscr[(half * 2) * 64 + sub] + sscr[(half * 2 + 1) * 64 + sub];
+        const float qv = qscr[(half * 2) * 64 + sub] + qscr[(half * 2 + 1) * 64 + sub];
+        float* bucket = stats + (size_t)(blockIdx.x & (kStatsBuckets - 1)) * 2 * N;
+        atomicAdd(bucket + gn, sv);
+        atomicAdd(bucket + N + gn, qv);
       }
     }
   }
 }
 
+// ---------------------------------------------------------------------------
+// bwd-weight: dW[N, K] = dy^T @ x, split over M into chunks of fp32 partials.
+// block 256 = 4 waves; block tile 64(N) x 64(K); wave 32x32; TM=64 per stage.
+// ---------------------------------------------------------------------------
+constexpr int WT = 64;        // block tile along N and K
+constexpr int TM = 64;        // m elements staged per iteration
+constexpr int LDM = TM + 8;   // padded LDS row (16-B aligned stride: 144 B)
+
+__global__ __launch_bounds__(256) void pw_wgrad_kernel(
+    const __hip_bfloat16* __restrict__ dy,  // [M, N]
+    const __hip_bfloat16* __restrict__ x,   // [M, K]
+    float* __restrict__ part,               // [chunks, N, K]
+    long long M, int N, int K, long long rows_per_chunk) {
+  __shared__ __bf16 dyt[WT * LDM];  // [n][m] transposed
+  __shared__ __bf16 xt[WT * LDM];   // [k][m] transposed
+
+  const int ktiles = (K + WT - 1) / WT;
+  const int n0 = (blockIdx.x / ktiles) * WT;
+  const int k0 = (blockIdx.x % ktiles) * WT;
+  const long long r0 = (long long)blockIdx.y * rows_per_chunk;
+  const long long r1 = min(r0 + rows_per_chunk, M);
+
+  const int tid = threadIdx.x;
+  const int lane = tid & (kWave - 1);
+  const int wid = tid / kWave;
+  const int wn = (wid & 1) * 32;
+  const int wk = (wid >> 1) * 32;
+  const int lrow = lane & 15;
+  const int lk = (lane >> 4) * 8;
+
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // stage thread plan: m-local = tid & 63, column quarter = tid >> 6
+  const int sm = tid & 63;
+  const int sq = tid >> 6;  // 0..3 -> 16-column group
+
+  for (long long m0 = r0; m0 < r1; m0 += TM) {
+    const long long gm = m0 + sm;
+    // dy tile: load vec8 along n, scatter-transpose into [n][m]
+    {
+      bf16x8 v[2];
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+        const int n = n0 + sq * 16 + h * 8;
+        v[h] = bf16x8{};
+        if (gm < r1 && n + 7 < N) {
+          v[h] = *reinterpret_cast<const bf16x8*>(dy + gm * N + n);
+        } else if (gm < r1) {
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            reinterpret_cast<__bf16*>(&v[h])[e] =
+                (n + e) < N ? *reinterpret_cast<const __bf16*>(dy + gm * N + n + e)
+                            : (__bf16)0.f;
+        }
+      }
+#pragma unroll
+      for (int h = 0; h < 2; ++h)
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          dyt[(sq * 16 + h * 8 + e) * LDM + sm] = reinterpret_cast<__bf16*>(&v[h])[e];
+    }
+    // x tile: same, into [k][m]
+    {
+      bf16x8 v[2];
+#pragma unroll
+      for (int h = 0; h < 2; ++h) {
+        const int k = k0 + sq * 16 + h * 8;
+        v[h] = bf16x8{};
+        if (gm < r1 && k + 7 < K) {
+          v[h] = *reinterpret_cast<const bf16x8*>(x + gm * K + k);
+        } else if (gm < r1) {
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            reinterpret_cast<__bf16*>(&v[h])[e] =
+                (k + e) < K ? *reinterpret_cast<const __bf16*>(x + gm * K + k + e)
+                            : (__bf16)0.f;
+        }
+      }
+#pragma unroll
+      for (int h = 0; h < 2; ++h)
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          xt[(sq * 16 + h * 8 + e) * LDM + sm] = reinterpret_cast<__bf16*>(&v[h])[e];
+    }
+    __syncthreads();
+
+    // 2 MFMA contraction steps of 32 m each
+#pragma unroll
+    for (int s = 0; s < 2; ++s) {
+      bf16x8 afrag[2], bfrag[2];
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+        afrag[i] = *reinterpret_cast<const bf16x8*>(
+            &dyt[(wn + i * 16 + lrow) * LDM + s * 32 + lk]);
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+        bfrag[j] = *reinterpret_cast<const bf16x8*>(
+            &xt[(wk + j * 16 + lrow) * LDM + s * 32 + lk]);
+#pragma unroll
+      for (int i = 0; i < 2; ++i)
+#pragma unroll
+        for (int j = 0; j < 2; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // partial writeback (fp32, per chunk; reduce kernel sums the chunk axis)
+  float* out = part + (size_t)blockIdx.y * N * K;
+  const int ccol = lane & 15;
+  const int crow0 = (lane >> 4) * 4;
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int gn = n0 + wn + i * 16 + crow0 + r;
+        const int gk = k0 + wk + j * 16 + ccol;
+        if (gn < N && gk < K) out[(size_t)gn * K + gk] = acc[i][j][r];
+      }
+    }
+  }
+}
+
+__global__ void pw_wgrad_reduce_kernel(const float* __restrict__ part,
+                                       __hip_bfloat16* __restrict__ dw,
+                                       long long nk, int chunks) {
+  const long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= nk) return;
+  float s = 0.f;
+  for (int c = 0; c < chunks; ++c) s += part[(size_t)c * nk + i];
+  dw[i] = __float2bfloat16(s);
+}
+
 }  // namespace
 
-#endif  // gfx950
-
-// x: (B, C_in, H, W) channels_last; w: (C_out, C_in, 1, 1). Returns NHWC y.
-at::Tensor pw_conv2d_fwd_mfma(at::Tensor x, at::Tensor w) {
+// x: (B, C_in, H, W) channels_last; w: (C_out, C_in, 1, 1). Returns NHWC y
+// (and fills `stats` [64, 2, C_out] fp32 zero-initialized, if provided).
+at::Tensor pw_conv2d_fwd_mfma(at::Tensor x, at::Tensor w,
+                              c10::optional<at::Tensor> stats_opt) {
   TORCH_CHECK(x.is_cuda() && w.is_cuda(), "pwconv: CUDA tensors required");
   TORCH_CHECK(x.scalar_type() == at::kBFloat16 && w.scalar_type() == at::kBFloat16,
               "pwconv: bf16 only");
@@ -164,12 +363,65 @@ at::Tensor pw_conv2d_fwd_mfma(at::Tensor x, at::Tensor w) {
   auto wc = w.contiguous();  // [N, K] row-major
   auto y = at::empty({x.size(0), (long long)N, x.size(2), x.size(3)},
                      x.options().memory_format(at::MemoryFormat::ChannelsLast));
-#if 1
   auto stream = at::hip::getCurrentHIPStream().stream();
-  dim3 grid((unsigned)((M + 127) / 128), (N + 127) / 128);
-  pw_gemm_bf16_kernel<<<grid, 256, 0, stream>>>(
-      (const __hip_bfloat16*)x.data_ptr(), (const __hip_bfloat16*)wc.data_ptr(),
-      (__hip_bfloat16*)y.data_ptr(), M, N, K);
-#endif
+  dim3 grid((unsigned)((M + BM - 1) / BM), (N + BN - 1) / BN);
+  if (stats_opt.has_value()) {
+    at::Tensor stats = *stats_opt;
+    TORCH_CHECK(stats.is_cuda() && stats.scalar_type() == at::kFloat &&
+                    stats.numel() == (long long)kStatsBuckets * 2 * N &&
+                    stats.is_contiguous(),
+                "pwconv: stats must be fp32 [64, 2, C_out] contiguous");
+    pw_gemm_bf16_kernel<true><<<grid, 256, 0, stream>>>(
+        (const __hip_bfloat16*)x.data_ptr(), (const __hip_bfloat16*)wc.data_ptr(),
+        (__hip_bfloat16*)y.data_ptr(), stats.data_ptr<float>(), M, N, K);
+  } else {
+    pw_gemm_bf16_kernel<false><<<grid, 256, 0, stream>>>(
+        (const __hip_bfloat16*)x.data_ptr(), (const __hip_bfloat16*)wc.data_ptr(),
+        (__hip_bfloat16*)y.data_ptr(), nullptr, M, N, K);
+  }
   return y;
+}
+
+// dy: (B, C_out, H, W) channels_last; x: (B, C_in, H, W) channels_last.
+// Returns dW (C_out, C_in, 1, 1) bf16.
+at::Tensor pw_conv2d_bwd_weight_mfma(at::Tensor dy, at::Tensor x) {
+  TORCH_CHECK(dy.is_cuda() && x.is_cuda(), "pw_wgrad: CUDA tensors required");
+  TORCH_CHECK(dy.scalar_type() == at::kBFloat16 && x.scalar_type() == at::kBFloat16,
+              "pw_wgrad: bf16 only");
+  TORCH_CHECK(dy.is_contiguous(at::MemoryFormat::ChannelsLast) &&
+                  x.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "pw_wgrad: channels_last required");
+  const long long M = (long long)x.size(0) * x.size(2) * x.size(3);
+  const int K = (int)x.size(1);
+  const int N = (int)dy.size(1);
+  TORCH_CHECK(dy.size(0) == x.size(0) && dy.size(2) == x.size(2) &&
+                  dy.size(3) == x.size(3),
+              "pw_wgrad: shape mismatch");
+
+  const int ntiles = (N + WT - 1) / WT;
+  const int ktiles = (K + WT - 1) / WT;
+  const long long kn = (long long)ntiles * ktiles;
+  // enough m-chunks to fill the chip, but >=8 stage iterations per block
+  long long chunks = kMaxGrid / kn;
+  const long long max_chunks = (M + 8 * TM - 1) / (8 * TM);
+  if (chunks > max_chunks) chunks = max_chunks;
+  if (chunks < 1) chunks = 1;
+  if (chunks > 2048) chunks = 2048;
+  const long long rows_per_chunk =
+      ((M + chunks - 1) / chunks + TM - 1) / TM * TM;  // TM-aligned
+  chunks = (M + rows_per_chunk - 1) / rows_per_chunk;
+
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  auto part = at::empty({chunks, (long long)N, (long long)K},
+                        x.options().dtype(at::kFloat));
+  dim3 grid((unsigned)kn, (unsigned)chunks);
+  pw_wgrad_kernel<<<grid, 256, 0, stream>>>(
+      (const __hip_bfloat16*)dy.data_ptr(), (const __hip_bfloat16*)x.data_ptr(),
+      part.data_ptr<float>(), M, N, K, rows_per_chunk);
+
+  auto dw = at::empty({(long long)N, (long long)K, 1, 1}, dy.options());
+  const long long nk = (long long)N * K;
+  pw_wgrad_reduce_kernel<<<dim3((unsigned)((nk + 255) / 256)), 256, 0, stream>>>(
+      part.data_ptr<float>(), (__hip_bfloat16*)dw.data_ptr(), nk, (int)chunks);
+  return dw;
 }

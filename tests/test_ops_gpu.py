@@ -387,11 +387,9 @@ def test_bn_act_fused_residual(act):
     assert x.grad is not None and torch.isfinite(x.grad).all()
 
 
-@pytest.mark.skipif(os.environ.get("DFD_AMD_PW_MFMA", "0") != "1",
-                    reason="experimental MFMA pointwise path (set DFD_AMD_PW_MFMA=1)")
 def test_pwconv_mfma_autograd():
-    """fwd + bwd-data (same kernel, transposed weight) + bwd-weight (rocBLAS)
-    vs fp32 torch conv."""
+    """fwd + bwd-data (same kernel, transposed weight) + bwd-weight (split-M
+    MFMA kernel) vs fp32 torch conv."""
     from deepfake_detection_amd.ops.pwconv import pw_conv2d
 
     torch.manual_seed(12)
@@ -425,3 +423,151 @@ def test_conv2d_same_depthwise_routes_to_hip():
         pad_same(x, [5, 5], [2, 2]), m.weight, None, 2, 0, 1, 32)
     assert y.shape == ref.shape
     assert torch.allclose(y, ref, atol=1e-4, rtol=1e-4)
+
+
+# ---------------------------------------------------------------------------
+# production pointwise conv path: bwd-weight kernel, stats epilogue, routing
+# ---------------------------------------------------------------------------
+@pytest.mark.parametrize("m,n,k", [(2 * 75 * 75, 144, 24), (4 * 19 * 19, 448, 1792),
+                                   (3 * 38 * 38, 48, 288), (2 * 10 * 10 + 3, 40, 80)])
+def test_pwconv_wgrad_kernel_matches_torch(m, n, k):
+    from deepfake_detection_amd.ops.extension import load_extension
+
+    ext = load_extension()
+    torch.manual_seed(5)
+    dy = _cl(torch.randn(1, n, 1, m, device="cuda", dtype=torch.bfloat16))
+    x = _cl(torch.randn(1, k, 1, m, device="cuda", dtype=torch.bfloat16))
+    dw = ext.pw_conv2d_bwd_weight_mfma(dy, x)
+    ref = torch.einsum("bnhw,bkhw->nk", dy.float(), x.float()).view(n, k, 1, 1)
+    tol = 0.5 + 0.02 * (m ** 0.5)
+    assert torch.allclose(dw.float(), ref, atol=tol, rtol=0.05), (
+        (dw.float() - ref).abs().max().item())
+
+
+def test_pwconv_stats_epilogue_matches_direct():
+    """The forward stats epilogue's folded per-channel sum/sumsq must match a
+    direct reduction over the stored y."""
+    from deepfake_detection_amd.ops.pwconv import pw_conv2d
+
+    torch.manual_seed(6)
+    B, K, N, H = 3, 96, 144, 23
+    x = _cl(torch.randn(B, K, H, H, device="cuda", dtype=torch.bfloat16))
+    w = torch.randn(N, K, 1, 1, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    y = pw_conv2d(x, w, want_stats=True)
+    assert hasattr(y, "_dfd_bn_stats")
+    buckets, m, c = y._dfd_bn_stats
+    assert m == B * H * H and c == N
+    s = buckets[:, 0].sum(dim=0)
+    q = buckets[:, 1].sum(dim=0)
+    ys = y.detach().float()
+    assert torch.allclose(s, ys.sum(dim=(0, 2, 3)), atol=0.5, rtol=1e-3)
+    assert torch.allclose(q, (ys * ys).sum(dim=(0, 2, 3)), atol=1.0, rtol=1e-3)
+
+
+def test_pointwise_module_routes_to_mfma_with_stats():
+    """PointwiseConv2d must run the MFMA kernel in bf16 training mode and
+    attach producer stats; the fused bn_act consumes them and produces the
+    same output/stats as the unfused stats pass."""
+    from deepfake_detection_amd.models.layers import PointwiseConv2d
+    from deepfake_detection_amd.ops import functional as O
+
+    torch.manual_seed(7)
+    conv = PointwiseConv2d(64, 128, 1, bias=False).cuda().to(torch.bfloat16)
+    conv.emit_bn_stats = True
+    bn = torch.nn.BatchNorm2d(128, momentum=0.01, eps=1e-3).cuda()
+    bn2 = torch.nn.BatchNorm2d(128, momentum=0.01, eps=1e-3).cuda()
+    x = _cl(torch.randn(4, 64, 17, 17, device="cuda", dtype=torch.bfloat16))
+
+    y = conv(x)
+    assert hasattr(y, "_dfd_bn_stats"), "MFMA+stats path not taken"
+    out_fused = O.bn_act(y, bn, "silu")
+    # unfused stats: same y pushed through bn_act WITHOUT attached stats
+    y2 = y.detach().clone()
+    out_plain = O.bn_act(y2, bn2, "silu")
+    assert torch.allclose(out_fused, out_plain, atol=2e-2, rtol=1e-2)
+    assert torch.allclose(bn.running_mean, bn2.running_mean, atol=1e-4, rtol=1e-4)
+    assert torch.allclose(bn.running_var, bn2.running_var, atol=1e-4, rtol=1e-4)
+
+
+def test_ir_block_train_step_with_stats_fusion():
+    """One InvertedResidual fwd+bwd in bf16 on the full fused path (MFMA pw,
+    producer stats, fused BN) against the CPU fp32 reference block."""
+    from deepfake_detection_amd.models.blocks import InvertedResidual
+
+    torch.manual_seed(8)
+    blk = InvertedResidual(32, 32, dw_kernel_size=3, exp_ratio=6.0,
+                           se_ratio=0.25, act_layer=torch.nn.SiLU).cuda()
+    ref = InvertedResidual(32, 32, dw_kernel_size=3, exp_ratio=6.0,
+                           se_ratio=0.25, act_layer=torch.nn.SiLU)
+    ref.load_state_dict(blk.state_dict())
+    blk.train()
+    ref.train()
+    x = torch.randn(6, 32, 19, 19)
+    xg = x.cuda().to(torch.bfloat16).contiguous(
+        memory_format=torch.channels_last).requires_grad_(True)
+    xr = x.clone().requires_grad_(True)
+    with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+        y = blk(xg)
+    yr = ref(xr)
+    assert torch.allclose(y.detach().float().cpu(), yr.detach(), atol=0.15, rtol=0.1)
+    y.float().sum().backward()
+    yr.sum().backward()
+    assert torch.allclose(xg.grad.float().cpu(), xr.grad, atol=0.3, rtol=0.1)
+    g1 = blk.conv_pw.weight.grad.float().cpu()
+    g2 = ref.conv_pw.weight.grad
+    assert torch.allclose(g1, g2, atol=0.5, rtol=0.1), (g1 - g2).abs().max().item()
+
+
+def test_bn_act_fused_drop_path():
+    """Fused per-sample drop_path scale in bn_act: forward equals
+    mask-scaled reference, backward scales dx/dgamma/dbeta and passes the
+    residual grad through unscaled."""
+    from deepfake_detection_amd.ops.bn_act import fused_bn_act
+
+    torch.manual_seed(9)
+    N, C, H, W = 6, 32, 13, 13
+    x = _cl(torch.randn(N, C, H, W, device="cuda")).requires_grad_(True)
+    res = _cl(torch.randn(N, C, H, W, device="cuda")).requires_grad_(True)
+    bn = torch.nn.BatchNorm2d(C, momentum=0.01, eps=1e-3).cuda()
+    keep = 0.75
+    mask = (torch.rand(N, device="cuda") < keep).float() / keep
+
+    y = fused_bn_act(x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
+                     True, 0.01, 1e-3, "silu", res, None, mask)
+
+    rx = x.detach().clone().requires_grad_(True)
+    rres = res.detach().clone().requires_grad_(True)
+    rm = torch.zeros(C, device="cuda")
+    rv = torch.ones(C, device="cuda")
+    g = bn.weight.detach().clone().requires_grad_(True)
+    b = bn.bias.detach().clone().requires_grad_(True)
+    ref = torch.nn.functional.silu(
+        torch.nn.functional.batch_norm(rx, rm, rv, g, b, True, 0.01, 1e-3))
+    ref = ref * mask.view(-1, 1, 1, 1) + rres
+    assert torch.allclose(y, ref, atol=5e-4, rtol=5e-4)
+
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    ref.backward(dy)
+    assert torch.allclose(res.grad, dy)
+    assert torch.allclose(x.grad, rx.grad, atol=5e-3, rtol=5e-3)
+
+
+def test_ir_block_drop_path_stays_fused_and_zeroes_samples():
+    """With drop_path_rate>0 in training the IR block must stay on the fused
+    GPU path and zero whole samples' block contribution (output == residual
+    for dropped samples)."""
+    from deepfake_detection_amd.models.blocks import InvertedResidual
+
+    torch.manual_seed(1)
+    blk = InvertedResidual(16, 16, exp_ratio=4.0, se_ratio=0.25,
+                           act_layer=torch.nn.SiLU, drop_path_rate=0.9).cuda()
+    blk.train()
+    x = _cl(torch.randn(32, 16, 9, 9, device="cuda", dtype=torch.bfloat16))
+    with torch.autocast(device_type="cuda", dtype=torch.bfloat16):
+        y = blk(x)
+    # with rate 0.9 some samples are certainly dropped: those outputs equal x
+    eq = [torch.allclose(y[i].float(), x[i].float(), atol=1e-3)
+          for i in range(32)]
+    assert any(eq), "no sample dropped at rate 0.9 (mask not applied)"
+    assert not all(eq), "all samples dropped (scale wrong)"
