@@ -186,6 +186,20 @@ def bench_pw(B=192):
         print(f"pw K={k:4d} N={n:4d} H={h:3d}: mfma {t_mfma:7.3f} miopen {t_mi:7.3f} "
               f"matmul {t_mm:7.3f} ms | mfma {gb/t_mfma*1000:5.0f} GB/s {tf/t_mfma*1000:6.1f} TF")
 
+    from deepfake_detection_amd.ops.extension import load_extension
+    ext = load_extension()
+    print(f"== pointwise conv bwd-weight (B={B}, bf16) | mfma ms | matmul ms")
+    for k, n, h in shapes:
+        x = cl(torch.randn(B, k, h, h, device="cuda", dtype=torch.bfloat16))
+        dy = cl(torch.randn(B, n, h, h, device="cuda", dtype=torch.bfloat16))
+        t_mfma = timeit(lambda: ext.pw_conv2d_bwd_weight_mfma(dy, x))
+        dy2 = dy.permute(0, 2, 3, 1).reshape(-1, n)
+        x2 = x.permute(0, 2, 3, 1).reshape(-1, k)
+        t_mm = timeit(lambda: dy2.t() @ x2)
+        gb = (x.numel() + dy.numel()) * 2 / 1e9
+        print(f"wg K={k:4d} N={n:4d} H={h:3d}: mfma {t_mfma:7.3f} matmul {t_mm:7.3f} ms "
+              f"| mfma {gb/t_mfma*1000:5.0f} GB/s")
+
 
 def main():
     p = argparse.ArgumentParser()
