@@ -40,101 +40,122 @@ using f32x4 = __attribute__((ext_vector_type(4))) float;
 constexpr int kStatsBuckets = 64;
 
 // ---------------------------------------------------------------------------
-// forward GEMM: block 256 = 4 waves; block tile 128(M) x 128(N); wave 64x64
+// forward GEMM: double-buffered LDS pipeline, BK=64 (two MFMA k-steps per
+// stage), templated block/wave tiling:
+//   big    128(M) x 128(N), 4 waves in 2x2, wave tile 64x64 (4x4 frags)
+//   skinny 128(M) x  32(N), 4 waves in 4x1, wave tile 32x32 (2x2 frags)
+//     (for the MBConv pw-linear projections: N = 24..48 would waste 3/4 of
+//      a 128-wide N tile)
+// Per iteration each wave issues its next-tile global loads into registers,
+// runs the MFMAs on the current LDS buffer, then writes the registers into
+// the other buffer — one __syncthreads per K step.
 // ---------------------------------------------------------------------------
-constexpr int BM = 128;  // block tile M
-constexpr int BN = 128;  // block tile N
-constexpr int BK = 32;   // K step (one MFMA K)
-constexpr int LDA = BK + 8;  // padded LDS row (elements)
+constexpr int BK = 64;       // staged K per iteration (2 MFMA k-steps)
+constexpr int LDK = BK + 8;  // padded LDS row stride (16-B aligned: 144 B)
 
-template <bool STATS>
+template <int ROWS>
+DFD_DEV void pw_load_tile(const __hip_bfloat16* __restrict__ src, long long row0,
+                          long long row_end, int k0, int K, int ld, int tid,
+                          bf16x8 (&v)[ROWS * 8 / 256]) {
+#pragma unroll
+  for (int u = 0; u < ROWS * 8 / 256; ++u) {
+    const int idx = tid + u * 256;
+    const int row = idx >> 3;
+    const int c = (idx & 7) * 8;
+    const long long gr = row0 + row;
+    v[u] = bf16x8{};
+    if (gr < row_end) {
+      if (k0 + c + 7 < K) {
+        v[u] = *reinterpret_cast<const bf16x8*>(src + gr * (long long)ld + k0 + c);
+      } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e) {
+          const int kk = k0 + c + e;
+          reinterpret_cast<__bf16*>(&v[u])[e] =
+              kk < K ? *reinterpret_cast<const __bf16*>(src + gr * (long long)ld + kk)
+                     : (__bf16)0.f;
+        }
+      }
+    }
+  }
+}
+
+template <int ROWS>
+DFD_DEV void pw_store_tile(__bf16* lds, int tid, const bf16x8 (&v)[ROWS * 8 / 256]) {
+#pragma unroll
+  for (int u = 0; u < ROWS * 8 / 256; ++u) {
+    const int idx = tid + u * 256;
+    *reinterpret_cast<bf16x8*>(&lds[(idx >> 3) * LDK + (idx & 7) * 8]) = v[u];
+  }
+}
+
+template <int BM, int BN, int WROWS, int WCOLS, bool STATS>
 __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
     const __hip_bfloat16* __restrict__ x,  // [M, K] row-major
     const __hip_bfloat16* __restrict__ w,  // [N, K] row-major
     __hip_bfloat16* __restrict__ y,        // [M, N] row-major
     float* __restrict__ stats,             // [kStatsBuckets, 2, N] or null
     long long M, int N, int K) {
-  __shared__ __hip_bfloat16 a_lds[BM * LDA];
-  __shared__ __hip_bfloat16 b_lds[BN * LDA];
+  constexpr int WTM = BM / WROWS;            // wave tile M
+  constexpr int WTN = BN / WCOLS;            // wave tile N
+  constexpr int FI = WTM / 16;               // row fragments per wave
+  constexpr int FJ = WTN / 16;               // col fragments per wave
+  __shared__ __bf16 a_lds[2][BM * LDK];
+  __shared__ __bf16 b_lds[2][BN * LDK];
 
   const long long m0 = (long long)blockIdx.x * BM;
   const int n0 = blockIdx.y * BN;
   const int tid = threadIdx.x;
   const int lane = tid & (kWave - 1);
   const int wid = tid / kWave;
-  // wave grid 2x2 over the 128x128 block tile
-  const int wm = (wid & 1) * 64;   // wave row offset
-  const int wn = (wid >> 1) * 64;  // wave col offset
+  const int wm = (wid % WROWS) * WTM;
+  const int wn = (wid / WROWS) * WTN;
 
-  f32x4 acc[4][4];
+  f32x4 acc[FI][FJ];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < FI; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < FJ; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  const int lrow = lane & 15;        // fragment row/col
-  const int lk = (lane >> 4) * 8;    // fragment k offset
+  const int lrow = lane & 15;      // fragment row/col
+  const int lk = (lane >> 4) * 8;  // fragment k offset within a 32-k step
 
-  for (int k0 = 0; k0 < K; k0 += BK) {
-    // stage A: 128 rows x 32 k. 256 threads x 2 vec8 loads.
-    {
-      const int r = tid >> 2;             // 0..63
-      const int c = (tid & 3) * 8;        // 0,8,16,24
-#pragma unroll
-      for (int half = 0; half < 2; ++half) {
-        const int row = r + half * 64;
-        const long long gm = m0 + row;
-        bf16x8 v = {};
-        if (gm < M && k0 + c + 7 < K) {
-          v = *reinterpret_cast<const bf16x8*>(x + gm * K + k0 + c);
-        } else if (gm < M) {
-#pragma unroll
-          for (int e = 0; e < 8; ++e) {
-            const int kk = k0 + c + e;
-            reinterpret_cast<__bf16*>(&v)[e] =
-                kk < K ? *reinterpret_cast<const __bf16*>(x + gm * K + kk) : (__bf16)0.f;
-          }
-        }
-        *reinterpret_cast<bf16x8*>(&a_lds[row * LDA + c]) = v;
-      }
+  bf16x8 av[BM * 8 / 256], bv[BN * 8 / 256];
+  pw_load_tile<BM>(x, m0, M, 0, K, K, tid, av);
+  pw_load_tile<BN>(w, n0, N, 0, K, K, tid, bv);
+  pw_store_tile<BM>(a_lds[0], tid, av);
+  pw_store_tile<BN>(b_lds[0], tid, bv);
+  __syncthreads();
+
+  int cur = 0;
+  for (int k0 = 0; k0 < K; k0 += BK, cur ^= 1) {
+    const bool has_next = k0 + BK < K;
+    if (has_next) {
+      pw_load_tile<BM>(x, m0, M, k0 + BK, K, K, tid, av);
+      pw_load_tile<BN>(w, n0, N, k0 + BK, K, K, tid, bv);
     }
-    // stage B: 128 n-rows x 32 k (w is [N, K] so this is a straight copy)
-    {
-      const int r = tid >> 2;
-      const int c = (tid & 3) * 8;
 #pragma unroll
-      for (int half = 0; half < 2; ++half) {
-        const int row = r + half * 64;
-        const int gn = n0 + row;
-        bf16x8 v = {};
-        if (gn < N && k0 + c + 7 < K) {
-          v = *reinterpret_cast<const bf16x8*>(w + (long long)gn * K + k0 + c);
-        } else if (gn < N) {
+    for (int s = 0; s < 2; ++s) {
+      bf16x8 afrag[FI], bfrag[FJ];
 #pragma unroll
-          for (int e = 0; e < 8; ++e) {
-            const int kk = k0 + c + e;
-            reinterpret_cast<__bf16*>(&v)[e] =
-                kk < K ? *reinterpret_cast<const __bf16*>(w + (long long)gn * K + kk) : (__bf16)0.f;
-          }
-        }
-        *reinterpret_cast<bf16x8*>(&b_lds[row * LDA + c]) = v;
-      }
+      for (int i = 0; i < FI; ++i)
+        afrag[i] = *reinterpret_cast<const bf16x8*>(
+            &a_lds[cur][(wm + i * 16 + lrow) * LDK + s * 32 + lk]);
+#pragma unroll
+      for (int j = 0; j < FJ; ++j)
+        bfrag[j] = *reinterpret_cast<const bf16x8*>(
+            &b_lds[cur][(wn + j * 16 + lrow) * LDK + s * 32 + lk]);
+#pragma unroll
+      for (int i = 0; i < FI; ++i)
+#pragma unroll
+        for (int j = 0; j < FJ; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
     }
-    __syncthreads();
-
-    bf16x8 afrag[4], bfrag[4];
-#pragma unroll
-    for (int i = 0; i < 4; ++i)
-      afrag[i] = *reinterpret_cast<const bf16x8*>(&a_lds[(wm + i * 16 + lrow) * LDA + lk]);
-#pragma unroll
-    for (int j = 0; j < 4; ++j)
-      bfrag[j] = *reinterpret_cast<const bf16x8*>(&b_lds[(wn + j * 16 + lrow) * LDA + lk]);
-#pragma unroll
-    for (int i = 0; i < 4; ++i)
-#pragma unroll
-      for (int j = 0; j < 4; ++j)
-        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-            afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+    if (has_next) {
+      pw_store_tile<BM>(a_lds[cur ^ 1], tid, av);
+      pw_store_tile<BN>(b_lds[cur ^ 1], tid, bv);
+    }
     __syncthreads();
   }
 
@@ -143,14 +164,14 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
   const int crow0 = (lane >> 4) * 4;
   // per-lane per-column stats of the ROUNDED outputs (so the sums match what
   // a separate pass over the stored bf16 y would produce)
-  float ssum[4], sq[4];
+  float ssum[FJ], sq[FJ];
 #pragma unroll
-  for (int j = 0; j < 4; ++j) { ssum[j] = 0.f; sq[j] = 0.f; }
+  for (int j = 0; j < FJ; ++j) { ssum[j] = 0.f; sq[j] = 0.f; }
 
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
+  for (int i = 0; i < FI; ++i) {
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
+    for (int j = 0; j < FJ; ++j) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const long long gm = m0 + wm + i * 16 + crow0 + r;
@@ -171,31 +192,35 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
   if (STATS) {
     // rows 16/32/48 fold onto lanes 0-15 (same output column)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
+    for (int j = 0; j < FJ; ++j) {
       ssum[j] += __shfl_down(ssum[j], 32);
       ssum[j] += __shfl_down(ssum[j], 16);
       sq[j] += __shfl_down(sq[j], 32);
       sq[j] += __shfl_down(sq[j], 16);
     }
-    __syncthreads();  // a_lds rewritten as fp32 scratch below
-    float* sscr = reinterpret_cast<float*>(a_lds);        // [4 waves][64]
-    float* qscr = sscr + 4 * 64;
+    __syncthreads();  // LDS rewritten as fp32 scratch below
+    float* sscr = reinterpret_cast<float*>(&a_lds[0][0]);  // [4 waves][WTN]
+    float* qscr = sscr + 4 * WTN;
     if (lane < 16) {
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        sscr[wid * 64 + j * 16 + lane] = ssum[j];
-        qscr[wid * 64 + j * 16 + lane] = sq[j];
+      for (int j = 0; j < FJ; ++j) {
+        sscr[wid * WTN + j * 16 + lane] = ssum[j];
+        qscr[wid * WTN + j * 16 + lane] = sq[j];
       }
     }
     __syncthreads();
     if (tid < BN) {
-      const int col = tid;               // n-column within the block tile
-      const int half = col >> 6;         // 0: waves {0,1}; 1: waves {2,3}
-      const int sub = col & 63;
+      const int col = tid;                 // n-column within the block tile
+      const int q = col / WTN;             // column-group index
+      const int sub = col % WTN;
       const int gn = n0 + col;
       if (gn < N) {
-        const float sv = sscr[(half * 2) * 64 + sub] + sscr[(half * 2 + 1) * 64 + sub];
-        const float qv = qscr[(half * 2) * 64 + sub] + qscr[(half * 2 + 1) * 64 + sub];
+        float sv = 0.f, qv = 0.f;
+#pragma unroll
+      for (int rr = 0; rr < WROWS; ++rr) {
+          sv += sscr[(q * WROWS + rr) * WTN + sub];
+          qv += qscr[(q * WROWS + rr) * WTN + sub];
+        }
         float* bucket = stats + (size_t)(blockIdx.x & (kStatsBuckets - 1)) * 2 * N;
         atomicAdd(bucket + gn, sv);
         atomicAdd(bucket + N + gn, qv);
@@ -204,6 +229,7 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
   }
 }
 
+constexpr int kPwBM = 128;
 // ---------------------------------------------------------------------------
 // bwd-weight: dW[N, K] = dy^T @ x, split over M into chunks of fp32 partials.
 // block 256 = 4 waves; block tile 64(N) x 64(K); wave 32x32; TM=64 per stage.
@@ -364,20 +390,35 @@ at::Tensor pw_conv2d_fwd_mfma(at::Tensor x, at::Tensor w,
   auto y = at::empty({x.size(0), (long long)N, x.size(2), x.size(3)},
                      x.options().memory_format(at::MemoryFormat::ChannelsLast));
   auto stream = at::hip::getCurrentHIPStream().stream();
-  dim3 grid((unsigned)((M + BM - 1) / BM), (N + BN - 1) / BN);
+  float* stats_p = nullptr;
   if (stats_opt.has_value()) {
     at::Tensor stats = *stats_opt;
     TORCH_CHECK(stats.is_cuda() && stats.scalar_type() == at::kFloat &&
                     stats.numel() == (long long)kStatsBuckets * 2 * N &&
                     stats.is_contiguous(),
                 "pwconv: stats must be fp32 [64, 2, C_out] contiguous");
-    pw_gemm_bf16_kernel<true><<<grid, 256, 0, stream>>>(
-        (const __hip_bfloat16*)x.data_ptr(), (const __hip_bfloat16*)wc.data_ptr(),
-        (__hip_bfloat16*)y.data_ptr(), stats.data_ptr<float>(), M, N, K);
+    stats_p = stats.data_ptr<float>();
+  }
+  const auto* xp = (const __hip_bfloat16*)x.data_ptr();
+  const auto* wp = (const __hip_bfloat16*)wc.data_ptr();
+  auto* yp = (__hip_bfloat16*)y.data_ptr();
+  // skinny-N config for the MBConv pw-linear projections (N = 24..48)
+  if (N <= 64) {
+    dim3 grid((unsigned)((M + kPwBM - 1) / kPwBM), (N + 31) / 32);
+    if (stats_p)
+      pw_gemm_bf16_kernel<128, 32, 4, 1, true><<<grid, 256, 0, stream>>>(
+          xp, wp, yp, stats_p, M, N, K);
+    else
+      pw_gemm_bf16_kernel<128, 32, 4, 1, false><<<grid, 256, 0, stream>>>(
+          xp, wp, yp, nullptr, M, N, K);
   } else {
-    pw_gemm_bf16_kernel<false><<<grid, 256, 0, stream>>>(
-        (const __hip_bfloat16*)x.data_ptr(), (const __hip_bfloat16*)wc.data_ptr(),
-        (__hip_bfloat16*)y.data_ptr(), nullptr, M, N, K);
+    dim3 grid((unsigned)((M + kPwBM - 1) / kPwBM), (N + 127) / 128);
+    if (stats_p)
+      pw_gemm_bf16_kernel<128, 128, 2, 2, true><<<grid, 256, 0, stream>>>(
+          xp, wp, yp, stats_p, M, N, K);
+    else
+      pw_gemm_bf16_kernel<128, 128, 2, 2, false><<<grid, 256, 0, stream>>>(
+          xp, wp, yp, nullptr, M, N, K);
   }
   return y;
 }
