@@ -65,7 +65,9 @@ DFD_DEV void pw_load_tile(const __hip_bfloat16* __restrict__ src, long long row0
     const long long gr = row0 + row;
     v[u] = bf16x8{};
     if (gr < row_end) {
-      if (k0 + c + 7 < K) {
+      // vec path needs 16-B alignment: row base gr*ld is only aligned when
+      // ld (=K) is a multiple of 8 elements
+      if ((ld & 7) == 0 && k0 + c + 7 < K) {
         v[u] = *reinterpret_cast<const bf16x8*>(src + gr * (long long)ld + k0 + c);
       } else {
 #pragma unroll
@@ -159,37 +161,31 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
     __syncthreads();
   }
 
-  // writeback: C/D col = lane&15, row = (lane>>4)*4 + reg
+  // ---- epilogue ----
+  // C/D fragment: col = lane&15, row = (lane>>4)*4 + reg.
   const int ccol = lane & 15;
   const int crow0 = (lane >> 4) * 4;
-  // per-lane per-column stats of the ROUNDED outputs (so the sums match what
-  // a separate pass over the stored bf16 y would produce)
-  float ssum[FJ], sq[FJ];
-#pragma unroll
-  for (int j = 0; j < FJ; ++j) { ssum[j] = 0.f; sq[j] = 0.f; }
 
+  if (STATS) {
+    // per-channel sum/sumsq of the ROUNDED outputs (matches a separate pass
+    // over the stored bf16 y)
+    float ssum[FJ], sq[FJ];
 #pragma unroll
-  for (int i = 0; i < FI; ++i) {
+    for (int j = 0; j < FJ; ++j) { ssum[j] = 0.f; sq[j] = 0.f; }
 #pragma unroll
-    for (int j = 0; j < FJ; ++j) {
+    for (int i = 0; i < FI; ++i)
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const long long gm = m0 + wm + i * 16 + crow0 + r;
-        const int gn = n0 + wn + j * 16 + ccol;
-        if (gm < M && gn < N) {
-          const __hip_bfloat16 v = __float2bfloat16(acc[i][j][r]);
-          y[gm * N + gn] = v;
-          if (STATS) {
-            const float f = __bfloat162float(v);
+      for (int j = 0; j < FJ; ++j)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const long long gm = m0 + wm + i * 16 + crow0 + r;
+          const int gn = n0 + wn + j * 16 + ccol;
+          if (gm < M && gn < N) {
+            const float f = __bfloat162float(__float2bfloat16(acc[i][j][r]));
             ssum[j] += f;
             sq[j] += f * f;
           }
         }
-      }
-    }
-  }
-
-  if (STATS) {
     // rows 16/32/48 fold onto lanes 0-15 (same output column)
 #pragma unroll
     for (int j = 0; j < FJ; ++j) {
@@ -217,13 +213,50 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
       if (gn < N) {
         float sv = 0.f, qv = 0.f;
 #pragma unroll
-      for (int rr = 0; rr < WROWS; ++rr) {
+        for (int rr = 0; rr < WROWS; ++rr) {
           sv += sscr[(q * WROWS + rr) * WTN + sub];
           qv += qscr[(q * WROWS + rr) * WTN + sub];
         }
         float* bucket = stats + (size_t)(blockIdx.x & (kStatsBuckets - 1)) * 2 * N;
         atomicAdd(bucket + gn, sv);
         atomicAdd(bucket + N + gn, qv);
+      }
+    }
+    __syncthreads();  // before the writeback re-uses the LDS
+  }
+
+  // Vectorized writeback: per-lane scalar b16 stores are store-ISSUE-bound
+  // (cdna_hip_programming.md T21), so each wave round-trips its tile through
+  // LDS and issues 16-B row-major stores instead.
+  {
+    __bf16* ctile = &a_lds[0][0] + wid * (WTM * WTN);  // 2*BM*LDK >= 4*WTM*WTN
+#pragma unroll
+    for (int i = 0; i < FI; ++i)
+#pragma unroll
+      for (int j = 0; j < FJ; ++j)
+#pragma unroll
+        for (int r = 0; r < 4; ++r)
+          ctile[(i * 16 + crow0 + r) * WTN + j * 16 + ccol] =
+              (__bf16)__float2bfloat16(acc[i][j][r]);
+    __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt: wave's own LDS stores landed
+    constexpr int CU = WTM * WTN / 64 / 8;  // vec8 units per lane
+    const bool full_cols = (n0 + wn + WTN) <= N && (N & 7) == 0;
+#pragma unroll
+    for (int u = 0; u < CU; ++u) {
+      const int unit = lane + u * 64;
+      const int row = unit / (WTN / 8);
+      const int c8 = (unit % (WTN / 8)) * 8;
+      const long long gm = m0 + wm + row;
+      if (gm >= M) continue;
+      const int gn = n0 + wn + c8;
+      if (full_cols) {
+        *reinterpret_cast<bf16x8*>(y + gm * N + gn) =
+            *reinterpret_cast<const bf16x8*>(&ctile[row * WTN + c8]);
+      } else {
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          if (gn + e < N)
+            *reinterpret_cast<__bf16*>(y + gm * N + gn + e) = ctile[row * WTN + c8 + e];
       }
     }
   }
@@ -279,7 +312,7 @@ __global__ __launch_bounds__(256) void pw_wgrad_kernel(
       for (int h = 0; h < 2; ++h) {
         const int n = n0 + sq * 16 + h * 8;
         v[h] = bf16x8{};
-        if (gm < r1 && n + 7 < N) {
+        if (gm < r1 && (N & 7) == 0 && n + 7 < N) {
           v[h] = *reinterpret_cast<const bf16x8*>(dy + gm * N + n);
         } else if (gm < r1) {
 #pragma unroll
@@ -302,7 +335,7 @@ __global__ __launch_bounds__(256) void pw_wgrad_kernel(
       for (int h = 0; h < 2; ++h) {
         const int k = k0 + sq * 16 + h * 8;
         v[h] = bf16x8{};
-        if (gm < r1 && k + 7 < K) {
+        if (gm < r1 && (K & 7) == 0 && k + 7 < K) {
           v[h] = *reinterpret_cast<const bf16x8*>(x + gm * K + k);
         } else if (gm < r1) {
 #pragma unroll
@@ -402,14 +435,24 @@ at::Tensor pw_conv2d_fwd_mfma(at::Tensor x, at::Tensor w,
   const auto* xp = (const __hip_bfloat16*)x.data_ptr();
   const auto* wp = (const __hip_bfloat16*)wc.data_ptr();
   auto* yp = (__hip_bfloat16*)y.data_ptr();
-  // skinny-N config for the MBConv pw-linear projections (N = 24..48)
-  if (N <= 64) {
+  // N-tile config: smallest width that doesn't add extra column tiles
+  // (every extra column tile re-reads the whole x) — 32 for the pw-linear
+  // projections (N<=32), 64 for N<=64, 128 otherwise.
+  if (N <= 32) {
     dim3 grid((unsigned)((M + kPwBM - 1) / kPwBM), (N + 31) / 32);
     if (stats_p)
       pw_gemm_bf16_kernel<128, 32, 4, 1, true><<<grid, 256, 0, stream>>>(
           xp, wp, yp, stats_p, M, N, K);
     else
       pw_gemm_bf16_kernel<128, 32, 4, 1, false><<<grid, 256, 0, stream>>>(
+          xp, wp, yp, nullptr, M, N, K);
+  } else if (N <= 64) {
+    dim3 grid((unsigned)((M + kPwBM - 1) / kPwBM), (N + 63) / 64);
+    if (stats_p)
+      pw_gemm_bf16_kernel<128, 64, 2, 2, true><<<grid, 256, 0, stream>>>(
+          xp, wp, yp, stats_p, M, N, K);
+    else
+      pw_gemm_bf16_kernel<128, 64, 2, 2, false><<<grid, 256, 0, stream>>>(
           xp, wp, yp, nullptr, M, N, K);
   } else {
     dim3 grid((unsigned)((M + kPwBM - 1) / kPwBM), (N + 127) / 128);
