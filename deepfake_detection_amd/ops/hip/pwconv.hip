@@ -91,7 +91,9 @@ DFD_DEV void pw_store_tile(__bf16* lds, int tid, const bf16x8 (&v)[ROWS * 8 / 25
   }
 }
 
-template <int BM, int BN, int WROWS, int WCOLS, bool STATS>
+// NBUF: LDS pipeline depth — 1 for single-iteration shapes (K <= BK), where
+// the 2nd buffer would only halve occupancy, 2 (double-buffered) otherwise.
+template <int BM, int BN, int WROWS, int WCOLS, int NBUF, bool STATS>
 __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
     const __hip_bfloat16* __restrict__ x,  // [M, K] row-major
     const __hip_bfloat16* __restrict__ w,  // [N, K] row-major
@@ -102,8 +104,11 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
   constexpr int WTN = BN / WCOLS;            // wave tile N
   constexpr int FI = WTM / 16;               // row fragments per wave
   constexpr int FJ = WTN / 16;               // col fragments per wave
-  __shared__ __bf16 a_lds[2][BM * LDK];
-  __shared__ __bf16 b_lds[2][BN * LDK];
+  static_assert(NBUF * (BM + BN) * LDK >= 4 * WTM * WTN * WROWS * WCOLS / 4,
+                "smem too small for the writeback staging tile");
+  __shared__ __bf16 smem[NBUF * (BM + BN) * LDK];
+  auto a_buf = [&](int b) { return &smem[b * BM * LDK]; };
+  auto b_buf = [&](int b) { return &smem[NBUF * BM * LDK + b * BN * LDK]; };
 
   const long long m0 = (long long)blockIdx.x * BM;
   const int n0 = blockIdx.y * BN;
@@ -125,12 +130,12 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
   bf16x8 av[BM * 8 / 256], bv[BN * 8 / 256];
   pw_load_tile<BM>(x, m0, M, 0, K, K, tid, av);
   pw_load_tile<BN>(w, n0, N, 0, K, K, tid, bv);
-  pw_store_tile<BM>(a_lds[0], tid, av);
-  pw_store_tile<BN>(b_lds[0], tid, bv);
+  pw_store_tile<BM>(a_buf(0), tid, av);
+  pw_store_tile<BN>(b_buf(0), tid, bv);
   __syncthreads();
 
   int cur = 0;
-  for (int k0 = 0; k0 < K; k0 += BK, cur ^= 1) {
+  for (int k0 = 0; k0 < K; k0 += BK, cur ^= (NBUF - 1)) {
     const bool has_next = k0 + BK < K;
     if (has_next) {
       pw_load_tile<BM>(x, m0, M, k0 + BK, K, K, tid, av);
@@ -142,11 +147,11 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
 #pragma unroll
       for (int i = 0; i < FI; ++i)
         afrag[i] = *reinterpret_cast<const bf16x8*>(
-            &a_lds[cur][(wm + i * 16 + lrow) * LDK + s * 32 + lk]);
+            &a_buf(cur)[(wm + i * 16 + lrow) * LDK + s * 32 + lk]);
 #pragma unroll
       for (int j = 0; j < FJ; ++j)
         bfrag[j] = *reinterpret_cast<const bf16x8*>(
-            &b_lds[cur][(wn + j * 16 + lrow) * LDK + s * 32 + lk]);
+            &b_buf(cur)[(wn + j * 16 + lrow) * LDK + s * 32 + lk]);
 #pragma unroll
       for (int i = 0; i < FI; ++i)
 #pragma unroll
@@ -155,8 +160,8 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
               afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
     }
     if (has_next) {
-      pw_store_tile<BM>(a_lds[cur ^ 1], tid, av);
-      pw_store_tile<BN>(b_lds[cur ^ 1], tid, bv);
+      pw_store_tile<BM>(a_buf(cur ^ 1), tid, av);
+      pw_store_tile<BN>(b_buf(cur ^ 1), tid, bv);
     }
     __syncthreads();
   }
@@ -195,7 +200,7 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
       sq[j] += __shfl_down(sq[j], 16);
     }
     __syncthreads();  // LDS rewritten as fp32 scratch below
-    float* sscr = reinterpret_cast<float*>(&a_lds[0][0]);  // [4 waves][WTN]
+    float* sscr = reinterpret_cast<float*>(&smem[0]);  // [4 waves][WTN]
     float* qscr = sscr + 4 * WTN;
     if (lane < 16) {
 #pragma unroll
@@ -229,7 +234,7 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
   // (cdna_hip_programming.md T21), so each wave round-trips its tile through
   // LDS and issues 16-B row-major stores instead.
   {
-    __bf16* ctile = &a_lds[0][0] + wid * (WTM * WTN);  // 2*BM*LDK >= 4*WTM*WTN
+    __bf16* ctile = &smem[0] + wid * (WTM * WTN);
 #pragma unroll
     for (int i = 0; i < FI; ++i)
 #pragma unroll
@@ -240,7 +245,7 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
               (__bf16)__float2bfloat16(acc[i][j][r]);
     __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt: wave's own LDS stores landed
     constexpr int CU = WTM * WTN / 64 / 8;  // vec8 units per lane
-    const bool full_cols = (n0 + wn + WTN) <= N && (N & 7) == 0;
+    const bool n_aligned = (N & 7) == 0;
 #pragma unroll
     for (int u = 0; u < CU; ++u) {
       const int unit = lane + u * 64;
@@ -249,7 +254,7 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
       const long long gm = m0 + wm + row;
       if (gm >= M) continue;
       const int gn = n0 + wn + c8;
-      if (full_cols) {
+      if (n_aligned && gn + 8 <= N) {
         *reinterpret_cast<bf16x8*>(y + gm * N + gn) =
             *reinterpret_cast<const bf16x8*>(&ctile[row * WTN + c8]);
       } else {
@@ -438,31 +443,34 @@ at::Tensor pw_conv2d_fwd_mfma(at::Tensor x, at::Tensor w,
   // N-tile config: smallest width that doesn't add extra column tiles
   // (every extra column tile re-reads the whole x) — 32 for the pw-linear
   // projections (N<=32), 64 for N<=64, 128 otherwise.
+#define PW_LAUNCH(BN_, WR_, WC_)                                              \
+  do {                                                                        \
+    dim3 grid((unsigned)((M + kPwBM - 1) / kPwBM), (N + (BN_)-1) / (BN_));    \
+    if (K <= 64) {                                                            \
+      if (stats_p)                                                            \
+        pw_gemm_bf16_kernel<128, BN_, WR_, WC_, 1, true>                      \
+            <<<grid, 256, 0, stream>>>(xp, wp, yp, stats_p, M, N, K);         \
+      else                                                                    \
+        pw_gemm_bf16_kernel<128, BN_, WR_, WC_, 1, false>                     \
+            <<<grid, 256, 0, stream>>>(xp, wp, yp, nullptr, M, N, K);         \
+    } else {                                                                  \
+      if (stats_p)                                                            \
+        pw_gemm_bf16_kernel<128, BN_, WR_, WC_, 2, true>                      \
+            <<<grid, 256, 0, stream>>>(xp, wp, yp, stats_p, M, N, K);         \
+      else                                                                    \
+        pw_gemm_bf16_kernel<128, BN_, WR_, WC_, 2, false>                     \
+            <<<grid, 256, 0, stream>>>(xp, wp, yp, nullptr, M, N, K);         \
+    }                                                                         \
+  } while (0)
+
   if (N <= 32) {
-    dim3 grid((unsigned)((M + kPwBM - 1) / kPwBM), (N + 31) / 32);
-    if (stats_p)
-      pw_gemm_bf16_kernel<128, 32, 4, 1, true><<<grid, 256, 0, stream>>>(
-          xp, wp, yp, stats_p, M, N, K);
-    else
-      pw_gemm_bf16_kernel<128, 32, 4, 1, false><<<grid, 256, 0, stream>>>(
-          xp, wp, yp, nullptr, M, N, K);
+    PW_LAUNCH(32, 4, 1);
   } else if (N <= 64) {
-    dim3 grid((unsigned)((M + kPwBM - 1) / kPwBM), (N + 63) / 64);
-    if (stats_p)
-      pw_gemm_bf16_kernel<128, 64, 2, 2, true><<<grid, 256, 0, stream>>>(
-          xp, wp, yp, stats_p, M, N, K);
-    else
-      pw_gemm_bf16_kernel<128, 64, 2, 2, false><<<grid, 256, 0, stream>>>(
-          xp, wp, yp, nullptr, M, N, K);
+    PW_LAUNCH(64, 2, 2);
   } else {
-    dim3 grid((unsigned)((M + kPwBM - 1) / kPwBM), (N + 127) / 128);
-    if (stats_p)
-      pw_gemm_bf16_kernel<128, 128, 2, 2, true><<<grid, 256, 0, stream>>>(
-          xp, wp, yp, stats_p, M, N, K);
-    else
-      pw_gemm_bf16_kernel<128, 128, 2, 2, false><<<grid, 256, 0, stream>>>(
-          xp, wp, yp, nullptr, M, N, K);
+    PW_LAUNCH(128, 2, 2);
   }
+#undef PW_LAUNCH
   return y;
 }
 
