@@ -49,23 +49,20 @@ int bn_pick_vec(long long c, int elem_size) {
   return 1;
 }
 
-int bn_pick_log2_cpb(int cv) {
-  int l = 0;
-  while ((1 << l) < cv && l < 6) ++l;
-  return l;
-}
-
 struct BnPlan {
-  int log2_cpb, ctiles, chunks, rows_per_chunk;
+  int cpb, ctiles, chunks, rows_per_chunk;
 };
 
+// Channel-slot count per block is NOT rounded up to a power of two: C=144
+// with VEC=8 (cv=18) previously ran with 18 of 32 slots active — 44% of
+// every block's load bandwidth idle (r01->r02 profile, bn kernels ~3x off
+// the streaming ceiling). Tiles are balanced across ctiles instead.
 BnPlan bn_plan(int cv, long long rows) {
   BnPlan p;
-  p.log2_cpb = bn_pick_log2_cpb(cv);
-  const int cpb = 1 << p.log2_cpb;
-  p.ctiles = (cv + cpb - 1) / cpb;
+  p.ctiles = (cv + 63) / 64;
+  p.cpb = (cv + p.ctiles - 1) / p.ctiles;
   long long want = (2048 + p.ctiles - 1) / p.ctiles;
-  const int nrg = 256 >> p.log2_cpb;
+  const int nrg = 256 / p.cpb;
   // floor of ~16 row-iterations per thread: small tensors otherwise explode
   // into 2048 blocks whose per-block atomics serialize on C addresses
   long long by_iters = rows / ((long long)nrg * 16);
@@ -84,15 +81,16 @@ BnPlan bn_plan(int cv, long long rows) {
 template <typename T, int VEC>
 __global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum,
                                 float* __restrict__ sumsq, long long M, int C,
-                                int log2_cpb, int rows_per_chunk) {
+                                int cpb, int rows_per_chunk) {
   extern __shared__ float lds[];  // [256 * VEC]
-  const int cpb = 1 << log2_cpb;
-  const int slot = threadIdx.x & (cpb - 1);
-  const int rg = threadIdx.x >> log2_cpb;
-  const int nrg = blockDim.x >> log2_cpb;
+  const int slot = threadIdx.x % cpb;
+  const int rg = threadIdx.x / cpb;
+  const int nrg = blockDim.x / cpb;
   const int cv = C / VEC;
   const int cvec = blockIdx.x * cpb + slot;
-  const bool active = cvec < cv;
+  // tail threads (cpb not dividing 256) must not stream rows: their rg
+  // aliases row-group 0's stride pattern and would double-count
+  const bool active = cvec < cv && rg < nrg;
   const int c = cvec * VEC;
 
   float s[VEC], q[VEC];
@@ -130,17 +128,28 @@ __global__ void bn_stats_kernel(const T* __restrict__ x, float* __restrict__ sum
     }
   }
 
-  float* my = lds + (size_t)threadIdx.x * VEC;
+  float* my = lds + (size_t)(rg * cpb + slot) * VEC;
+  int p2 = 1;
+  while (p2 * 2 <= nrg) p2 *= 2;
+  const bool in_block = rg < nrg;  // cpb may not divide 256: tail threads idle
   // reduce s then q through the same LDS buffer
 #pragma unroll
   for (int pass = 0; pass < 2; ++pass) {
     __syncthreads();
+    if (in_block) {
 #pragma unroll
-    for (int j = 0; j < VEC; ++j) my[j] = pass == 0 ? s[j] : q[j];
+      for (int j = 0; j < VEC; ++j) my[j] = pass == 0 ? s[j] : q[j];
+    }
     __syncthreads();
-    for (int step = nrg >> 1; step > 0; step >>= 1) {
+    if (in_block && rg >= p2) {  // fold the non-pow2 excess (unique dst)
+      float* dst = lds + (size_t)((rg - p2) * cpb + slot) * VEC;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) dst[j] += my[j];
+    }
+    __syncthreads();
+    for (int step = p2 >> 1; step > 0; step >>= 1) {
       if (rg < step) {
-        const float* other = lds + ((size_t)((rg + step) << log2_cpb) + slot) * VEC;
+        const float* other = lds + ((size_t)((rg + step) * cpb) + slot) * VEC;
 #pragma unroll
         for (int j = 0; j < VEC; ++j) my[j] += other[j];
       }
@@ -230,14 +239,13 @@ __global__ void bn_act_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
                                   const float* __restrict__ dp,  // [B] per-sample
                                   long long hw,                  // rows per sample
                                   long long M, int C,
-                                  int log2_cpb, int rows_per_chunk) {
-  const int cpb = 1 << log2_cpb;
-  const int slot = threadIdx.x & (cpb - 1);
-  const int rg = threadIdx.x >> log2_cpb;
-  const int nrg = blockDim.x >> log2_cpb;
+                                  int cpb, int rows_per_chunk) {
+  const int slot = threadIdx.x % cpb;
+  const int rg = threadIdx.x / cpb;
+  const int nrg = blockDim.x / cpb;
   const int cv = C / VEC;
   const int cvec = blockIdx.x * cpb + slot;
-  if (cvec >= cv) return;
+  if (cvec >= cv || rg >= nrg) return;
   const int c = cvec * VEC;
 
   float sc[VEC], sh[VEC];
@@ -297,15 +305,14 @@ __global__ void bn_act_bwd_reduce_kernel(
     const float* __restrict__ weight, const float* __restrict__ bias,
     float* __restrict__ dgamma, float* __restrict__ dbeta,
     const float* __restrict__ dp, long long hw, long long M, int C,
-    int log2_cpb, int rows_per_chunk) {
+    int cpb, int rows_per_chunk) {
   extern __shared__ float lds[];
-  const int cpb = 1 << log2_cpb;
-  const int slot = threadIdx.x & (cpb - 1);
-  const int rg = threadIdx.x >> log2_cpb;
-  const int nrg = blockDim.x >> log2_cpb;
+  const int slot = threadIdx.x % cpb;
+  const int rg = threadIdx.x / cpb;
+  const int nrg = blockDim.x / cpb;
   const int cv = C / VEC;
   const int cvec = blockIdx.x * cpb + slot;
-  const bool active = cvec < cv;
+  const bool active = cvec < cv && rg < nrg;
   const int c = cvec * VEC;
 
   float sg[VEC], sgx[VEC], mn[VEC], is[VEC], ga[VEC], be[VEC];
@@ -363,16 +370,27 @@ __global__ void bn_act_bwd_reduce_kernel(
     }
   }
 
-  float* my = lds + (size_t)threadIdx.x * VEC;
+  float* my = lds + (size_t)(rg * cpb + slot) * VEC;
+  int p2 = 1;
+  while (p2 * 2 <= nrg) p2 *= 2;
+  const bool in_block = rg < nrg;
 #pragma unroll
   for (int pass = 0; pass < 2; ++pass) {
     __syncthreads();
+    if (in_block) {
 #pragma unroll
-    for (int j = 0; j < VEC; ++j) my[j] = pass == 0 ? sg[j] : sgx[j];
+      for (int j = 0; j < VEC; ++j) my[j] = pass == 0 ? sg[j] : sgx[j];
+    }
     __syncthreads();
-    for (int step = nrg >> 1; step > 0; step >>= 1) {
+    if (in_block && rg >= p2) {  // fold the non-pow2 excess (unique dst)
+      float* dst = lds + (size_t)((rg - p2) * cpb + slot) * VEC;
+#pragma unroll
+      for (int j = 0; j < VEC; ++j) dst[j] += my[j];
+    }
+    __syncthreads();
+    for (int step = p2 >> 1; step > 0; step >>= 1) {
       if (rg < step) {
-        const float* other = lds + ((size_t)((rg + step) << log2_cpb) + slot) * VEC;
+        const float* other = lds + ((size_t)((rg + step) * cpb) + slot) * VEC;
 #pragma unroll
         for (int j = 0; j < VEC; ++j) my[j] += other[j];
       }
@@ -399,14 +417,13 @@ __global__ void bn_act_bwd_dx_kernel(
     const float* __restrict__ weight, const float* __restrict__ bias,
     const float* __restrict__ dgamma, const float* __restrict__ dbeta,
     const float* __restrict__ dp, long long hw,
-    long long M, int C, float invM, int log2_cpb, int rows_per_chunk) {
-  const int cpb = 1 << log2_cpb;
-  const int slot = threadIdx.x & (cpb - 1);
-  const int rg = threadIdx.x >> log2_cpb;
-  const int nrg = blockDim.x >> log2_cpb;
+    long long M, int C, float invM, int cpb, int rows_per_chunk) {
+  const int slot = threadIdx.x % cpb;
+  const int rg = threadIdx.x / cpb;
+  const int nrg = blockDim.x / cpb;
   const int cv = C / VEC;
   const int cvec = blockIdx.x * cpb + slot;
-  if (cvec >= cv) return;
+  if (cvec >= cv || rg >= nrg) return;
   const int c = cvec * VEC;
 
   float sc[VEC], sh[VEC], mn[VEC], k1[VEC], k2[VEC], k3[VEC];
@@ -593,7 +610,7 @@ std::vector<at::Tensor> bn_act_fwd(
         DISPATCH_VEC(vec, [&] {
           hipLaunchKernelGGL((bn_stats_kernel<T, KVEC>), grid, dim3(256), lds, stream,
                              (const T*)x.data_ptr(), sum.data_ptr<float>(),
-                             sumsq.data_ptr<float>(), M, C, plan.log2_cpb,
+                             sumsq.data_ptr<float>(), M, C, plan.cpb,
                              plan.rows_per_chunk);
         });
       });
@@ -621,13 +638,13 @@ std::vector<at::Tensor> bn_act_fwd(
                              stream, (const T*)x.data_ptr(), (T*)y.data_ptr(),
                              scale.data_ptr<float>(), shift.data_ptr<float>(),
                              (const T*)residual.data_ptr(), dp_p, hw, M, C,
-                             plan.log2_cpb, plan.rows_per_chunk);
+                             plan.cpb, plan.rows_per_chunk);
         } else {
           hipLaunchKernelGGL((bn_act_fwd_kernel<T, ACT, KVEC, false>), grid, dim3(256), 0,
                              stream, (const T*)x.data_ptr(), (T*)y.data_ptr(),
                              scale.data_ptr<float>(), shift.data_ptr<float>(),
                              (const T*)nullptr, dp_p, hw, M, C,
-                             plan.log2_cpb, plan.rows_per_chunk);
+                             plan.cpb, plan.rows_per_chunk);
         }
       });
     });
@@ -678,7 +695,7 @@ std::vector<at::Tensor> bn_act_bwd(
                            stream, (const T*)dy.data_ptr(), (const T*)x.data_ptr(),
                            save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),
                            w_p, b_p, dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
-                           dp_p, hw_bn, M, C, plan.log2_cpb, plan.rows_per_chunk);
+                           dp_p, hw_bn, M, C, plan.cpb, plan.rows_per_chunk);
       });
     });
   });
@@ -692,14 +709,14 @@ std::vector<at::Tensor> bn_act_bwd(
                              (T*)dx.data_ptr(), save_mean.data_ptr<float>(),
                              save_invstd.data_ptr<float>(), w_p, b_p,
                              dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
-                             dp_p, hw_bn, M, C, invM, plan.log2_cpb, plan.rows_per_chunk);
+                             dp_p, hw_bn, M, C, invM, plan.cpb, plan.rows_per_chunk);
         } else {
           hipLaunchKernelGGL((bn_act_bwd_dx_kernel<T, ACT, false, KVEC>), grid, dim3(256), 0,
                              stream, (const T*)dy.data_ptr(), (const T*)x.data_ptr(),
                              (T*)dx.data_ptr(), save_mean.data_ptr<float>(),
                              save_invstd.data_ptr<float>(), w_p, b_p,
                              dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
-                             dp_p, hw_bn, M, C, invM, plan.log2_cpb, plan.rows_per_chunk);
+                             dp_p, hw_bn, M, C, invM, plan.cpb, plan.rows_per_chunk);
         }
       });
     });

@@ -50,18 +50,21 @@ constexpr int kStatsBuckets = 64;
 // runs the MFMAs on the current LDS buffer, then writes the registers into
 // the other buffer — one __syncthreads per K step.
 // ---------------------------------------------------------------------------
-constexpr int BK = 64;       // staged K per iteration (2 MFMA k-steps)
-constexpr int LDK = BK + 8;  // padded LDS row stride (16-B aligned: 144 B)
+// KSTEP: staged K per iteration (one or two MFMA k-steps); 32 for the
+// low-K expansion convs (e.g. 24->144) where a 64-wide stage is half zeros.
+template <int R, int KS>
+constexpr int pw_units() { return (R * KS / 8 + 255) / 256; }
 
-template <int ROWS>
+template <int ROWS, int KSTEP>
 DFD_DEV void pw_load_tile(const __hip_bfloat16* __restrict__ src, long long row0,
                           long long row_end, int k0, int K, int ld, int tid,
-                          bf16x8 (&v)[ROWS * 8 / 256]) {
+                          bf16x8 (&v)[pw_units<ROWS, KSTEP>()]) {
 #pragma unroll
-  for (int u = 0; u < ROWS * 8 / 256; ++u) {
+  for (int u = 0; u < pw_units<ROWS, KSTEP>(); ++u) {
     const int idx = tid + u * 256;
-    const int row = idx >> 3;
-    const int c = (idx & 7) * 8;
+    if (idx >= ROWS * KSTEP / 8) break;
+    const int row = idx / (KSTEP / 8);
+    const int c = (idx % (KSTEP / 8)) * 8;
     const long long gr = row0 + row;
     v[u] = bf16x8{};
     if (gr < row_end) {
@@ -82,18 +85,21 @@ DFD_DEV void pw_load_tile(const __hip_bfloat16* __restrict__ src, long long row0
   }
 }
 
-template <int ROWS>
-DFD_DEV void pw_store_tile(__bf16* lds, int tid, const bf16x8 (&v)[ROWS * 8 / 256]) {
+template <int ROWS, int KSTEP>
+DFD_DEV void pw_store_tile(__bf16* lds, int tid,
+                           const bf16x8 (&v)[pw_units<ROWS, KSTEP>()]) {
 #pragma unroll
-  for (int u = 0; u < ROWS * 8 / 256; ++u) {
+  for (int u = 0; u < pw_units<ROWS, KSTEP>(); ++u) {
     const int idx = tid + u * 256;
-    *reinterpret_cast<bf16x8*>(&lds[(idx >> 3) * LDK + (idx & 7) * 8]) = v[u];
+    if (idx >= ROWS * KSTEP / 8) break;
+    *reinterpret_cast<bf16x8*>(
+        &lds[(idx / (KSTEP / 8)) * (KSTEP + 8) + (idx % (KSTEP / 8)) * 8]) = v[u];
   }
 }
 
 // NBUF: LDS pipeline depth — 1 for single-iteration shapes (K <= BK), where
 // the 2nd buffer would only halve occupancy, 2 (double-buffered) otherwise.
-template <int BM, int BN, int WROWS, int WCOLS, int NBUF, bool STATS>
+template <int BM, int BN, int WROWS, int WCOLS, int NBUF, int KSTEP, bool STATS>
 __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
     const __hip_bfloat16* __restrict__ x,  // [M, K] row-major
     const __hip_bfloat16* __restrict__ w,  // [N, K] row-major
@@ -104,9 +110,11 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
   constexpr int WTN = BN / WCOLS;            // wave tile N
   constexpr int FI = WTM / 16;               // row fragments per wave
   constexpr int FJ = WTN / 16;               // col fragments per wave
-  static_assert(NBUF * (BM + BN) * LDK >= 4 * WTM * WTN * WROWS * WCOLS / 4,
-                "smem too small for the writeback staging tile");
-  __shared__ __bf16 smem[NBUF * (BM + BN) * LDK];
+  constexpr int LDK = KSTEP + 8;             // padded LDS row stride
+  constexpr int STAGE = NBUF * (BM + BN) * LDK;
+  constexpr int CTILE = 4 * WTM * WTN;       // writeback staging
+  constexpr int SMEM = STAGE > CTILE ? STAGE : CTILE;
+  __shared__ __bf16 smem[SMEM];
   auto a_buf = [&](int b) { return &smem[b * BM * LDK]; };
   auto b_buf = [&](int b) { return &smem[NBUF * BM * LDK + b * BN * LDK]; };
 
@@ -127,22 +135,22 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
   const int lrow = lane & 15;      // fragment row/col
   const int lk = (lane >> 4) * 8;  // fragment k offset within a 32-k step
 
-  bf16x8 av[BM * 8 / 256], bv[BN * 8 / 256];
-  pw_load_tile<BM>(x, m0, M, 0, K, K, tid, av);
-  pw_load_tile<BN>(w, n0, N, 0, K, K, tid, bv);
-  pw_store_tile<BM>(a_buf(0), tid, av);
-  pw_store_tile<BN>(b_buf(0), tid, bv);
+  bf16x8 av[pw_units<BM, KSTEP>()], bv[pw_units<BN, KSTEP>()];
+  pw_load_tile<BM, KSTEP>(x, m0, M, 0, K, K, tid, av);
+  pw_load_tile<BN, KSTEP>(w, n0, N, 0, K, K, tid, bv);
+  pw_store_tile<BM, KSTEP>(a_buf(0), tid, av);
+  pw_store_tile<BN, KSTEP>(b_buf(0), tid, bv);
   __syncthreads();
 
   int cur = 0;
-  for (int k0 = 0; k0 < K; k0 += BK, cur ^= (NBUF - 1)) {
-    const bool has_next = k0 + BK < K;
+  for (int k0 = 0; k0 < K; k0 += KSTEP, cur ^= (NBUF - 1)) {
+    const bool has_next = k0 + KSTEP < K;
     if (has_next) {
-      pw_load_tile<BM>(x, m0, M, k0 + BK, K, K, tid, av);
-      pw_load_tile<BN>(w, n0, N, k0 + BK, K, K, tid, bv);
+      pw_load_tile<BM, KSTEP>(x, m0, M, k0 + KSTEP, K, K, tid, av);
+      pw_load_tile<BN, KSTEP>(w, n0, N, k0 + KSTEP, K, K, tid, bv);
     }
 #pragma unroll
-    for (int s = 0; s < 2; ++s) {
+    for (int s = 0; s < KSTEP / 32; ++s) {
       bf16x8 afrag[FI], bfrag[FJ];
 #pragma unroll
       for (int i = 0; i < FI; ++i)
@@ -160,8 +168,8 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
               afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
     }
     if (has_next) {
-      pw_store_tile<BM>(a_buf(cur ^ 1), tid, av);
-      pw_store_tile<BN>(b_buf(cur ^ 1), tid, bv);
+      pw_store_tile<BM, KSTEP>(a_buf(cur ^ 1), tid, av);
+      pw_store_tile<BN, KSTEP>(b_buf(cur ^ 1), tid, bv);
     }
     __syncthreads();
   }
@@ -443,24 +451,22 @@ at::Tensor pw_conv2d_fwd_mfma(at::Tensor x, at::Tensor w,
   // N-tile config: smallest width that doesn't add extra column tiles
   // (every extra column tile re-reads the whole x) — 32 for the pw-linear
   // projections (N<=32), 64 for N<=64, 128 otherwise.
-#define PW_LAUNCH(BN_, WR_, WC_)                                              \
+#define PW_KERNEL(BN_, WR_, WC_, NBUF_, KS_)                                  \
   do {                                                                        \
     dim3 grid((unsigned)((M + kPwBM - 1) / kPwBM), (N + (BN_)-1) / (BN_));    \
-    if (K <= 64) {                                                            \
-      if (stats_p)                                                            \
-        pw_gemm_bf16_kernel<128, BN_, WR_, WC_, 1, true>                      \
-            <<<grid, 256, 0, stream>>>(xp, wp, yp, stats_p, M, N, K);         \
-      else                                                                    \
-        pw_gemm_bf16_kernel<128, BN_, WR_, WC_, 1, false>                     \
-            <<<grid, 256, 0, stream>>>(xp, wp, yp, nullptr, M, N, K);         \
-    } else {                                                                  \
-      if (stats_p)                                                            \
-        pw_gemm_bf16_kernel<128, BN_, WR_, WC_, 2, true>                      \
-            <<<grid, 256, 0, stream>>>(xp, wp, yp, stats_p, M, N, K);         \
-      else                                                                    \
-        pw_gemm_bf16_kernel<128, BN_, WR_, WC_, 2, false>                     \
-            <<<grid, 256, 0, stream>>>(xp, wp, yp, nullptr, M, N, K);         \
-    }                                                                         \
+    if (stats_p)                                                              \
+      pw_gemm_bf16_kernel<128, BN_, WR_, WC_, NBUF_, KS_, true>               \
+          <<<grid, 256, 0, stream>>>(xp, wp, yp, stats_p, M, N, K);           \
+    else                                                                      \
+      pw_gemm_bf16_kernel<128, BN_, WR_, WC_, NBUF_, KS_, false>              \
+          <<<grid, 256, 0, stream>>>(xp, wp, yp, nullptr, M, N, K);           \
+  } while (0)
+
+#define PW_LAUNCH(BN_, WR_, WC_)                                              \
+  do {                                                                        \
+    if (K <= 32) PW_KERNEL(BN_, WR_, WC_, 1, 32);                             \
+    else if (K <= 64) PW_KERNEL(BN_, WR_, WC_, 1, 64);                        \
+    else PW_KERNEL(BN_, WR_, WC_, 2, 64);                                     \
   } while (0)
 
   if (N <= 32) {
@@ -471,6 +477,7 @@ at::Tensor pw_conv2d_fwd_mfma(at::Tensor x, at::Tensor w,
     PW_LAUNCH(128, 2, 2);
   }
 #undef PW_LAUNCH
+#undef PW_KERNEL
   return y;
 }
 
