@@ -12,6 +12,8 @@
 
 #include <vector>
 
+#include <cstring>
+
 #include "common.h"
 
 namespace {
@@ -53,6 +55,7 @@ ChunkTable build_chunks(const std::vector<at::Tensor>& a,
   // per step and keeps optimizer steps hipGraph-capturable.
   struct Cache {
     std::vector<long long> key;
+    at::Tensor host;  // pinned staging — must outlive any captured memcpy node
     at::Tensor dev;
   };
   static thread_local std::vector<Cache> cache;
@@ -64,14 +67,18 @@ ChunkTable build_chunks(const std::vector<at::Tensor>& a,
       return t;
     }
   }
-  auto host = at::from_blob(rows.data(), {(long long)nchunks, 5},
-                            at::TensorOptions().dtype(at::kLong))
-                  .clone();
+  // PINNED host staging: under hipGraph capture the backward's grads live at
+  // graph-pool addresses, so the table is (re)built inside capture — a
+  // pageable H2D there is illegal. A pinned copy captures as a legal memcpy
+  // node; the cache keeps the pinned tensor alive for replays.
+  auto host = at::empty({(long long)nchunks, 5},
+                        at::TensorOptions().dtype(at::kLong).pinned_memory(true));
+  std::memcpy(host.data_ptr(), rows.data(), rows.size() * sizeof(long long));
   ChunkTable t;
   t.dev = host.to(opts.device(), /*non_blocking=*/true);
   t.nchunks = nchunks;
   if (cache.size() > 8) cache.clear();  // bound: a few optimizers/EMA per process
-  cache.push_back({std::move(rows), t.dev});
+  cache.push_back({std::move(rows), host, t.dev});
   return t;
 }
 
