@@ -8,7 +8,10 @@ small. Usage (on a GPU box):
   rocprofv3 --pmc TCC_EA0_RDREQ_sum TCC_EA0_WRREQ_sum -d out -o hbm -- \
       python tools/pmc_probe.py dw_fwd_k5
 """
+import os
 import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
