@@ -103,43 +103,53 @@ __global__ void dw_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
 }
 
 // ---------------------------------------------------------------------------
-// forward, stride 1, TW consecutive outputs per thread.
-// Loads per output drop from K*K to ~K*(K+TW-1)/TW (x) and weight loads
-// amortize by TW; bwd-data for stride 1 reuses this kernel with the packed
-// weight flipped in (kh,kw) and padding (K-1-p) — it is the same stride-1
-// correlation.
+// forward, stride 1: TH x TW output tile per thread (2-D).
+// PMC evidence (r02): the 1-D TW=4 version re-fetched x ~4x from HBM at
+// ~5.2 TB/s effective — bandwidth-saturated on redundant row re-reads. A
+// TH-row tile cuts the per-output x-row fetches from K to (K+TH-1)/TH; the
+// whole K*K weight tile is loaded once per grid-stride iteration (L1-hot)
+// instead of per (col, t) pair. bwd-data for stride 1 reuses this kernel
+// with the packed weight flipped in (kh,kw) and padding (K-1-p) — the same
+// stride-1 correlation.
 // ---------------------------------------------------------------------------
-template <typename T, int K, int VEC, int TW>
+template <typename T, int K, int VEC, int TW, int TH>
 __global__ void dw_fwd_s1_kernel(const T* __restrict__ x, const T* __restrict__ w,
                                  T* __restrict__ y, int N, int C, int H, int W,
                                  int Ho, int Wo, int ph, int pw) {
   const int cv = C / VEC;
   const int wt = (Wo + TW - 1) / TW;  // wo tiles per row
-  const long long total = (long long)N * Ho * wt * cv;
+  const int ht = (Ho + TH - 1) / TH;  // ho tiles per column
+  const long long total = (long long)N * ht * wt * cv;
   for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
        idx += (long long)gridDim.x * blockDim.x) {
     const int c = (int)(idx % cv) * VEC;
     long long p = idx / cv;
     const int wo0 = (int)(p % wt) * TW;
     p /= wt;
-    const int ho = (int)(p % Ho);
-    const int n = (int)(p / Ho);
+    const int ho0 = (int)(p % ht) * TH;
+    const int n = (int)(p / ht);
 
-    float acc[TW][VEC];
+    // whole K*K weight tile once per iteration (stays in registers / L1)
+    TVec<T, VEC> wv[K * K];
 #pragma unroll
-    for (int t = 0; t < TW; ++t)
-#pragma unroll
-      for (int i = 0; i < VEC; ++i) acc[t][i] = 0.f;
+    for (int i = 0; i < K * K; ++i) wv[i] = vload<T, VEC>(w + (long long)i * C + c);
 
-    const int hi0 = ho - ph;
+    float acc[TH][TW][VEC];
+#pragma unroll
+    for (int th = 0; th < TH; ++th)
+#pragma unroll
+      for (int t = 0; t < TW; ++t)
+#pragma unroll
+        for (int i = 0; i < VEC; ++i) acc[th][t][i] = 0.f;
+
+    const int hi0 = ho0 - ph;
     const int wi0 = wo0 - pw;
     const bool interior = wi0 >= 0 && wi0 + K + TW - 2 < W;
 #pragma unroll
-    for (int kh = 0; kh < K; ++kh) {
-      const int hi = hi0 + kh;
+    for (int row = 0; row < K + TH - 1; ++row) {
+      const int hi = hi0 + row;
       if (hi < 0 || hi >= H) continue;
       const T* xrow = x + (((long long)n * H + hi) * W) * C + c;
-      const T* wrow = w + ((long long)kh * K) * C + c;
       if (interior) {
         // branchless: issue all K+TW-1 column loads, then consume (per-load
         // guards would serialize each load behind s_waitcnt vmcnt(0))
@@ -148,16 +158,20 @@ __global__ void dw_fwd_s1_kernel(const T* __restrict__ x, const T* __restrict__ 
         for (int col = 0; col < K + TW - 1; ++col)
           xv[col] = vload<T, VEC>(xrow + (long long)(wi0 + col) * C);
 #pragma unroll
-        for (int col = 0; col < K + TW - 1; ++col) {
+        for (int th = 0; th < TH; ++th) {
+          const int kh = row - th;
+          if (kh < 0 || kh >= K) continue;
 #pragma unroll
-          for (int t = 0; t < TW; ++t) {
-            const int kw = col - t;
-            if (kw < 0 || kw >= K) continue;
-            const TVec<T, VEC> wv = vload<T, VEC>(wrow + (long long)kw * C);
+          for (int col = 0; col < K + TW - 1; ++col)
 #pragma unroll
-            for (int i = 0; i < VEC; ++i)
-              acc[t][i] += DfdCvt<T>::to_f32(xv[col].v[i]) * DfdCvt<T>::to_f32(wv.v[i]);
-          }
+            for (int t = 0; t < TW; ++t) {
+              const int kw = col - t;
+              if (kw < 0 || kw >= K) continue;
+#pragma unroll
+              for (int i = 0; i < VEC; ++i)
+                acc[th][t][i] += DfdCvt<T>::to_f32(xv[col].v[i]) *
+                                 DfdCvt<T>::to_f32(wv[kh * K + kw].v[i]);
+            }
         }
       } else {
 #pragma unroll
@@ -165,27 +179,35 @@ __global__ void dw_fwd_s1_kernel(const T* __restrict__ x, const T* __restrict__ 
           const int wi = wi0 + col;
           if (wi < 0 || wi >= W) continue;
           const TVec<T, VEC> xv = vload<T, VEC>(xrow + (long long)wi * C);
-          // this column contributes to outputs t with 0 <= col - t < K
 #pragma unroll
-          for (int t = 0; t < TW; ++t) {
-            const int kw = col - t;
-            if (kw < 0 || kw >= K) continue;
-            const TVec<T, VEC> wv = vload<T, VEC>(wrow + (long long)kw * C);
+          for (int th = 0; th < TH; ++th) {
+            const int kh = row - th;
+            if (kh < 0 || kh >= K) continue;
 #pragma unroll
-            for (int i = 0; i < VEC; ++i)
-              acc[t][i] += DfdCvt<T>::to_f32(xv.v[i]) * DfdCvt<T>::to_f32(wv.v[i]);
+            for (int t = 0; t < TW; ++t) {
+              const int kw = col - t;
+              if (kw < 0 || kw >= K) continue;
+#pragma unroll
+              for (int i = 0; i < VEC; ++i)
+                acc[th][t][i] += DfdCvt<T>::to_f32(xv.v[i]) *
+                                 DfdCvt<T>::to_f32(wv[kh * K + kw].v[i]);
+            }
           }
         }
       }
     }
-    T* yrow = y + (((long long)n * Ho + ho) * Wo) * C + c;
 #pragma unroll
-    for (int t = 0; t < TW; ++t) {
-      if (wo0 + t >= Wo) break;
-      TVec<T, VEC> yv;
+    for (int th = 0; th < TH; ++th) {
+      if (ho0 + th >= Ho) break;
+      T* yrow = y + (((long long)n * Ho + ho0 + th) * Wo) * C + c;
 #pragma unroll
-      for (int i = 0; i < VEC; ++i) yv.v[i] = DfdCvt<T>::from_f32(acc[t][i]);
-      vstore<T, VEC>(yrow + (long long)(wo0 + t) * C, yv);
+      for (int t = 0; t < TW; ++t) {
+        if (wo0 + t >= Wo) break;
+        TVec<T, VEC> yv;
+#pragma unroll
+        for (int i = 0; i < VEC; ++i) yv.v[i] = DfdCvt<T>::from_f32(acc[th][t][i]);
+        vstore<T, VEC>(yrow + (long long)(wo0 + t) * C, yv);
+      }
     }
   }
 }
@@ -400,9 +422,11 @@ void launch_fwd(const at::Tensor& x, const at::Tensor& w, at::Tensor& y, const G
   const int block = 256;
   if (g.sh == 1 && g.sw == 1) {
     constexpr int TW = 4;
+    constexpr int TH = 2;
     const int wt = (g.Wo + TW - 1) / TW;
-    const long long total = (long long)g.N * g.Ho * wt * (g.C / VEC);
-    dw_fwd_s1_kernel<T, K, VEC, TW><<<dfd_grid(total, block), block, 0, stream>>>(
+    const int ht = (g.Ho + TH - 1) / TH;
+    const long long total = (long long)g.N * ht * wt * (g.C / VEC);
+    dw_fwd_s1_kernel<T, K, VEC, TW, TH><<<dfd_grid(total, block), block, 0, stream>>>(
         (const T*)x.data_ptr(), (const T*)w.data_ptr(), (T*)y.data_ptr(), g.N, g.C, g.H,
         g.W, g.Ho, g.Wo, g.ph, g.pw);
     return;

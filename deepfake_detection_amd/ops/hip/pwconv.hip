@@ -275,6 +275,114 @@ __global__ __launch_bounds__(256) void pw_gemm_bf16_kernel(
   }
 }
 
+// ---------------------------------------------------------------------------
+// small-K expansion GEMM (K <= 32, 64 < N <= 192): the whole contraction fits
+// one MFMA k-step, and the A/B fragment layouts are row-major vec8 per lane —
+// so fragments load DIRECTLY from global memory. No LDS, no barriers; weight
+// fragments load once per thread and m-tiles grid-stride. Stats accumulate in
+// registers across all m-tiles (one bucket atomic set per block at the end).
+// PMC (r02): the staged kernel ran this class at 34% of roofline, 64%
+// wave-parked on its barriers.
+// ---------------------------------------------------------------------------
+template <int FJ, bool STATS>
+__global__ __launch_bounds__(256) void pw_gemm_smallk_kernel(
+    const __hip_bfloat16* __restrict__ x,  // [M, K] row-major
+    const __hip_bfloat16* __restrict__ w,  // [N, K] row-major
+    __hip_bfloat16* __restrict__ y,        // [M, N] row-major
+    float* __restrict__ stats,             // [kStatsBuckets, 2, N] or null
+    long long M, int N, int K) {
+  const int tid = threadIdx.x;
+  const int lane = tid & (kWave - 1);
+  const int wid = tid / kWave;
+  const int lrow = lane & 15;
+  const int lk = (lane >> 4) * 8;
+  const bool kin = lk < K;
+
+  // B fragments once per thread (w is tiny and L1/L2-hot)
+  bf16x8 bfrag[FJ];
+#pragma unroll
+  for (int j = 0; j < FJ; ++j) {
+    const int gn = j * 16 + lrow;
+    bfrag[j] = bf16x8{};
+    if (gn < N && kin)
+      bfrag[j] = *reinterpret_cast<const bf16x8*>(w + (long long)gn * K + lk);
+  }
+
+  float ssum[FJ], sq[FJ];
+#pragma unroll
+  for (int j = 0; j < FJ; ++j) { ssum[j] = 0.f; sq[j] = 0.f; }
+
+  const int ccol = lane & 15;
+  const int crow0 = (lane >> 4) * 4;
+  const long long mtiles = (M + 255) / 256;  // 4 waves x 64 rows
+  for (long long mt = blockIdx.x; mt < mtiles; mt += gridDim.x) {
+    const long long m0 = mt * 256 + wid * 64;
+    f32x4 acc[4][FJ];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < FJ; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+    bf16x8 afrag[4];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const long long gm = m0 + i * 16 + lrow;
+      afrag[i] = bf16x8{};
+      if (gm < M && kin)
+        afrag[i] = *reinterpret_cast<const bf16x8*>(x + gm * K + lk);
+    }
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < FJ; ++j)
+        acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
+
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+#pragma unroll
+      for (int j = 0; j < FJ; ++j) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const long long gm = m0 + i * 16 + crow0 + r;
+          const int gn = j * 16 + ccol;
+          if (gm < M && gn < N) {
+            const __hip_bfloat16 v = __float2bfloat16(acc[i][j][r]);
+            y[gm * N + gn] = v;
+            if (STATS) {
+              const float f = __bfloat162float(v);
+              ssum[j] += f;
+              sq[j] += f * f;
+            }
+          }
+        }
+      }
+    }
+  }
+
+  if (STATS) {
+#pragma unroll
+    for (int j = 0; j < FJ; ++j) {
+      ssum[j] += __shfl_down(ssum[j], 32);
+      ssum[j] += __shfl_down(ssum[j], 16);
+      sq[j] += __shfl_down(sq[j], 32);
+      sq[j] += __shfl_down(sq[j], 16);
+    }
+    if (lane < 16) {
+      float* bucket =
+          stats + (size_t)((blockIdx.x * 4 + wid) & (kStatsBuckets - 1)) * 2 * N;
+#pragma unroll
+      for (int j = 0; j < FJ; ++j) {
+        const int gn = j * 16 + lane;
+        if (gn < N) {
+          atomicAdd(bucket + gn, ssum[j]);
+          atomicAdd(bucket + N + gn, sq[j]);
+        }
+      }
+    }
+  }
+}
+
 constexpr int kPwBM = 128;
 // ---------------------------------------------------------------------------
 // bwd-weight: dW[N, K] = dy^T @ x, split over M into chunks of fp32 partials.
@@ -469,7 +577,22 @@ at::Tensor pw_conv2d_fwd_mfma(at::Tensor x, at::Tensor w,
     else PW_KERNEL(BN_, WR_, WC_, 2, 64);                                     \
   } while (0)
 
-  if (N <= 32) {
+  if (K <= 32 && (K & 7) == 0 && N > 64 && N <= 192) {
+    // one-k-step expansion: direct fragment loads, no LDS/barriers
+    const long long mtiles = (M + 255) / 256;
+    const unsigned grid = (unsigned)(mtiles < 8192 ? mtiles : 8192);
+    if (N <= 144) {
+      if (stats_p)
+        pw_gemm_smallk_kernel<9, true><<<grid, 256, 0, stream>>>(xp, wp, yp, stats_p, M, N, K);
+      else
+        pw_gemm_smallk_kernel<9, false><<<grid, 256, 0, stream>>>(xp, wp, yp, nullptr, M, N, K);
+    } else {
+      if (stats_p)
+        pw_gemm_smallk_kernel<12, true><<<grid, 256, 0, stream>>>(xp, wp, yp, stats_p, M, N, K);
+      else
+        pw_gemm_smallk_kernel<12, false><<<grid, 256, 0, stream>>>(xp, wp, yp, nullptr, M, N, K);
+    }
+  } else if (N <= 32) {
     PW_LAUNCH(32, 4, 1);
   } else if (N <= 64) {
     PW_LAUNCH(64, 2, 2);
