@@ -113,9 +113,10 @@ __global__ void dw_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
 // stride-1 correlation.
 // ---------------------------------------------------------------------------
 template <typename T, int K, int VEC, int TW, int TH>
-__global__ void dw_fwd_s1_kernel(const T* __restrict__ x, const T* __restrict__ w,
-                                 T* __restrict__ y, int N, int C, int H, int W,
-                                 int Ho, int Wo, int ph, int pw) {
+__global__ __launch_bounds__(256, 2) void dw_fwd_s1_kernel(
+    const T* __restrict__ x, const T* __restrict__ w,
+    T* __restrict__ y, int N, int C, int H, int W,
+    int Ho, int Wo, int ph, int pw) {
   const int cv = C / VEC;
   const int wt = (Wo + TW - 1) / TW;  // wo tiles per row
   const int ht = (Ho + TH - 1) / TH;  // ho tiles per column
@@ -128,11 +129,6 @@ __global__ void dw_fwd_s1_kernel(const T* __restrict__ x, const T* __restrict__ 
     p /= wt;
     const int ho0 = (int)(p % ht) * TH;
     const int n = (int)(p / ht);
-
-    // whole K*K weight tile once per iteration (stays in registers / L1)
-    TVec<T, VEC> wv[K * K];
-#pragma unroll
-    for (int i = 0; i < K * K; ++i) wv[i] = vload<T, VEC>(w + (long long)i * C + c);
 
     float acc[TH][TW][VEC];
 #pragma unroll
@@ -161,16 +157,20 @@ __global__ void dw_fwd_s1_kernel(const T* __restrict__ x, const T* __restrict__ 
         for (int th = 0; th < TH; ++th) {
           const int kh = row - th;
           if (kh < 0 || kh >= K) continue;
+          // weight loads stay per-use: the K*K tile is L1-hot, and holding
+          // it in registers (K*K*4 VGPRs at k5) spilled and ran 3x slower
+          const T* wrow = w + ((long long)kh * K) * C + c;
 #pragma unroll
           for (int col = 0; col < K + TW - 1; ++col)
 #pragma unroll
             for (int t = 0; t < TW; ++t) {
               const int kw = col - t;
               if (kw < 0 || kw >= K) continue;
+              const TVec<T, VEC> wv = vload<T, VEC>(wrow + (long long)kw * C);
 #pragma unroll
               for (int i = 0; i < VEC; ++i)
                 acc[th][t][i] += DfdCvt<T>::to_f32(xv[col].v[i]) *
-                                 DfdCvt<T>::to_f32(wv[kh * K + kw].v[i]);
+                                 DfdCvt<T>::to_f32(wv.v[i]);
             }
         }
       } else {
@@ -187,10 +187,12 @@ __global__ void dw_fwd_s1_kernel(const T* __restrict__ x, const T* __restrict__ 
             for (int t = 0; t < TW; ++t) {
               const int kw = col - t;
               if (kw < 0 || kw >= K) continue;
+              const TVec<T, VEC> wv =
+                  vload<T, VEC>(w + ((long long)kh * K + kw) * C + c);
 #pragma unroll
               for (int i = 0; i < VEC; ++i)
                 acc[th][t][i] += DfdCvt<T>::to_f32(xv.v[i]) *
-                                 DfdCvt<T>::to_f32(wv[kh * K + kw].v[i]);
+                                 DfdCvt<T>::to_f32(wv.v[i]);
             }
           }
         }
