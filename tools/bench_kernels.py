@@ -167,7 +167,7 @@ def bench_pw(B=192):
     on the B4-299 pointwise shapes."""
     from deepfake_detection_amd.ops.pwconv import pw_conv2d_fwd
     shapes = [  # (Cin, Cout, H)
-        (48, 24, 150), (24, 144, 150), (144, 32, 75), (192, 32, 75),
+        (48, 24, 150), (24, 144, 150), (32, 192, 75), (144, 32, 75), (192, 32, 75),
         (192, 288, 38), (288, 48, 38), (672, 112, 19), (960, 160, 19),
         (960, 272, 10), (1632, 272, 10), (1632, 448, 10), (448, 2688, 10),
         (2688, 448, 10),
