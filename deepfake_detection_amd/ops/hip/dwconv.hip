@@ -113,7 +113,7 @@ __global__ void dw_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
 // stride-1 correlation.
 // ---------------------------------------------------------------------------
 template <typename T, int K, int VEC, int TW, int TH>
-__global__ __launch_bounds__(256, 2) void dw_fwd_s1_kernel(
+__global__ void dw_fwd_s1_kernel(
     const T* __restrict__ x, const T* __restrict__ w,
     T* __restrict__ y, int N, int C, int H, int W,
     int Ho, int Wo, int ph, int pw) {
@@ -423,8 +423,11 @@ void launch_fwd(const at::Tensor& x, const at::Tensor& w, at::Tensor& y, const G
                 hipStream_t stream) {
   const int block = 256;
   if (g.sh == 1 && g.sw == 1) {
-    constexpr int TW = 4;
-    constexpr int TH = 2;
+    // TH=1: the 2-row tile halved traffic but cost occupancy (570 vs 433 us
+    // avg k5 in-bench) — these kernels are latency-bound, so widen TW for
+    // more independent column loads in flight instead.
+    constexpr int TW = K == 3 ? 8 : 6;
+    constexpr int TH = 1;
     const int wt = (g.Wo + TW - 1) / TW;
     const int ht = (g.Ho + TH - 1) / TH;
     const long long total = (long long)g.N * ht * wt * (g.C / VEC);
