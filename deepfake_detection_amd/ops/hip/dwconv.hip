@@ -423,10 +423,10 @@ void launch_fwd(const at::Tensor& x, const at::Tensor& w, at::Tensor& y, const G
                 hipStream_t stream) {
   const int block = 256;
   if (g.sh == 1 && g.sw == 1) {
-    // TH=1: the 2-row tile halved traffic but cost occupancy (570 vs 433 us
-    // avg k5 in-bench) — these kernels are latency-bound, so widen TW for
-    // more independent column loads in flight instead.
-    constexpr int TW = K == 3 ? 8 : 6;
+    // TW=4/TH=1 is the measured optimum so far: TH=2 halved x traffic but
+    // cost occupancy (570 vs 433 us avg k5 in-bench); TW=6/8 re-triggered
+    // the compiler's full weight-tile register hoist (spills).
+    constexpr int TW = 4;
     constexpr int TH = 1;
     const int wt = (g.Wo + TW - 1) / TW;
     const int ht = (g.Ho + TH - 1) / TH;
