@@ -18,7 +18,8 @@ import torch
 import torch.nn as nn
 
 from ..ops import functional as O
-from .layers import PointwiseConv2d, create_conv2d, drop_path, sigmoid
+from .layers import (DepthwiseConv2d, PointwiseConv2d, create_conv2d,
+                     drop_path, sigmoid)
 
 __all__ = [
     "BN_MOMENTUM_TF_DEFAULT",
@@ -158,9 +159,9 @@ class ConvBnAct(nn.Module):
 
 
 def _mark_bn_producer(conv):
-    """Ask a PointwiseConv2d to emit BN stats from its epilogue: its output
-    feeds a fused BatchNorm, which then skips its own stats pass."""
-    if isinstance(conv, PointwiseConv2d):
+    """Ask a conv whose output feeds a fused BatchNorm to emit the BN stats
+    from its epilogue — the BN then skips its own stats pass over y."""
+    if isinstance(conv, (PointwiseConv2d, DepthwiseConv2d)):
         conv.emit_bn_stats = True
 
 
@@ -203,6 +204,7 @@ class DepthwiseSeparableConv(nn.Module):
         self.conv_dw = create_conv2d(
             in_chs, in_chs, dw_kernel_size, stride=stride, dilation=dilation,
             padding=pad_type, depthwise=True)
+        _mark_bn_producer(self.conv_dw)
         self.bn1 = norm_layer(in_chs, **norm_kwargs)
         self.act1 = act_layer(inplace=True)
 
@@ -270,6 +272,7 @@ class InvertedResidual(nn.Module):
         self.conv_dw = create_conv2d(
             mid_chs, mid_chs, dw_kernel_size, stride=stride, dilation=dilation,
             padding=pad_type, depthwise=True, **conv_kwargs)
+        _mark_bn_producer(self.conv_dw)
         self.bn2 = norm_layer(mid_chs, **norm_kwargs)
         self.act2 = act_layer(inplace=True)
 

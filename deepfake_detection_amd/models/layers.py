@@ -117,14 +117,19 @@ class Conv2dSame(nn.Conv2d):
 
 class DepthwiseConv2d(nn.Conv2d):
     """nn.Conv2d with groups == in == out, routed to the gfx950 HIP depthwise
-    kernels on ROCm devices (same state_dict as nn.Conv2d)."""
+    kernels on ROCm devices (same state_dict as nn.Conv2d). `emit_bn_stats`
+    (set by the owning block) makes the k3 s1 forward also emit the
+    per-channel stats the following BatchNorm needs."""
+
+    emit_bn_stats = False
 
     def forward(self, x):
         if _dw_hip_path(x, self.weight, self.stride, self.padding, self.dilation, self.groups):
             from ..ops.dwconv import dw_conv2d
 
             return dw_conv2d(x, self.weight, self.bias, self.stride, self.padding,
-                             self.dilation)
+                             self.dilation,
+                             want_stats=self.emit_bn_stats and self.training)
         return super().forward(x)
 
 

@@ -33,14 +33,24 @@ def dw_supported(weight, stride, padding, dilation, groups) -> bool:
     return True
 
 
+STATS_BUCKETS = 64
+
+
 class _DwConv2d(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, weight, sh, sw, ph, pw):
+    def forward(ctx, x, weight, sh, sw, ph, pw, stats_out):
         ext = load_extension()
         x = x.contiguous(memory_format=torch.channels_last)
         C, _, K, _ = weight.shape
         w_packed = weight.reshape(C, K, K).permute(1, 2, 0).contiguous()
-        y = ext.dw_conv2d_fwd(x, w_packed, sh, sw, ph, pw)
+        stats = None
+        if stats_out is not None and K == 3 and sh == 1 and sw == 1:
+            # k3 s1 only: the stats variant fixes each thread's channel; at
+            # k5 the register cost measured slower than the pass it saves
+            stats = torch.zeros(STATS_BUCKETS, 2, C, device=x.device,
+                                dtype=torch.float32)
+            stats_out.append(stats)
+        y = ext.dw_conv2d_fwd(x, w_packed, sh, sw, ph, pw, stats)
         ctx.save_for_backward(x, w_packed)
         ctx.geom = (sh, sw, ph, pw, K)
         ctx.w_dtype = weight.dtype
@@ -58,25 +68,33 @@ class _DwConv2d(torch.autograd.Function):
         if ctx.needs_input_grad[1]:
             dw_kkc = ext.dw_conv2d_bwd_weight(dy, x, K, sh, sw, ph, pw)  # (K,K,C) fp32
             dw = dw_kkc.permute(2, 0, 1).reshape(-1, 1, K, K).to(ctx.w_dtype).contiguous()
-        return dx, dw, None, None, None, None
+        return dx, dw, None, None, None, None, None
 
 
 def _pair(v):
     return (v, v) if isinstance(v, int) else tuple(v)
 
 
-def dw_conv2d(x, weight, bias=None, stride=1, padding=0, dilation=1):
+def dw_conv2d(x, weight, bias=None, stride=1, padding=0, dilation=1,
+              want_stats=False):
     """Depthwise conv2d (groups == C) on the HIP NHWC kernels.
 
     `weight` is the torch-native (C, 1, K, K) tensor; dtype mismatch with x
     (fp32 master weights under bf16 autocast) is resolved by a differentiable
     cast so weight grads land on the master dtype.
+
+    want_stats: for k3 stride-1, also emit per-channel (sum, sumsq) of y for
+    the following fused BatchNorm (attached as ``y._dfd_bn_stats``).
     """
     sh, sw = _pair(stride)
     ph, pw = _pair(padding)
     if weight.dtype != x.dtype:
         weight = weight.to(x.dtype)
-    y = _DwConv2d.apply(x, weight, sh, sw, ph, pw)
+    holder = [] if (want_stats and bias is None) else None
+    y = _DwConv2d.apply(x, weight, sh, sw, ph, pw, holder)
     if bias is not None:
         y = y + bias.to(y.dtype).view(1, -1, 1, 1)
+    elif holder:
+        y._dfd_bn_stats = (holder[0],
+                           y.shape[0] * y.shape[2] * y.shape[3], y.shape[1])
     return y

@@ -121,20 +121,19 @@ static inline int dfd_pick_vec(long long c, int elem_size) {
 // rows into chunks. Chunks aim for ~kMaxGrid blocks but keep ≥min_iters
 // row-iterations per thread so per-block reductions/atomics amortize.
 struct DfdPlan {
-  int log2_cpb, ctiles, chunks, rows_per_chunk;
+  int cpb, ctiles, chunks, rows_per_chunk;
 };
 
+// Channel slots are NOT rounded up to a power of two (that idled up to 44%
+// of a block's load bandwidth on C=144-class layers); tiles balance.
 static inline DfdPlan dfd_plan(int cv, long long rows, long long base_blocks = 1,
                                int min_iters = 16) {
   DfdPlan p;
-  int l = 0;
-  while ((1 << l) < cv && l < 6) ++l;
-  p.log2_cpb = l;
-  const int cpb = 1 << l;
-  p.ctiles = (cv + cpb - 1) / cpb;
+  p.ctiles = (cv + 63) / 64;
+  p.cpb = (cv + p.ctiles - 1) / p.ctiles;
   const long long base = base_blocks * p.ctiles;
   long long want = (kMaxGrid + base - 1) / base;
-  const int nrg = 256 >> l;
+  const int nrg = 256 / p.cpb;
   long long by_iters = rows / ((long long)nrg * min_iters);
   if (want > by_iters) want = by_iters;
   long long max_chunks = (rows + nrg - 1) / nrg;

@@ -215,6 +215,149 @@ __global__ void dw_fwd_s1_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// stats-emitting stride-1 forward (k3): bn-style (channel-slot, row-group)
+// block layout — each thread's channel slice is FIXED, so the per-channel
+// sum/sumsq of y fold in LDS and land in the bucketed stats buffers the
+// following BatchNorm consumes (skipping its stats pass over y). k5 is
+// excluded: a fixed channel makes the compiler hoist the whole K*K weight
+// tile (k5: 100 VGPRs) and the occupancy loss measured slower than the
+// separate stats pass it saves.
+// ---------------------------------------------------------------------------
+constexpr int kDwStatsBuckets = 64;
+
+template <typename T, int K, int VEC, int TW>
+__global__ void dw_fwd_s1_stats_kernel(
+    const T* __restrict__ x, const T* __restrict__ w, T* __restrict__ y,
+    float* __restrict__ stats,  // [kDwStatsBuckets, 2, C]
+    int N, int C, int H, int W, int Ho, int Wo, int ph, int pw,
+    int cpb, long long rows_per_chunk) {
+  extern __shared__ float lds[];  // [256 * VEC]
+  const int slot = threadIdx.x % cpb;
+  const int rg = threadIdx.x / cpb;
+  const int nrg = blockDim.x / cpb;
+  const int cv = C / VEC;
+  const int cvec = blockIdx.x * cpb + slot;
+  const bool active = cvec < cv && rg < nrg;
+  const int c = cvec * VEC;
+  const int wt = (Wo + TW - 1) / TW;
+  const long long rows = (long long)N * Ho * wt;
+
+  float s[VEC], q[VEC];
+#pragma unroll
+  for (int i = 0; i < VEC; ++i) { s[i] = 0.f; q[i] = 0.f; }
+
+  if (active) {
+    const long long r0 = (long long)blockIdx.y * rows_per_chunk;
+    const long long r1 = min(r0 + rows_per_chunk, rows);
+    for (long long r = r0 + rg; r < r1; r += nrg) {
+      const int wo0 = (int)(r % wt) * TW;
+      long long p = r / wt;
+      const int ho = (int)(p % Ho);
+      const int n = (int)(p / Ho);
+
+      float acc[TW][VEC];
+#pragma unroll
+      for (int t = 0; t < TW; ++t)
+#pragma unroll
+        for (int i = 0; i < VEC; ++i) acc[t][i] = 0.f;
+
+      const int hi0 = ho - ph;
+      const int wi0 = wo0 - pw;
+      const bool interior = wi0 >= 0 && wi0 + K + TW - 2 < W;
+#pragma unroll
+      for (int kh = 0; kh < K; ++kh) {
+        const int hi = hi0 + kh;
+        if (hi < 0 || hi >= H) continue;
+        const T* xrow = x + (((long long)n * H + hi) * W) * C + c;
+        const T* wrow = w + ((long long)kh * K) * C + c;
+        if (interior) {
+          TVec<T, VEC> xv[K + TW - 1];
+#pragma unroll
+          for (int col = 0; col < K + TW - 1; ++col)
+            xv[col] = vload<T, VEC>(xrow + (long long)(wi0 + col) * C);
+#pragma unroll
+          for (int col = 0; col < K + TW - 1; ++col)
+#pragma unroll
+            for (int t = 0; t < TW; ++t) {
+              const int kw = col - t;
+              if (kw < 0 || kw >= K) continue;
+              const TVec<T, VEC> wv = vload<T, VEC>(wrow + (long long)kw * C);
+#pragma unroll
+              for (int i = 0; i < VEC; ++i)
+                acc[t][i] += DfdCvt<T>::to_f32(xv[col].v[i]) * DfdCvt<T>::to_f32(wv.v[i]);
+            }
+        } else {
+#pragma unroll
+          for (int col = 0; col < K + TW - 1; ++col) {
+            const int wi = wi0 + col;
+            if (wi < 0 || wi >= W) continue;
+            const TVec<T, VEC> xv = vload<T, VEC>(xrow + (long long)wi * C);
+#pragma unroll
+            for (int t = 0; t < TW; ++t) {
+              const int kw = col - t;
+              if (kw < 0 || kw >= K) continue;
+              const TVec<T, VEC> wv = vload<T, VEC>(wrow + (long long)kw * C);
+#pragma unroll
+              for (int i = 0; i < VEC; ++i)
+                acc[t][i] += DfdCvt<T>::to_f32(xv.v[i]) * DfdCvt<T>::to_f32(wv.v[i]);
+            }
+          }
+        }
+      }
+      T* yrow = y + (((long long)n * Ho + ho) * Wo) * C + c;
+#pragma unroll
+      for (int t = 0; t < TW; ++t) {
+        if (wo0 + t >= Wo) break;
+        TVec<T, VEC> yv;
+#pragma unroll
+        for (int i = 0; i < VEC; ++i) {
+          yv.v[i] = DfdCvt<T>::from_f32(acc[t][i]);
+          const float f = DfdCvt<T>::to_f32(yv.v[i]);  // rounded value
+          s[i] += f;
+          q[i] += f * f;
+        }
+        vstore<T, VEC>(yrow + (long long)(wo0 + t) * C, yv);
+      }
+    }
+  }
+
+  // fold row-groups in LDS (non-pow2-aware), one bucketed atomic set per block
+  float* my = lds + (size_t)(rg * cpb + slot) * VEC;
+  int p2 = 1;
+  while (p2 * 2 <= nrg) p2 *= 2;
+  const bool in_block = rg < nrg;
+#pragma unroll
+  for (int pass = 0; pass < 2; ++pass) {
+    __syncthreads();
+    if (in_block) {
+#pragma unroll
+      for (int i = 0; i < VEC; ++i) my[i] = pass == 0 ? s[i] : q[i];
+    }
+    __syncthreads();
+    if (in_block && rg >= p2) {
+      float* dst = lds + (size_t)((rg - p2) * cpb + slot) * VEC;
+#pragma unroll
+      for (int i = 0; i < VEC; ++i) dst[i] += my[i];
+    }
+    __syncthreads();
+    for (int step = p2 >> 1; step > 0; step >>= 1) {
+      if (rg < step) {
+        const float* other = lds + ((size_t)((rg + step) * cpb) + slot) * VEC;
+#pragma unroll
+        for (int i = 0; i < VEC; ++i) my[i] += other[i];
+      }
+      __syncthreads();
+    }
+    if (rg == 0 && cvec < cv) {
+      float* out = stats + (size_t)(blockIdx.y & (kDwStatsBuckets - 1)) * 2 * C +
+                   (pass == 0 ? 0 : C);
+#pragma unroll
+      for (int i = 0; i < VEC; ++i) atomicAdd(out + c + i, my[i]);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // backward data:
 // dx[n,hi,wi,c] = sum over (kh,kw) with hi = ho*sh-ph+kh solvable:
 //                 dy[n,ho,wo,c] * w[kh,kw,c]
@@ -584,8 +727,32 @@ Geom make_geom(int64_t N, int64_t C, int64_t H, int64_t W, int64_t K, int64_t sh
 }  // namespace
 
 // x: NCHW logical, channels_last physical. w_packed: (K, K, C) same dtype.
+template <typename T>
+void launch_fwd_s1_stats(const at::Tensor& x, const at::Tensor& w, at::Tensor& y,
+                         at::Tensor& stats, const Geom& g, int vec, hipStream_t stream) {
+  constexpr int TW = 4;
+  const int wt = (g.Wo + TW - 1) / TW;
+  const long long rows = (long long)g.N * g.Ho * wt;
+  const int cv = g.C / vec;
+  DfdPlan plan = dfd_plan(cv, rows);
+  dim3 grid(plan.ctiles, plan.chunks);
+  const int lds = 256 * vec * sizeof(float);
+#define DW_STATS_V(V)                                                          \
+  dw_fwd_s1_stats_kernel<T, 3, V, TW><<<grid, 256, lds, stream>>>(             \
+      (const T*)x.data_ptr(), (const T*)w.data_ptr(), (T*)y.data_ptr(),        \
+      stats.data_ptr<float>(), g.N, g.C, g.H, g.W, g.Ho, g.Wo, g.ph, g.pw,     \
+      plan.cpb, plan.rows_per_chunk)
+  switch (vec) {
+    case 8: DW_STATS_V(8); break;
+    case 4: DW_STATS_V(4); break;
+    case 2: DW_STATS_V(2); break;
+    default: DW_STATS_V(1); break;
+  }
+#undef DW_STATS_V
+}
+
 at::Tensor dw_conv2d_fwd(at::Tensor x, at::Tensor w_packed, int64_t sh, int64_t sw,
-                         int64_t ph, int64_t pw) {
+                         int64_t ph, int64_t pw, c10::optional<at::Tensor> stats_opt) {
   TORCH_CHECK(x.is_cuda() && w_packed.is_cuda(), "dwconv: CUDA tensors required");
   TORCH_CHECK(x.is_contiguous(at::MemoryFormat::ChannelsLast), "dwconv: x must be channels_last");
   TORCH_CHECK(w_packed.is_contiguous(), "dwconv: packed weight must be contiguous");
@@ -599,6 +766,23 @@ at::Tensor dw_conv2d_fwd(at::Tensor x, at::Tensor w_packed, int64_t sh, int64_t 
   auto stream = at::hip::getCurrentHIPStream().stream();
   const auto stype = x.scalar_type();
   const int Ki = (int)K;
+  if (stats_opt.has_value()) {
+    at::Tensor stats = *stats_opt;
+    TORCH_CHECK(Ki == 3 && sh == 1 && sw == 1,
+                "dwconv: stats epilogue only for k3 stride-1");
+    TORCH_CHECK(stats.scalar_type() == at::kFloat && stats.is_contiguous() &&
+                    stats.numel() == (long long)kDwStatsBuckets * 2 * g.C,
+                "dwconv: stats must be fp32 [64, 2, C]");
+    switch (stype) {
+      case at::kBFloat16:
+        launch_fwd_s1_stats<__hip_bfloat16>(x, w_packed, y, stats, g, vec, stream);
+        break;
+      case at::kHalf: launch_fwd_s1_stats<__half>(x, w_packed, y, stats, g, vec, stream); break;
+      case at::kFloat: launch_fwd_s1_stats<float>(x, w_packed, y, stats, g, vec, stream); break;
+      default: TORCH_CHECK(false, "dwconv: unsupported dtype");
+    }
+    return y;
+  }
   switch (stype) {
     case at::kBFloat16: fwd_ktype<__hip_bfloat16>(x, w_packed, y, g, Ki, vec, stream); break;
     case at::kHalf: fwd_ktype<__half>(x, w_packed, y, g, Ki, vec, stream); break;

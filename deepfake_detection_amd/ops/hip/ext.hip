@@ -35,7 +35,7 @@ void adamw_multi_tensor(std::vector<at::Tensor> params, std::vector<at::Tensor> 
 void ema_multi_tensor(std::vector<at::Tensor> ema_params, std::vector<at::Tensor> model_params,
                       double decay);
 at::Tensor dw_conv2d_fwd(at::Tensor x, at::Tensor w_packed, int64_t sh, int64_t sw,
-                         int64_t ph, int64_t pw);
+                         int64_t ph, int64_t pw, c10::optional<at::Tensor> stats);
 at::Tensor dw_conv2d_bwd_data(at::Tensor dy, at::Tensor w_packed, int64_t H, int64_t W,
                               int64_t sh, int64_t sw, int64_t ph, int64_t pw);
 at::Tensor dw_conv2d_bwd_weight(at::Tensor dy, at::Tensor x, int64_t K, int64_t sh,
@@ -64,7 +64,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("rmsprop_tf_multi_tensor", &rmsprop_tf_multi_tensor, "fused RMSpropTF step");
   m.def("adamw_multi_tensor", &adamw_multi_tensor, "fused AdamW step");
   m.def("ema_multi_tensor", &ema_multi_tensor, "fused EMA update");
-  m.def("dw_conv2d_fwd", &dw_conv2d_fwd, "depthwise conv2d forward (NHWC)");
+  m.def("dw_conv2d_fwd", &dw_conv2d_fwd, "depthwise conv2d forward (NHWC)",
+        py::arg("x"), py::arg("w_packed"), py::arg("sh"), py::arg("sw"),
+        py::arg("ph"), py::arg("pw"), py::arg("stats") = py::none());
   m.def("dw_conv2d_bwd_data", &dw_conv2d_bwd_data, "depthwise conv2d bwd data (NHWC)");
   m.def("dw_conv2d_bwd_weight", &dw_conv2d_bwd_weight, "depthwise conv2d bwd weight (NHWC)");
   m.def("pw_conv2d_fwd_mfma", &pw_conv2d_fwd_mfma, "1x1 conv as MFMA GEMM (NHWC)",

@@ -571,3 +571,29 @@ def test_ir_block_drop_path_stays_fused_and_zeroes_samples():
           for i in range(32)]
     assert any(eq), "no sample dropped at rate 0.9 (mask not applied)"
     assert not all(eq), "all samples dropped (scale wrong)"
+
+
+def test_dwconv_stats_epilogue_matches_direct():
+    """k3 s1 depthwise stats variant: output must match the plain kernel and
+    the folded per-channel sums must match a direct reduction over y."""
+    from deepfake_detection_amd.ops.dwconv import dw_conv2d
+
+    torch.manual_seed(13)
+    B, C, H = 3, 96, 37
+    x = _cl(torch.randn(B, C, H, H, device="cuda", dtype=torch.bfloat16))
+    w = torch.randn(C, 1, 3, 3, device="cuda", dtype=torch.bfloat16)
+    y = dw_conv2d(x, w, None, 1, 1, 1, want_stats=True)
+    y0 = dw_conv2d(x, w, None, 1, 1, 1)
+    assert torch.equal(y, y0), "stats variant changed the conv output"
+    assert hasattr(y, "_dfd_bn_stats")
+    buckets, m, c = y._dfd_bn_stats
+    assert m == B * H * H and c == C
+    ys = y.detach().float()
+    assert torch.allclose(buckets[:, 0].sum(0), ys.sum(dim=(0, 2, 3)),
+                          atol=0.5, rtol=1e-3)
+    assert torch.allclose(buckets[:, 1].sum(0), (ys * ys).sum(dim=(0, 2, 3)),
+                          atol=1.0, rtol=1e-3)
+    # k5 must NOT emit (excluded by design)
+    w5 = torch.randn(C, 1, 5, 5, device="cuda", dtype=torch.bfloat16)
+    y5 = dw_conv2d(x, w5, None, 1, 2, 1, want_stats=True)
+    assert not hasattr(y5, "_dfd_bn_stats")
