@@ -87,11 +87,16 @@ def distribute_bn(model, world_size, reduce=False):
                 dist.broadcast(bn_buf, 0)
 
 
-def wrap_ddp(model, device, find_unused_parameters=False, bucket_cap_mb=XGMI_BUCKET_CAP_MB,
+def wrap_ddp(model, device, find_unused_parameters=False, bucket_cap_mb=None,
              static_graph=True):
     """Wrap in torch DDP configured for RCCL/xGMI: bucketed all-reduce
     overlapped with backward, gradients viewed into flat buckets (no extra
-    copy), static graph (fixed CNN) enables bucket-order capture."""
+    copy), static graph (fixed CNN) enables bucket-order capture.
+
+    Bucket size: DFD_AMD_BUCKET_MB env > explicit arg > 60 MB xGMI default
+    (tools/sweep_buckets.sh measures the curve on an 8-GPU node)."""
+    if bucket_cap_mb is None:
+        bucket_cap_mb = int(os.environ.get("DFD_AMD_BUCKET_MB", XGMI_BUCKET_CAP_MB))
     device_ids = [device.index] if device.type == "cuda" else None
     ddp = torch.nn.parallel.DistributedDataParallel(
         model,
