@@ -18,8 +18,8 @@ import torch
 import torch.nn as nn
 
 from ..ops import functional as O
-from .layers import (DepthwiseConv2d, PointwiseConv2d, create_conv2d,
-                     drop_path, sigmoid)
+from .layers import (DepthwiseConv2d, PointwiseConv2d, StemConv2d,
+                     create_conv2d, drop_path, sigmoid)
 
 __all__ = [
     "BN_MOMENTUM_TF_DEFAULT",
@@ -161,7 +161,7 @@ class ConvBnAct(nn.Module):
 def _mark_bn_producer(conv):
     """Ask a conv whose output feeds a fused BatchNorm to emit the BN stats
     from its epilogue — the BN then skips its own stats pass over y."""
-    if isinstance(conv, (PointwiseConv2d, DepthwiseConv2d)):
+    if isinstance(conv, (PointwiseConv2d, DepthwiseConv2d, StemConv2d)):
         conv.emit_bn_stats = True
 
 

@@ -125,6 +125,7 @@ class EfficientNet(nn.Module):
         self.conv_head = create_conv2d(self._in_chs, self.num_features, 1, padding=pad_type)
         from .blocks import _mark_bn_producer as _mark
         _mark(self.conv_head)
+        _mark(self.conv_stem)
         self.bn2 = norm_layer(self.num_features, **norm_kwargs)
         self.act2 = act_layer(inplace=True)
         self.global_pool = SelectAdaptivePool2d(pool_type=global_pool)

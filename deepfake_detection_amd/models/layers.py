@@ -26,6 +26,7 @@ __all__ = [
     "Conv2dSame",
     "DepthwiseConv2d",
     "PointwiseConv2d",
+    "StemConv2d",
     "create_conv2d_pad",
     "MixedConv2d",
     "CondConv2d",
@@ -160,6 +161,31 @@ class PointwiseConv2d(nn.Conv2d):
         return super().forward(x)
 
 
+class StemConv2d(nn.Conv2d):
+    """Small-C_in strided conv (the EfficientNet stem) routed to the gfx950
+    implicit-GEMM MFMA kernels (ops/stemconv.py) — the last convs that
+    otherwise fell to MIOpen. Same state_dict as nn.Conv2d. `emit_bn_stats`
+    (set by the owning model) also emits the following BatchNorm's
+    per-channel stats from the epilogue."""
+
+    emit_bn_stats = False
+
+    def forward(self, x):
+        if x.is_cuda and not x.requires_grad:
+            from ..ops.extension import gpu_ops_required
+            from ..ops.stemconv import stem_supported
+
+            if gpu_ops_required() and stem_supported(
+                    x, self.weight, self.stride, self.padding, self.dilation,
+                    self.groups):
+                from ..ops.stemconv import stem_conv2d
+
+                return stem_conv2d(x, self.weight, self.bias, self.stride,
+                                   self.padding,
+                                   want_stats=self.emit_bn_stats and self.training)
+        return super().forward(x)
+
+
 def get_padding_value(padding, kernel_size, **kwargs) -> Tuple[object, bool]:
     stride = kwargs.get("stride", 1)
     dilation = kwargs.get("dilation", 1)
@@ -190,6 +216,9 @@ def create_conv2d_pad(in_chs, out_chs, kernel_size, **kwargs):
     ks = kernel_size[0] if isinstance(kernel_size, (tuple, list)) else kernel_size
     if ks == 1 and kwargs.get("groups", 1) == 1:
         return PointwiseConv2d(in_chs, out_chs, kernel_size, padding=padding, **kwargs)
+    if (ks == 3 and kwargs.get("stride", 1) == 2 and in_chs <= 16
+            and kwargs.get("groups", 1) == 1):
+        return StemConv2d(in_chs, out_chs, kernel_size, padding=padding, **kwargs)
     return nn.Conv2d(in_chs, out_chs, kernel_size, padding=padding, **kwargs)
 
 

@@ -43,6 +43,12 @@ at::Tensor dw_conv2d_bwd_weight(at::Tensor dy, at::Tensor x, int64_t K, int64_t 
 at::Tensor pw_conv2d_fwd_mfma(at::Tensor x, at::Tensor w,
                               c10::optional<at::Tensor> stats);
 at::Tensor pw_conv2d_bwd_weight_mfma(at::Tensor dy, at::Tensor x);
+at::Tensor stem_conv2d_fwd(at::Tensor x, at::Tensor w_packed, int64_t n_out,
+                           int64_t kh, int64_t kw, int64_t sh, int64_t sw,
+                           int64_t ph, int64_t pw, c10::optional<at::Tensor> stats);
+at::Tensor stem_conv2d_bwd_weight(at::Tensor dy, at::Tensor x, int64_t kh,
+                                  int64_t kw, int64_t sh, int64_t sw, int64_t ph,
+                                  int64_t pw);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "deepfake_detection_amd gfx950 (MI355X/CDNA4) kernels";
@@ -73,4 +79,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("x"), py::arg("w"), py::arg("stats") = py::none());
   m.def("pw_conv2d_bwd_weight_mfma", &pw_conv2d_bwd_weight_mfma,
         "1x1 conv weight grad: split-M MFMA + fp32 chunk reduce");
+  m.def("stem_conv2d_fwd", &stem_conv2d_fwd, "stem conv implicit-GEMM fwd (NHWC)",
+        py::arg("x"), py::arg("w_packed"), py::arg("n_out"), py::arg("kh"),
+        py::arg("kw"), py::arg("sh"), py::arg("sw"), py::arg("ph"), py::arg("pw"),
+        py::arg("stats") = py::none());
+  m.def("stem_conv2d_bwd_weight", &stem_conv2d_bwd_weight,
+        "stem conv implicit-GEMM weight grad");
 }

@@ -597,3 +597,52 @@ def test_dwconv_stats_epilogue_matches_direct():
     w5 = torch.randn(C, 1, 5, 5, device="cuda", dtype=torch.bfloat16)
     y5 = dw_conv2d(x, w5, None, 1, 2, 1, want_stats=True)
     assert not hasattr(y5, "_dfd_bn_stats")
+
+
+@pytest.mark.parametrize("cin,n,img", [(3, 48, 61), (12, 256, 38)])
+def test_stem_conv_matches_torch(cin, n, img):
+    """Implicit-GEMM stem fwd + bwd-weight vs fp32 torch conv (k3 s2 p1)."""
+    from deepfake_detection_amd.ops.stemconv import stem_conv2d
+
+    torch.manual_seed(14)
+    B = 3
+    x = _cl(torch.randn(B, cin, img, img, device="cuda", dtype=torch.bfloat16))
+    w = torch.randn(n, cin, 3, 3, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = stem_conv2d(x, w, stride=2, padding=1, want_stats=True)
+    ref = torch.nn.functional.conv2d(x.float(), w.detach().float(), None, 2, 1)
+    assert y.shape == ref.shape
+    assert torch.allclose(y.float(), ref, atol=0.25, rtol=0.05), (
+        (y.float() - ref).abs().max().item())
+    # stats match a direct reduction
+    buckets, m, c = y._dfd_bn_stats
+    ys = y.detach().float()
+    assert torch.allclose(buckets[:, 0].sum(0), ys.sum(dim=(0, 2, 3)),
+                          atol=0.5, rtol=1e-3)
+    # bwd-weight
+    dy = torch.randn_like(y)
+    y.backward(dy)
+    rw = w.detach().float().requires_grad_(True)
+    ref2 = torch.nn.functional.conv2d(x.float(), rw, None, 2, 1)
+    ref2.backward(dy.float())
+    assert torch.allclose(w.grad.float(), rw.grad, atol=3.0, rtol=0.05), (
+        (w.grad.float() - rw.grad).abs().max().item())
+
+
+def test_stem_module_routes_and_model_runs():
+    """B4's conv_stem is a StemConv2d; a full bf16 train microstep runs the
+    MFMA stem (no-grad input) and produces finite grads."""
+    import deepfake_detection_amd as dfd
+    from deepfake_detection_amd.models.layers import StemConv2d
+
+    torch.manual_seed(15)
+    m = dfd.create_model("efficientnet_b0", num_classes=2).cuda()
+    m = m.to(memory_format=torch.channels_last)
+    assert isinstance(m.conv_stem, StemConv2d)
+    x = torch.randn(4, 3, 65, 65, device="cuda").to(
+        memory_format=torch.channels_last)
+    with torch.autocast("cuda", torch.bfloat16):
+        out = m(x)
+        loss = out.float().sum()
+    loss.backward()
+    assert torch.isfinite(m.conv_stem.weight.grad).all()
