@@ -85,9 +85,10 @@ __global__ __launch_bounds__(256) void stem_fwd_kernel(
 
   const int ccol = lane & 15;
   const int crow0 = (lane >> 4) * 4;
-  const long long mtiles = (M + 255) / 256;
+  // block M-tile = 4 waves x FI*16 = 128 rows
+  const long long mtiles = (M + 127) / 128;
   for (long long mt = blockIdx.x; mt < mtiles; mt += gridDim.x) {
-    const long long m0 = mt * 256 + wid * (FI * 16);
+    const long long m0 = mt * 128 + wid * (FI * 16);
     f32x4 acc[FI][FJ];
 #pragma unroll
     for (int i = 0; i < FI; ++i)
@@ -339,7 +340,7 @@ at::Tensor stem_conv2d_fwd(at::Tensor x, at::Tensor w_packed, int64_t n_out,
   auto y = at::empty({(long long)g.B, (long long)g.N, (long long)g.Ho, (long long)g.Wo},
                      x.options().memory_format(at::MemoryFormat::ChannelsLast));
   auto stream = at::hip::getCurrentHIPStream().stream();
-  const long long mtiles = (M + 255) / 256;
+  const long long mtiles = (M + 127) / 128;
   dim3 grid((unsigned)(mtiles < 8192 ? mtiles : 8192), (g.N + 63) / 64);
   float* stats_p = nullptr;
   if (stats_opt.has_value()) {
