@@ -41,23 +41,41 @@ struct StemGeom {
   int sh, sw, ph, pw;
 };
 
-// gather one patch element: output pixel m, contraction index k
-template <typename T>
-DFD_DEV float stem_patch_el(const T* __restrict__ x, const StemGeom& g,
-                            long long m, int k) {
+// decode helpers: the k-triples (c, kw, kh) are loop-invariant per thread —
+// hoist them out of the m-loop (packed c | kw<<8 | kh<<16, or -1 for the
+// zero-pad tail); the (b, ho, wo) decode happens once per output row.
+DFD_DEV int stem_ktrip(const StemGeom& g, int k) {
+  if (k >= g.KH * g.KW * g.Cin) return -1;
   const int c = k % g.Cin;
   const int cell = k / g.Cin;
-  const int kw = cell % g.KW;
-  const int kh = cell / g.KW;
-  const int wo = (int)(m % g.Wo);
-  long long t = m / g.Wo;
-  const int ho = (int)(t % g.Ho);
-  const int b = (int)(t / g.Ho);
-  const int hi = ho * g.sh - g.ph + kh;
-  const int wi = wo * g.sw - g.pw + kw;
+  return c | ((cell % g.KW) << 8) | ((cell / g.KW) << 16);
+}
+
+struct StemRow {
+  int b, hi0, wi0;  // input-space origin of the output pixel's patch
+};
+
+DFD_DEV StemRow stem_row(const StemGeom& g, long long m) {
+  StemRow r;
+  const unsigned mu = (unsigned)m;  // M < 2^32 for any realistic batch
+  const unsigned wo = mu % (unsigned)g.Wo;
+  const unsigned t = mu / (unsigned)g.Wo;
+  const unsigned ho = t % (unsigned)g.Ho;
+  r.b = (int)(t / (unsigned)g.Ho);
+  r.hi0 = (int)ho * g.sh - g.ph;
+  r.wi0 = (int)wo * g.sw - g.pw;
+  return r;
+}
+
+template <typename T>
+DFD_DEV float stem_patch_el(const T* __restrict__ x, const StemGeom& g,
+                            const StemRow& r, int trip) {
+  if (trip < 0) return 0.f;
+  const int hi = r.hi0 + ((trip >> 16) & 0xff);
+  const int wi = r.wi0 + ((trip >> 8) & 0xff);
   if (hi < 0 || hi >= g.H || wi < 0 || wi >= g.W) return 0.f;
   return DfdCvt<T>::to_f32(
-      x[(((long long)b * g.H + hi) * g.W + wi) * g.Cin + c]);
+      x[(((long long)r.b * g.H + hi) * g.W + wi) * g.Cin + (trip & 0xff)]);
 }
 
 // ---------------------------------------------------------------------------
@@ -95,7 +113,16 @@ __global__ __launch_bounds__(256) void stem_fwd_kernel(
 #pragma unroll
       for (int j = 0; j < FJ; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
+    StemRow rows[FI];
+#pragma unroll
+    for (int i = 0; i < FI; ++i) {
+      const long long gm = m0 + i * 16 + lrow;
+      rows[i] = stem_row(g, gm < M ? gm : 0);
+    }
     for (int k0 = 0; k0 < g.Kpad; k0 += 32) {
+      int trips[8];
+#pragma unroll
+      for (int e = 0; e < 8; ++e) trips[e] = stem_ktrip(g, k0 + lk + e);
       bf16x8 afrag[FI], bfrag[FJ];
 #pragma unroll
       for (int i = 0; i < FI; ++i) {
@@ -103,12 +130,9 @@ __global__ __launch_bounds__(256) void stem_fwd_kernel(
         afrag[i] = bf16x8{};
         if (gm < M) {
 #pragma unroll
-          for (int e = 0; e < 8; ++e) {
-            const int k = k0 + lk + e;
+          for (int e = 0; e < 8; ++e)
             reinterpret_cast<__bf16*>(&afrag[i])[e] =
-                (__bf16)__float2bfloat16(
-                    k < g.KH * g.KW * g.Cin ? stem_patch_el<T>(x, g, gm, k) : 0.f);
-          }
+                (__bf16)__float2bfloat16(stem_patch_el<T>(x, g, rows[i], trips[e]));
         }
       }
 #pragma unroll
@@ -212,7 +236,6 @@ __global__ __launch_bounds__(256) void stem_wgrad_kernel(
   const int sm = tid & 63;       // m-local
   const int sq_ = tid >> 6;      // 16-col group
 
-  const int kk_real = g.KH * g.KW * g.Cin;
   for (long long m0 = r0; m0 < r1; m0 += STM) {
     const long long gm = m0 + sm;
     // dy tile: vec8 along n, transposed into [n][m]
@@ -238,16 +261,17 @@ __global__ __launch_bounds__(256) void stem_wgrad_kernel(
           dyt[(sq_ * 16 + h * 8 + e) * SLDM + sm] = reinterpret_cast<__bf16*>(&v)[e];
       }
     }
-    // patch tile: gathered scalar into [k][m]
+    // patch tile: gathered scalar into [k][m] (k-triples hoisted, row
+    // decode once per m)
     {
+      const StemRow row = stem_row(g, gm < r1 ? gm : 0);
 #pragma unroll
       for (int h = 0; h < 2; ++h) {
         const int kbase = k0 + sq_ * 16 + h * 8;
 #pragma unroll
         for (int e = 0; e < 8; ++e) {
-          const int k = kbase + e;
           float f = 0.f;
-          if (gm < r1 && k < kk_real) f = stem_patch_el<T>(x, g, gm, k);
+          if (gm < r1) f = stem_patch_el<T>(x, g, row, stem_ktrip(g, kbase + e));
           xt[(sq_ * 16 + h * 8 + e) * SLDM + sm] = (__bf16)__float2bfloat16(f);
         }
       }
