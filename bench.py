@@ -27,17 +27,19 @@ def parse_args():
     p.add_argument("--model", default="efficientnet_b4")
     p.add_argument("--img-size", type=int, default=299)
     p.add_argument("--in-chans", type=int, default=3)
-    # 288 GB HBM3E fits >=768/GPU at B4-299 bf16 (2151 img/s measured) but
-    # MIOpen's one-time find for the batch-768 1x1 conv shapes costs several
-    # minutes on a fresh box; 384 keeps the default run inside a few minutes
-    # (1870 img/s). Pass --batch-size 768 for the full-memory number.
-    p.add_argument("--batch-size", type=int, default=384, help="per-GPU micro-batch")
+    # per-GPU micro-batch sized for 288 GB HBM3E: 768 at B4-299 bf16
+    # (2336 img/s r02). The r01 default of 384 only existed to dodge MIOpen
+    # find time — every conv now runs on in-tree HIP kernels, so a fresh
+    # process reaches the first timed step in seconds.
+    p.add_argument("--batch-size", type=int, default=768, help="per-GPU micro-batch")
     p.add_argument("--num-classes", type=int, default=2)
     p.add_argument("--opt", default="rmsproptf", choices=["rmsproptf", "adamw"])
     p.add_argument("--lr", type=float, default=1e-4)
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
-    p.add_argument("--graph", action="store_true", default=False,
-                   help="capture the train step in a hipGraph (single-GPU only)")
+    p.add_argument("--graph", action="store_true", default=True,
+                   help="capture the train step in a hipGraph (single-GPU only; "
+                        "auto-disabled when distributed)")
+    p.add_argument("--no-graph", dest="graph", action="store_false")
     p.add_argument("--channels-last", dest="channels_last", action="store_true", default=True)
     p.add_argument("--no-fused-ops", action="store_true", default=False,
                    help="A/B: run plain torch ops instead of the HIP kernels")
