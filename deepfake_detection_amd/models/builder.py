@@ -140,18 +140,21 @@ def _scale_stage_depth(stack_args, repeats, depth_multiplier=1.0, depth_trunc="c
     else:
         num_repeat_scaled = int(math.ceil(num_repeat * depth_multiplier))
 
-    repeats_scaled = []
-    for r in repeats[::-1]:
-        rs = max(1, round((r / num_repeat * num_repeat_scaled)))
-        repeats_scaled.append(rs)
-        num_repeat -= r
-        num_repeat_scaled -= rs
-    repeats_scaled = repeats_scaled[::-1]
+    # distribute the scaled budget back-to-front (exact-parity algorithm:
+    # the LAST block definitions absorb rounding first, so the stage's first
+    # block scales last)
+    remaining, budget = num_repeat, num_repeat_scaled
+    scaled = {}
+    for idx in range(len(repeats) - 1, -1, -1):
+        share = max(1, round(repeats[idx] / remaining * budget))
+        scaled[idx] = share
+        remaining -= repeats[idx]
+        budget -= share
 
-    sa_scaled = []
-    for ba, rep in zip(stack_args, repeats_scaled):
-        sa_scaled.extend([deepcopy(ba) for _ in range(rep)])
-    return sa_scaled
+    expanded = []
+    for idx, ba in enumerate(stack_args):
+        expanded.extend(deepcopy(ba) for _ in range(scaled[idx]))
+    return expanded
 
 
 def decode_arch_def(arch_def, depth_multiplier=1.0, depth_trunc="ceil", experts_multiplier=1):
@@ -300,35 +303,36 @@ class EfficientNetBuilder:
         return stages
 
 
+def _conv_fan_out(conv, fix_group_fanout):
+    fan = conv.kernel_size[0] * conv.kernel_size[1] * conv.out_channels
+    return fan // conv.groups if fix_group_fanout else fan
+
+
 def _init_weight_goog(m, n="", fix_group_fanout=True):
-    """TF-official fan-out normal init (reference efficientnet_builder.py:537-575)."""
+    """TF-official fan-out normal init; exact semantic parity with reference
+    efficientnet_builder.py:537-575 (normal(0, sqrt(2/fan_out)) convs,
+    unit-BN, uniform classifier with routing_fn fan-in special case)."""
     if isinstance(m, CondConv2d):
-        fan_out = m.kernel_size[0] * m.kernel_size[1] * m.out_channels
-        if fix_group_fanout:
-            fan_out //= m.groups
-        init_weight_fn = get_condconv_initializer(
-            lambda w: w.data.normal_(0, math.sqrt(2.0 / fan_out)), m.num_experts, m.weight_shape)
-        init_weight_fn(m.weight)
+        std = math.sqrt(2.0 / _conv_fan_out(m, fix_group_fanout))
+        expert_init = get_condconv_initializer(
+            lambda w: nn.init.normal_(w, 0, std), m.num_experts, m.weight_shape)
+        expert_init(m.weight)
         if m.bias is not None:
-            m.bias.data.zero_()
+            nn.init.zeros_(m.bias)
     elif isinstance(m, nn.Conv2d):
-        fan_out = m.kernel_size[0] * m.kernel_size[1] * m.out_channels
-        if fix_group_fanout:
-            fan_out //= m.groups
-        m.weight.data.normal_(0, math.sqrt(2.0 / fan_out))
+        std = math.sqrt(2.0 / _conv_fan_out(m, fix_group_fanout))
+        nn.init.normal_(m.weight, 0, std)
         if m.bias is not None:
-            m.bias.data.zero_()
+            nn.init.zeros_(m.bias)
     elif isinstance(m, nn.BatchNorm2d):
-        m.weight.data.fill_(1.0)
-        m.bias.data.zero_()
+        nn.init.ones_(m.weight)
+        nn.init.zeros_(m.bias)
     elif isinstance(m, nn.Linear):
         fan_out = m.weight.size(0)
-        fan_in = 0
-        if "routing_fn" in n:
-            fan_in = m.weight.size(1)
-        init_range = 1.0 / math.sqrt(fan_in + fan_out)
-        m.weight.data.uniform_(-init_range, init_range)
-        m.bias.data.zero_()
+        fan_in = m.weight.size(1) if "routing_fn" in n else 0
+        bound = 1.0 / math.sqrt(fan_in + fan_out)
+        nn.init.uniform_(m.weight, -bound, bound)
+        nn.init.zeros_(m.bias)
 
 
 def efficientnet_init_weights(model: nn.Module, init_fn=None):
