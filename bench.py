@@ -118,11 +118,29 @@ def main():
     std = std.view(1, args.in_chans, 1, 1)
     norm_dtype = torch.bfloat16 if use_bf16 else torch.float32
 
+    # fused classifier+CE head (ops/head.py) on the single-GPU path; DDP
+    # must go through run_model.forward for the reducer hooks
+    from deepfake_detection_amd.ops import functional as O
+    from deepfake_detection_amd.ops.extension import gpu_ops_required
+
+    use_fused_head = (use_cuda and not distributed and not args.no_fused_ops
+                      and gpu_ops_required()
+                      and hasattr(model, "forward_features")
+                      and isinstance(model.classifier, torch.nn.Linear))
+
     def one_step(i):
         x = normalize_uint8(pool[i % n_pool], mean, std, out_dtype=norm_dtype,
                             channels_last=args.channels_last and use_cuda)
         t = targets[i % n_pool]
-        if use_bf16:
+        if use_fused_head:
+            from deepfake_detection_amd.ops.head import fused_head_ce
+
+            with torch.autocast("cuda", torch.bfloat16, enabled=use_bf16):
+                feats = model.forward_features(x)
+                pooled = O.global_avg_pool(feats)
+            loss, _ = fused_head_ce(pooled, model.classifier.weight,
+                                    model.classifier.bias, t)
+        elif use_bf16:
             with torch.autocast("cuda", torch.bfloat16):
                 out = run_model(x)
                 loss = loss_fn(out, t)

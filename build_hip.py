@@ -16,7 +16,7 @@ SRC_DIR = os.path.join(REPO, "deepfake_detection_amd", "ops", "hip")
 OUT_SO = os.path.join(REPO, "deepfake_detection_amd", "_hip_ops.so")
 BUILD_DIR = os.path.join(REPO, "build", "hip")
 
-SOURCES = ["bn_act.hip", "normalize.hip", "pool.hip", "se.hip", "optim.hip", "dwconv.hip", "pwconv.hip", "stemconv.hip", "ext.hip"]
+SOURCES = ["bn_act.hip", "normalize.hip", "pool.hip", "se.hip", "optim.hip", "dwconv.hip", "pwconv.hip", "stemconv.hip", "head.hip", "ext.hip"]
 
 ARCH = os.environ.get("PYTORCH_ROCM_ARCH", "gfx950")
 

@@ -49,6 +49,11 @@ at::Tensor stem_conv2d_fwd(at::Tensor x, at::Tensor w_packed, int64_t n_out,
 at::Tensor stem_conv2d_bwd_weight(at::Tensor dy, at::Tensor x, int64_t kh,
                                   int64_t kw, int64_t sh, int64_t sw, int64_t ph,
                                   int64_t pw);
+std::vector<at::Tensor> head_ce_fwd(at::Tensor x, at::Tensor w, at::Tensor bias,
+                                    at::Tensor target, double smoothing);
+std::vector<at::Tensor> head_ce_bwd(at::Tensor dloss, at::Tensor logits,
+                                    at::Tensor x, at::Tensor w, at::Tensor target,
+                                    double smoothing);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "deepfake_detection_amd gfx950 (MI355X/CDNA4) kernels";
@@ -85,4 +90,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("stats") = py::none());
   m.def("stem_conv2d_bwd_weight", &stem_conv2d_bwd_weight,
         "stem conv implicit-GEMM weight grad");
+  m.def("head_ce_fwd", &head_ce_fwd, "fused classifier GEMM + smoothed-CE fwd");
+  m.def("head_ce_bwd", &head_ce_bwd, "fused classifier + CE backward");
 }

@@ -646,3 +646,34 @@ def test_stem_module_routes_and_model_runs():
         loss = out.float().sum()
     loss.backward()
     assert torch.isfinite(m.conv_stem.weight.grad).all()
+
+
+@pytest.mark.parametrize("smoothing", [0.0, 0.1])
+def test_fused_head_ce_matches_torch(smoothing):
+    """Fused classifier GEMM + smoothed-CE (fwd loss/logits + dx/dW/db) vs
+    the eager torch chain."""
+    from deepfake_detection_amd.ops.head import fused_head_ce
+
+    torch.manual_seed(16)
+    B, F, C = 37, 448, 2
+    x = torch.randn(B, F, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    w = torch.randn(C, F, device="cuda", requires_grad=True)
+    b = torch.randn(C, device="cuda", requires_grad=True)
+    t = torch.randint(0, C, (B,), device="cuda")
+
+    loss, logits = fused_head_ce(x, w, b, t, smoothing)
+    rx = x.detach().float().requires_grad_(True)
+    rw = w.detach().clone().requires_grad_(True)
+    rb = b.detach().clone().requires_grad_(True)
+    rlogits = torch.nn.functional.linear(rx, rw, rb)
+    logp = torch.nn.functional.log_softmax(rlogits, dim=-1)
+    nll = torch.nn.functional.nll_loss(logp, t)
+    rloss = (1 - smoothing) * nll + smoothing * (-logp.mean(dim=-1)).mean()
+    assert torch.allclose(logits, rlogits, atol=0.2, rtol=0.02)
+    assert abs(loss.item() - rloss.item()) < 0.02 * max(1.0, rloss.item())
+
+    loss.backward()
+    rloss.backward()
+    assert torch.allclose(w.grad, rw.grad, atol=0.02, rtol=0.05)
+    assert torch.allclose(b.grad, rb.grad, atol=0.01, rtol=0.05)
+    assert torch.allclose(x.grad.float(), rx.grad, atol=0.05, rtol=0.05)
