@@ -26,9 +26,12 @@ from .extension import load_extension
 STATS_BUCKETS = 64
 
 # Shapes (C_in, C_out) where MIOpen measured FASTER than the MFMA kernel on
-# MI355X (tools/bench_kernels.py --ops pw). Empty = MFMA wins everywhere
-# measured; entries are exceptions, not an allowlist, so unmeasured models
-# still get the native path.
+# MI355X (tools/bench_kernels.py --ops pw, 2026-09-14 sweep over every
+# B4-299 and deepfake_v4 shape). Only (288,48) and (672,112) were within
+# ~25% of MIOpen; keeping them on MFMA preserves the BN-stats epilogue
+# (worth more than the gap) and keeps MIOpen out of the step entirely.
+# Entries are exceptions, not an allowlist, so unmeasured models still get
+# the native path.
 _MIOPEN_FASTER: set = set()
 
 

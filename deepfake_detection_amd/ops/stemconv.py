@@ -17,7 +17,9 @@ STATS_BUCKETS = 64
 
 
 def stem_supported(x, weight, stride, padding, dilation, groups) -> bool:
-    if groups != 1 or x.dtype != torch.bfloat16:
+    # bf16/fp16 (fragments compute in bf16 either way); fp32 falls back to
+    # torch for full-precision parity
+    if groups != 1 or x.dtype not in (torch.bfloat16, torch.float16):
         return False
     n, cin, kh, kw = weight.shape
     if kh != kw or cin > 16:
