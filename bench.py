@@ -166,15 +166,24 @@ def main():
 
     graph = None
     if args.graph and use_cuda and not distributed:
-        # capture the whole train step in a hipGraph; replay K times
-        static_idx = 0
-        graph = torch.cuda.CUDAGraph()
-        # a couple extra eager steps on a side stream pool for allocator warmup
-        one_step(static_idx)
-        torch.cuda.synchronize()
-        with torch.cuda.graph(graph):
+        # capture the whole train step in a hipGraph; replay K times. If a
+        # fallback op in this model config is not capture-safe, log and run
+        # eager — the bench must never die on the capture path.
+        try:
+            static_idx = 0
+            graph = torch.cuda.CUDAGraph()
             one_step(static_idx)
-        barrier_sync()
+            torch.cuda.synchronize()
+            with torch.cuda.graph(graph):
+                one_step(static_idx)
+            barrier_sync()
+        except Exception as e:  # noqa: BLE001
+            import sys
+
+            print(f"[bench] hipGraph capture failed ({e}); running eager",
+                  file=sys.stderr)
+            graph = None
+            torch.cuda.synchronize()
 
     t0 = time.perf_counter()
     for i in range(args.steps):

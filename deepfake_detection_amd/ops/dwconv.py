@@ -11,11 +11,22 @@ want (K, K, C) so channel loads vectorize. The pack/unpack permutes touch
 C*K*K elements (≤ ~70 KB) — negligible next to the conv itself.
 """
 
+import os
+
 import torch
 
 from .extension import load_extension
 
 _SUPPORTED_K = (3, 5, 7, 9, 11)
+
+
+def dw_stats_enabled() -> bool:
+    """The k3 stats-epilogue variant is OFF by default: its fixed-channel
+    layout costs ~4 waves of occupancy (1451 vs 702 us on the bs768 k3
+    layers — r02 profile) and the dw kernels are latency-bound, so it loses
+    more than the separate bn_stats pass it saves. Kept behind
+    DFD_AMD_DW_STATS=1 for re-evaluation after a register-pressure fix."""
+    return os.environ.get("DFD_AMD_DW_STATS", "0") == "1"
 
 
 def dw_supported(weight, stride, padding, dilation, groups) -> bool:
@@ -44,7 +55,8 @@ class _DwConv2d(torch.autograd.Function):
         C, _, K, _ = weight.shape
         w_packed = weight.reshape(C, K, K).permute(1, 2, 0).contiguous()
         stats = None
-        if stats_out is not None and K == 3 and sh == 1 and sw == 1:
+        if (stats_out is not None and K == 3 and sh == 1 and sw == 1
+                and dw_stats_enabled()):
             # k3 s1 only: the stats variant fixes each thread's channel; at
             # k5 the register cost measured slower than the pass it saves
             stats = torch.zeros(STATS_BUCKETS, 2, C, device=x.device,

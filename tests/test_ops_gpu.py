@@ -573,9 +573,11 @@ def test_ir_block_drop_path_stays_fused_and_zeroes_samples():
     assert not all(eq), "all samples dropped (scale wrong)"
 
 
-def test_dwconv_stats_epilogue_matches_direct():
-    """k3 s1 depthwise stats variant: output must match the plain kernel and
-    the folded per-channel sums must match a direct reduction over y."""
+def test_dwconv_stats_epilogue_matches_direct(monkeypatch):
+    """k3 s1 depthwise stats variant (opt-in): output must match the plain
+    kernel and the folded per-channel sums must match a direct reduction
+    over y."""
+    monkeypatch.setenv("DFD_AMD_DW_STATS", "1")
     from deepfake_detection_amd.ops.dwconv import dw_conv2d
 
     torch.manual_seed(13)
