@@ -29,6 +29,19 @@ struct ChunkTable {
   int nchunks;
 };
 
+// Capture-safe table fill: the values travel as KERNEL ARGUMENTS (embedded
+// in the graph node), so building a table inside an active hipGraph capture
+// needs no host allocation and no memcpy — both of which invalidate a
+// capture (hipHostMalloc/pageable H2D).
+struct TableSlice {
+  long long v[480];
+};
+
+__global__ void fill_table_kernel(long long* __restrict__ dst, TableSlice s, int n) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) dst[i] = s.v[i];
+}
+
 ChunkTable build_chunks(const std::vector<at::Tensor>& a,
                         const std::vector<at::Tensor>& b,
                         const std::vector<at::Tensor>& c,
@@ -67,16 +80,32 @@ ChunkTable build_chunks(const std::vector<at::Tensor>& a,
       return t;
     }
   }
-  // PINNED host staging: under hipGraph capture the backward's grads live at
-  // graph-pool addresses, so the table is (re)built inside capture — a
-  // pageable H2D there is illegal. A pinned copy captures as a legal memcpy
-  // node; the cache keeps the pinned tensor alive for replays.
-  auto host = at::empty({(long long)nchunks, 5},
-                        at::TensorOptions().dtype(at::kLong).pinned_memory(true));
-  std::memcpy(host.data_ptr(), rows.data(), rows.size() * sizeof(long long));
   ChunkTable t;
-  t.dev = host.to(opts.device(), /*non_blocking=*/true);
   t.nchunks = nchunks;
+  hipStreamCaptureStatus cap = hipStreamCaptureStatusNone;
+  auto stream = at::hip::getCurrentHIPStream().stream();
+  (void)hipStreamIsCapturing(stream, &cap);
+  at::Tensor host;  // kept alive by the cache when the pinned path is used
+  if (cap != hipStreamCaptureStatusNone) {
+    // Inside capture (grads live at graph-pool addresses -> cache miss):
+    // device alloc is legal (graph pool); ship the values as kernel args.
+    t.dev = at::empty({(long long)nchunks, 5},
+                      at::TensorOptions().dtype(at::kLong).device(opts.device()));
+    long long* dst = (long long*)t.dev.data_ptr();
+    const long long total = (long long)rows.size();
+    for (long long off = 0; off < total; off += 480) {
+      TableSlice s;
+      const int n = (int)std::min<long long>(480, total - off);
+      std::memcpy(s.v, rows.data() + off, n * sizeof(long long));
+      fill_table_kernel<<<dim3((n + 255) / 256), 256, 0, stream>>>(dst + off, s, n);
+    }
+  } else {
+    // Eager path: pinned staging + one async H2D.
+    host = at::empty({(long long)nchunks, 5},
+                     at::TensorOptions().dtype(at::kLong).pinned_memory(true));
+    std::memcpy(host.data_ptr(), rows.data(), rows.size() * sizeof(long long));
+    t.dev = host.to(opts.device(), /*non_blocking=*/true);
+  }
   if (cache.size() > 8) cache.clear();  // bound: a few optimizers/EMA per process
   cache.push_back({std::move(rows), host, t.dev});
   return t;
