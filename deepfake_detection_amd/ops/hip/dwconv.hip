@@ -141,48 +141,6 @@ __global__ void dw_fwd_s1_kernel(
     const int hi0 = ho0 - ph;
     const int wi0 = wo0 - pw;
     const bool interior = wi0 >= 0 && wi0 + K + TW - 2 < W;
-    const bool vinterior = hi0 >= 0 && hi0 + K + TH - 2 < H;
-    // k3 only: at k5 the double row buffer (2 x 8 x 4 VGPRs) spills
-    if (K == 3 && interior && vinterior) {
-      // fully-interior fast path: software-pipeline the row loads — issue
-      // row r+1's K+TW-1 column loads into the alternate buffer while
-      // consuming row r (the single-row version ran 74% wave-parked on
-      // load latency; this doubles the bytes in flight per wave)
-      const T* xbase = x + (((long long)n * H + hi0) * W + wi0) * C + c;
-      TVec<T, VEC> xv[2][K + TW - 1];
-#pragma unroll
-      for (int col = 0; col < K + TW - 1; ++col)
-        xv[0][col] = vload<T, VEC>(xbase + (long long)col * C);
-#pragma unroll
-      for (int row = 0; row < K + TH - 1; ++row) {
-        const int cur = row & 1;
-        if (row + 1 < K + TH - 1) {
-          const T* xnext = xbase + (long long)(row + 1) * W * C;
-#pragma unroll
-          for (int col = 0; col < K + TW - 1; ++col)
-            xv[cur ^ 1][col] = vload<T, VEC>(xnext + (long long)col * C);
-        }
-#pragma unroll
-        for (int th = 0; th < TH; ++th) {
-          const int kh = row - th;
-          if (kh < 0 || kh >= K) continue;
-          const T* wrow = w + ((long long)kh * K) * C + c;
-#pragma unroll
-          for (int col = 0; col < K + TW - 1; ++col)
-#pragma unroll
-            for (int t = 0; t < TW; ++t) {
-              const int kw = col - t;
-              if (kw < 0 || kw >= K) continue;
-              const TVec<T, VEC> wv = vload<T, VEC>(wrow + (long long)kw * C);
-#pragma unroll
-              for (int i = 0; i < VEC; ++i)
-                acc[th][t][i] += DfdCvt<T>::to_f32(xv[cur][col].v[i]) *
-                                 DfdCvt<T>::to_f32(wv.v[i]);
-            }
-        }
-      }
-      goto writeback;
-    }
 #pragma unroll
     for (int row = 0; row < K + TH - 1; ++row) {
       const int hi = hi0 + row;
@@ -240,7 +198,6 @@ __global__ void dw_fwd_s1_kernel(
         }
       }
     }
-  writeback:
 #pragma unroll
     for (int th = 0; th < TH; ++th) {
       if (ho0 + th >= Ho) break;
@@ -609,11 +566,13 @@ void launch_fwd(const at::Tensor& x, const at::Tensor& w, at::Tensor& y, const G
                 hipStream_t stream) {
   const int block = 256;
   if (g.sh == 1 && g.sw == 1) {
-    // TW=4/TH=1 is the measured optimum so far: TH=2 halved x traffic but
-    // cost occupancy (570 vs 433 us avg k5 in-bench); TW=6/8 re-triggered
-    // the compiler's full weight-tile register hoist (spills).
+    // Per-K measured optimum (bs384 microbench, r02): k3 wants the 2-row
+    // tile (C=192: 0.695 ms vs 1.0+ at TH=1 — 104 VGPRs, 4 waves/SIMD);
+    // k5 wants max occupancy (TH=1, 82 VGPRs, 6 waves — its 2-row tile
+    // needs 172 VGPRs and ran 570 vs 433 us avg). Wider TW and pipelined
+    // row loads both re-triggered register hoists/spills and lost.
     constexpr int TW = 4;
-    constexpr int TH = 1;
+    constexpr int TH = K == 3 ? 2 : 1;
     const int wt = (g.Wo + TW - 1) / TW;
     const int ht = (g.Ho + TH - 1) / TH;
     const long long total = (long long)g.N * ht * wt * (g.C / VEC);
