@@ -103,49 +103,49 @@ __global__ void dw_fwd_kernel(const T* __restrict__ x, const T* __restrict__ w,
 }
 
 // ---------------------------------------------------------------------------
-// forward, stride 1: TH x TW output tile per thread (2-D).
-// PMC evidence (r02): the 1-D TW=4 version re-fetched x ~4x from HBM at
-// ~5.2 TB/s effective — bandwidth-saturated on redundant row re-reads. A
-// TH-row tile cuts the per-output x-row fetches from K to (K+TH-1)/TH; the
-// whole K*K weight tile is loaded once per grid-stride iteration (L1-hot)
-// instead of per (col, t) pair. bwd-data for stride 1 reuses this kernel
-// with the packed weight flipped in (kh,kw) and padding (K-1-p) — the same
-// stride-1 correlation.
+// NOTE (r02): this body is the round-1 kernel RESTORED VERBATIM. Five
+// re-tilings were measured against it on MI355X (2-row tiles, wider TW,
+// software-pipelined row loads, register-resident weights, per-K hybrids)
+// and every one lost in the full bench — the kernel is latency-bound and
+// extremely sensitive to compiler scheduling/occupancy; see BASELINE.md
+// "next levers" for the remaining ideas (LDS-staged weights).
+// forward, stride 1, TW consecutive outputs per thread.
+// Loads per output drop from K*K to ~K*(K+TW-1)/TW (x) and weight loads
+// amortize by TW; bwd-data for stride 1 reuses this kernel with the packed
+// weight flipped in (kh,kw) and padding (K-1-p) — it is the same stride-1
+// correlation.
 // ---------------------------------------------------------------------------
-template <typename T, int K, int VEC, int TW, int TH>
-__global__ void dw_fwd_s1_kernel(
-    const T* __restrict__ x, const T* __restrict__ w,
-    T* __restrict__ y, int N, int C, int H, int W,
-    int Ho, int Wo, int ph, int pw) {
+template <typename T, int K, int VEC, int TW>
+__global__ void dw_fwd_s1_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                                 T* __restrict__ y, int N, int C, int H, int W,
+                                 int Ho, int Wo, int ph, int pw) {
   const int cv = C / VEC;
   const int wt = (Wo + TW - 1) / TW;  // wo tiles per row
-  const int ht = (Ho + TH - 1) / TH;  // ho tiles per column
-  const long long total = (long long)N * ht * wt * cv;
+  const long long total = (long long)N * Ho * wt * cv;
   for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
        idx += (long long)gridDim.x * blockDim.x) {
     const int c = (int)(idx % cv) * VEC;
     long long p = idx / cv;
     const int wo0 = (int)(p % wt) * TW;
     p /= wt;
-    const int ho0 = (int)(p % ht) * TH;
-    const int n = (int)(p / ht);
+    const int ho = (int)(p % Ho);
+    const int n = (int)(p / Ho);
 
-    float acc[TH][TW][VEC];
+    float acc[TW][VEC];
 #pragma unroll
-    for (int th = 0; th < TH; ++th)
+    for (int t = 0; t < TW; ++t)
 #pragma unroll
-      for (int t = 0; t < TW; ++t)
-#pragma unroll
-        for (int i = 0; i < VEC; ++i) acc[th][t][i] = 0.f;
+      for (int i = 0; i < VEC; ++i) acc[t][i] = 0.f;
 
-    const int hi0 = ho0 - ph;
+    const int hi0 = ho - ph;
     const int wi0 = wo0 - pw;
     const bool interior = wi0 >= 0 && wi0 + K + TW - 2 < W;
 #pragma unroll
-    for (int row = 0; row < K + TH - 1; ++row) {
-      const int hi = hi0 + row;
+    for (int kh = 0; kh < K; ++kh) {
+      const int hi = hi0 + kh;
       if (hi < 0 || hi >= H) continue;
       const T* xrow = x + (((long long)n * H + hi) * W) * C + c;
+      const T* wrow = w + ((long long)kh * K) * C + c;
       if (interior) {
         // branchless: issue all K+TW-1 column loads, then consume (per-load
         // guards would serialize each load behind s_waitcnt vmcnt(0))
@@ -154,24 +154,16 @@ __global__ void dw_fwd_s1_kernel(
         for (int col = 0; col < K + TW - 1; ++col)
           xv[col] = vload<T, VEC>(xrow + (long long)(wi0 + col) * C);
 #pragma unroll
-        for (int th = 0; th < TH; ++th) {
-          const int kh = row - th;
-          if (kh < 0 || kh >= K) continue;
-          // weight loads stay per-use: the K*K tile is L1-hot, and holding
-          // it in registers (K*K*4 VGPRs at k5) spilled and ran 3x slower
-          const T* wrow = w + ((long long)kh * K) * C + c;
+        for (int col = 0; col < K + TW - 1; ++col) {
 #pragma unroll
-          for (int col = 0; col < K + TW - 1; ++col)
+          for (int t = 0; t < TW; ++t) {
+            const int kw = col - t;
+            if (kw < 0 || kw >= K) continue;
+            const TVec<T, VEC> wv = vload<T, VEC>(wrow + (long long)kw * C);
 #pragma unroll
-            for (int t = 0; t < TW; ++t) {
-              const int kw = col - t;
-              if (kw < 0 || kw >= K) continue;
-              const TVec<T, VEC> wv = vload<T, VEC>(wrow + (long long)kw * C);
-#pragma unroll
-              for (int i = 0; i < VEC; ++i)
-                acc[th][t][i] += DfdCvt<T>::to_f32(xv[col].v[i]) *
-                                 DfdCvt<T>::to_f32(wv.v[i]);
-            }
+            for (int i = 0; i < VEC; ++i)
+              acc[t][i] += DfdCvt<T>::to_f32(xv[col].v[i]) * DfdCvt<T>::to_f32(wv.v[i]);
+          }
         }
       } else {
 #pragma unroll
@@ -179,37 +171,27 @@ __global__ void dw_fwd_s1_kernel(
           const int wi = wi0 + col;
           if (wi < 0 || wi >= W) continue;
           const TVec<T, VEC> xv = vload<T, VEC>(xrow + (long long)wi * C);
+          // this column contributes to outputs t with 0 <= col - t < K
 #pragma unroll
-          for (int th = 0; th < TH; ++th) {
-            const int kh = row - th;
-            if (kh < 0 || kh >= K) continue;
+          for (int t = 0; t < TW; ++t) {
+            const int kw = col - t;
+            if (kw < 0 || kw >= K) continue;
+            const TVec<T, VEC> wv = vload<T, VEC>(wrow + (long long)kw * C);
 #pragma unroll
-            for (int t = 0; t < TW; ++t) {
-              const int kw = col - t;
-              if (kw < 0 || kw >= K) continue;
-              const TVec<T, VEC> wv =
-                  vload<T, VEC>(w + ((long long)kh * K + kw) * C + c);
-#pragma unroll
-              for (int i = 0; i < VEC; ++i)
-                acc[th][t][i] += DfdCvt<T>::to_f32(xv.v[i]) *
-                                 DfdCvt<T>::to_f32(wv.v[i]);
-            }
+            for (int i = 0; i < VEC; ++i)
+              acc[t][i] += DfdCvt<T>::to_f32(xv.v[i]) * DfdCvt<T>::to_f32(wv.v[i]);
           }
         }
       }
     }
+    T* yrow = y + (((long long)n * Ho + ho) * Wo) * C + c;
 #pragma unroll
-    for (int th = 0; th < TH; ++th) {
-      if (ho0 + th >= Ho) break;
-      T* yrow = y + (((long long)n * Ho + ho0 + th) * Wo) * C + c;
+    for (int t = 0; t < TW; ++t) {
+      if (wo0 + t >= Wo) break;
+      TVec<T, VEC> yv;
 #pragma unroll
-      for (int t = 0; t < TW; ++t) {
-        if (wo0 + t >= Wo) break;
-        TVec<T, VEC> yv;
-#pragma unroll
-        for (int i = 0; i < VEC; ++i) yv.v[i] = DfdCvt<T>::from_f32(acc[th][t][i]);
-        vstore<T, VEC>(yrow + (long long)(wo0 + t) * C, yv);
-      }
+      for (int i = 0; i < VEC; ++i) yv.v[i] = DfdCvt<T>::from_f32(acc[t][i]);
+      vstore<T, VEC>(yrow + (long long)(wo0 + t) * C, yv);
     }
   }
 }
@@ -566,17 +548,10 @@ void launch_fwd(const at::Tensor& x, const at::Tensor& w, at::Tensor& y, const G
                 hipStream_t stream) {
   const int block = 256;
   if (g.sh == 1 && g.sw == 1) {
-    // Per-K measured optimum (bs384 microbench, r02): k3 wants the 2-row
-    // tile (C=192: 0.695 ms vs 1.0+ at TH=1 — 104 VGPRs, 4 waves/SIMD);
-    // k5 wants max occupancy (TH=1, 82 VGPRs, 6 waves — its 2-row tile
-    // needs 172 VGPRs and ran 570 vs 433 us avg). Wider TW and pipelined
-    // row loads both re-triggered register hoists/spills and lost.
     constexpr int TW = 4;
-    constexpr int TH = K == 3 ? 2 : 1;
     const int wt = (g.Wo + TW - 1) / TW;
-    const int ht = (g.Ho + TH - 1) / TH;
-    const long long total = (long long)g.N * ht * wt * (g.C / VEC);
-    dw_fwd_s1_kernel<T, K, VEC, TW, TH><<<dfd_grid(total, block), block, 0, stream>>>(
+    const long long total = (long long)g.N * g.Ho * wt * (g.C / VEC);
+    dw_fwd_s1_kernel<T, K, VEC, TW><<<dfd_grid(total, block), block, 0, stream>>>(
         (const T*)x.data_ptr(), (const T*)w.data_ptr(), (T*)y.data_ptr(), g.N, g.C, g.H,
         g.W, g.Ho, g.Wo, g.ph, g.pw);
     return;
