@@ -397,8 +397,8 @@ __global__ __launch_bounds__(256) void pw_wgrad_kernel(
     const __hip_bfloat16* __restrict__ x,   // [M, K]
     float* __restrict__ part,               // [chunks, N, K]
     long long M, int N, int K, long long rows_per_chunk) {
-  __shared__ __bf16 dyt[2][WT * LDM];  // [n][m] transposed, double-buffered
-  __shared__ __bf16 xt[2][WT * LDM];    // [k][m] transposed
+  __shared__ __bf16 dyt[WT * LDM];  // [n][m] transposed
+  __shared__ __bf16 xt[WT * LDM];   // [k][m] transposed
 
   const int ktiles = (K + WT - 1) / WT;
   const int n0 = (blockIdx.x / ktiles) * WT;
@@ -424,57 +424,55 @@ __global__ __launch_bounds__(256) void pw_wgrad_kernel(
   const int sm = tid & 63;
   const int sq = tid >> 6;  // 0..3 -> 16-column group
 
-  // per-iteration staging registers (double-buffered pipeline: issue the
-  // NEXT m-tile's global loads while the MFMAs consume the current buffer)
-  auto load_mtile = [&](long long m0, bf16x8 (&dv)[2], bf16x8 (&xv)[2]) {
+  for (long long m0 = r0; m0 < r1; m0 += TM) {
     const long long gm = m0 + sm;
+    // dy tile: load vec8 along n, scatter-transpose into [n][m]
+    {
+      bf16x8 v[2];
 #pragma unroll
-    for (int h = 0; h < 2; ++h) {
-      const int n = n0 + sq * 16 + h * 8;
-      dv[h] = bf16x8{};
-      if (gm < r1 && (N & 7) == 0 && n + 7 < N) {
-        dv[h] = *reinterpret_cast<const bf16x8*>(dy + gm * N + n);
-      } else if (gm < r1) {
+      for (int h = 0; h < 2; ++h) {
+        const int n = n0 + sq * 16 + h * 8;
+        v[h] = bf16x8{};
+        if (gm < r1 && (N & 7) == 0 && n + 7 < N) {
+          v[h] = *reinterpret_cast<const bf16x8*>(dy + gm * N + n);
+        } else if (gm < r1) {
+#pragma unroll
+          for (int e = 0; e < 8; ++e)
+            reinterpret_cast<__bf16*>(&v[h])[e] =
+                (n + e) < N ? *reinterpret_cast<const __bf16*>(dy + gm * N + n + e)
+                            : (__bf16)0.f;
+        }
+      }
+#pragma unroll
+      for (int h = 0; h < 2; ++h)
 #pragma unroll
         for (int e = 0; e < 8; ++e)
-          reinterpret_cast<__bf16*>(&dv[h])[e] =
-              (n + e) < N ? *reinterpret_cast<const __bf16*>(dy + gm * N + n + e)
-                          : (__bf16)0.f;
-      }
-      const int k = k0 + sq * 16 + h * 8;
-      xv[h] = bf16x8{};
-      if (gm < r1 && (K & 7) == 0 && k + 7 < K) {
-        xv[h] = *reinterpret_cast<const bf16x8*>(x + gm * K + k);
-      } else if (gm < r1) {
-#pragma unroll
-        for (int e = 0; e < 8; ++e)
-          reinterpret_cast<__bf16*>(&xv[h])[e] =
-              (k + e) < K ? *reinterpret_cast<const __bf16*>(x + gm * K + k + e)
-                          : (__bf16)0.f;
-      }
+          dyt[(sq * 16 + h * 8 + e) * LDM + sm] = reinterpret_cast<__bf16*>(&v[h])[e];
     }
-  };
-  auto store_mtile = [&](int buf, const bf16x8 (&dv)[2], const bf16x8 (&xv)[2]) {
+    // x tile: same, into [k][m]
+    {
+      bf16x8 v[2];
 #pragma unroll
-    for (int h = 0; h < 2; ++h)
+      for (int h = 0; h < 2; ++h) {
+        const int k = k0 + sq * 16 + h * 8;
+        v[h] = bf16x8{};
+        if (gm < r1 && (K & 7) == 0 && k + 7 < K) {
+          v[h] = *reinterpret_cast<const bf16x8*>(x + gm * K + k);
+        } else if (gm < r1) {
 #pragma unroll
-      for (int e = 0; e < 8; ++e) {
-        dyt[buf][(sq * 16 + h * 8 + e) * LDM + sm] =
-            reinterpret_cast<const __bf16*>(&dv[h])[e];
-        xt[buf][(sq * 16 + h * 8 + e) * LDM + sm] =
-            reinterpret_cast<const __bf16*>(&xv[h])[e];
+          for (int e = 0; e < 8; ++e)
+            reinterpret_cast<__bf16*>(&v[h])[e] =
+                (k + e) < K ? *reinterpret_cast<const __bf16*>(x + gm * K + k + e)
+                            : (__bf16)0.f;
+        }
       }
-  };
-
-  bf16x8 dv[2], xv[2];
-  load_mtile(r0, dv, xv);
-  store_mtile(0, dv, xv);
-  __syncthreads();
-
-  int cur = 0;
-  for (long long m0 = r0; m0 < r1; m0 += TM, cur ^= 1) {
-    const bool has_next = m0 + TM < r1;
-    if (has_next) load_mtile(m0 + TM, dv, xv);
+#pragma unroll
+      for (int h = 0; h < 2; ++h)
+#pragma unroll
+        for (int e = 0; e < 8; ++e)
+          xt[(sq * 16 + h * 8 + e) * LDM + sm] = reinterpret_cast<__bf16*>(&v[h])[e];
+    }
+    __syncthreads();
 
     // 2 MFMA contraction steps of 32 m each
 #pragma unroll
@@ -483,11 +481,11 @@ __global__ __launch_bounds__(256) void pw_wgrad_kernel(
 #pragma unroll
       for (int i = 0; i < 2; ++i)
         afrag[i] = *reinterpret_cast<const bf16x8*>(
-            &dyt[cur][(wn + i * 16 + lrow) * LDM + s * 32 + lk]);
+            &dyt[(wn + i * 16 + lrow) * LDM + s * 32 + lk]);
 #pragma unroll
       for (int j = 0; j < 2; ++j)
         bfrag[j] = *reinterpret_cast<const bf16x8*>(
-            &xt[cur][(wk + j * 16 + lrow) * LDM + s * 32 + lk]);
+            &xt[(wk + j * 16 + lrow) * LDM + s * 32 + lk]);
 #pragma unroll
       for (int i = 0; i < 2; ++i)
 #pragma unroll
@@ -495,7 +493,6 @@ __global__ __launch_bounds__(256) void pw_wgrad_kernel(
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afrag[i], bfrag[j], acc[i][j], 0, 0, 0);
     }
-    if (has_next) store_mtile(cur ^ 1, dv, xv);
     __syncthreads();
   }
 
