@@ -45,16 +45,15 @@ DFD_DEV void svstore(T* p, const SVec<T, N>& x) {
 // ---------------------------------------------------------------------------
 template <typename T, int VEC>
 __global__ void se_pool_kernel(const T* __restrict__ x, float* __restrict__ s, int C,
-                               long long HW, int log2_cpb, int rows_per_chunk) {
+                               long long HW, int cpb, int rows_per_chunk) {
   extern __shared__ float lds[];  // [blockDim.x * VEC]
-  const int cpb = 1 << log2_cpb;
-  const int slot = threadIdx.x & (cpb - 1);
-  const int rg = threadIdx.x >> log2_cpb;
-  const int nrg = blockDim.x >> log2_cpb;
+  const int slot = threadIdx.x % cpb;
+  const int rg = threadIdx.x / cpb;
+  const int nrg = blockDim.x / cpb;
   const int cv = C / VEC;
   const int cvec = blockIdx.y * cpb + slot;
   const long long n = blockIdx.x;
-  const bool active = cvec < cv;
+  const bool active = cvec < cv && rg < nrg;  // tail threads must not stream
   const int c = cvec * VEC;
 
   float acc[VEC];
@@ -65,27 +64,48 @@ __global__ void se_pool_kernel(const T* __restrict__ x, float* __restrict__ s, i
     const T* xn = x + n * HW * C + c;
     const long long r0 = (long long)blockIdx.z * rows_per_chunk;
     const long long r1 = min(r0 + rows_per_chunk, HW);
-    for (long long r = r0 + rg; r < r1; r += nrg) {
+    long long r = r0 + rg;
+    for (; r + 3 * (long long)nrg < r1; r += 4 * (long long)nrg) {
+      SVec<T, VEC> xv[4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) xv[u] = svload<T, VEC>(xn + (r + u * (long long)nrg) * C);
+#pragma unroll
+      for (int u = 0; u < 4; ++u)
+#pragma unroll
+        for (int i = 0; i < VEC; ++i) acc[i] += DfdCvt<T>::to_f32(xv[u].v[i]);
+    }
+    for (; r < r1; r += nrg) {
       const SVec<T, VEC> xv = svload<T, VEC>(xn + r * C);
 #pragma unroll
       for (int i = 0; i < VEC; ++i) acc[i] += DfdCvt<T>::to_f32(xv.v[i]);
     }
   }
 
-  // cross-row-group tree reduction in LDS
-  float* my = lds + (size_t)threadIdx.x * VEC;
+  // cross-row-group tree reduction in LDS (non-pow2 nrg: fold excess first)
+  float* my = lds + (size_t)(rg * cpb + slot) * VEC;
+  int p2 = 1;
+  while (p2 * 2 <= nrg) p2 *= 2;
+  const bool in_block = rg < nrg;
+  if (in_block) {
 #pragma unroll
-  for (int i = 0; i < VEC; ++i) my[i] = acc[i];
+    for (int i = 0; i < VEC; ++i) my[i] = acc[i];
+  }
   __syncthreads();
-  for (int step = nrg >> 1; step > 0; step >>= 1) {
+  if (in_block && rg >= p2) {
+    float* dst = lds + (size_t)((rg - p2) * cpb + slot) * VEC;
+#pragma unroll
+    for (int i = 0; i < VEC; ++i) dst[i] += my[i];
+  }
+  __syncthreads();
+  for (int step = p2 >> 1; step > 0; step >>= 1) {
     if (rg < step) {
-      const float* other = lds + ((size_t)((rg + step) << log2_cpb) + slot) * VEC;
+      const float* other = lds + ((size_t)((rg + step) * cpb) + slot) * VEC;
 #pragma unroll
       for (int i = 0; i < VEC; ++i) my[i] += other[i];
     }
     __syncthreads();
   }
-  if (rg == 0 && active) {
+  if (rg == 0 && cvec < cv) {
 #pragma unroll
     for (int i = 0; i < VEC; ++i) atomicAdd(s + n * C + c + i, my[i]);
   }
@@ -166,16 +186,15 @@ template <typename T, int VEC>
 __global__ void se_bwd_reduce_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                                      const float* __restrict__ g, T* __restrict__ dx,
                                      float* __restrict__ dg, int C, long long HW,
-                                     int log2_cpb, int rows_per_chunk) {
+                                     int cpb, int rows_per_chunk) {
   extern __shared__ float lds[];
-  const int cpb = 1 << log2_cpb;
-  const int slot = threadIdx.x & (cpb - 1);
-  const int rg = threadIdx.x >> log2_cpb;
-  const int nrg = blockDim.x >> log2_cpb;
+  const int slot = threadIdx.x % cpb;
+  const int rg = threadIdx.x / cpb;
+  const int nrg = blockDim.x / cpb;
   const int cv = C / VEC;
   const int cvec = blockIdx.y * cpb + slot;
   const long long n = blockIdx.x;
-  const bool active = cvec < cv;
+  const bool active = cvec < cv && rg < nrg;
   const int c = cvec * VEC;
 
   float acc[VEC], gv[VEC];
@@ -190,7 +209,28 @@ __global__ void se_bwd_reduce_kernel(const T* __restrict__ dy, const T* __restri
     T* dxn = dx + n * HW * C + c;
     const long long r0 = (long long)blockIdx.z * rows_per_chunk;
     const long long r1 = min(r0 + rows_per_chunk, HW);
-    for (long long r = r0 + rg; r < r1; r += nrg) {
+    long long r = r0 + rg;
+    // 4-row batches: 8 independent loads in flight per iteration
+    for (; r + 3 * (long long)nrg < r1; r += 4 * (long long)nrg) {
+      SVec<T, VEC> xv[4], dv[4];
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        xv[u] = svload<T, VEC>(xn + (r + u * (long long)nrg) * C);
+        dv[u] = svload<T, VEC>(dn + (r + u * (long long)nrg) * C);
+      }
+#pragma unroll
+      for (int u = 0; u < 4; ++u) {
+        SVec<T, VEC> ov;
+#pragma unroll
+        for (int i = 0; i < VEC; ++i) {
+          const float d = DfdCvt<T>::to_f32(dv[u].v[i]);
+          acc[i] += d * DfdCvt<T>::to_f32(xv[u].v[i]);
+          ov.v[i] = DfdCvt<T>::from_f32(d * gv[i]);
+        }
+        svstore<T, VEC>(dxn + (r + u * (long long)nrg) * C, ov);
+      }
+    }
+    for (; r < r1; r += nrg) {
       const SVec<T, VEC> xv = svload<T, VEC>(xn + r * C);
       const SVec<T, VEC> dv = svload<T, VEC>(dn + r * C);
       SVec<T, VEC> ov;
@@ -204,19 +244,30 @@ __global__ void se_bwd_reduce_kernel(const T* __restrict__ dy, const T* __restri
     }
   }
 
-  float* my = lds + (size_t)threadIdx.x * VEC;
+  float* my = lds + (size_t)(rg * cpb + slot) * VEC;
+  int p2 = 1;
+  while (p2 * 2 <= nrg) p2 *= 2;
+  const bool in_block = rg < nrg;
+  if (in_block) {
 #pragma unroll
-  for (int i = 0; i < VEC; ++i) my[i] = acc[i];
+    for (int i = 0; i < VEC; ++i) my[i] = acc[i];
+  }
   __syncthreads();
-  for (int step = nrg >> 1; step > 0; step >>= 1) {
+  if (in_block && rg >= p2) {
+    float* dst = lds + (size_t)((rg - p2) * cpb + slot) * VEC;
+#pragma unroll
+    for (int i = 0; i < VEC; ++i) dst[i] += my[i];
+  }
+  __syncthreads();
+  for (int step = p2 >> 1; step > 0; step >>= 1) {
     if (rg < step) {
-      const float* other = lds + ((size_t)((rg + step) << log2_cpb) + slot) * VEC;
+      const float* other = lds + ((size_t)((rg + step) * cpb) + slot) * VEC;
 #pragma unroll
       for (int i = 0; i < VEC; ++i) my[i] += other[i];
     }
     __syncthreads();
   }
-  if (rg == 0 && active) {
+  if (rg == 0 && cvec < cv) {
 #pragma unroll
     for (int i = 0; i < VEC; ++i) atomicAdd(dg + n * C + c + i, my[i]);
   }
@@ -251,25 +302,25 @@ int se_pick_vec(long long c, int elem_size) {
   return 1;
 }
 
-// power-of-two channel slots per block, ≤64, ≥ enough to cover cv when small
-int pick_log2_cpb(int cv) {
-  int l = 0;
-  while ((1 << l) < cv && l < 6) ++l;
-  return l;
+// channel slots per block: exact (non-pow2) balanced tiles — rounding up to
+// a power of two idled up to 44% of a block's load bandwidth (C=336: 42 of
+// 64 slots active)
+int pick_cpb(int cv, int* ctiles_out) {
+  const int ctiles = (cv + 63) / 64;
+  *ctiles_out = ctiles;
+  return (cv + ctiles - 1) / ctiles;
 }
 
 struct ChunkPlan {
-  int log2_cpb, ctiles, chunks, rows_per_chunk;
+  int cpb, ctiles, chunks, rows_per_chunk;
 };
 
 ChunkPlan plan_chunks(long long N, int cv, long long HW) {
   ChunkPlan p;
-  p.log2_cpb = pick_log2_cpb(cv);
-  const int cpb = 1 << p.log2_cpb;
-  p.ctiles = (cv + cpb - 1) / cpb;
+  p.cpb = pick_cpb(cv, &p.ctiles);
   const long long base = N * p.ctiles;
   long long want = (2048 + base - 1) / base;  // chunks to reach ~2048 blocks
-  const int nrg = 256 >> p.log2_cpb;
+  const int nrg = 256 / p.cpb;
   // keep ≥~16 row-iterations per thread to amortize the per-block atomics
   long long by_iters = HW / ((long long)nrg * 16);
   if (want > by_iters) want = by_iters;
@@ -316,7 +367,7 @@ std::vector<at::Tensor> se_fwd(at::Tensor x, at::Tensor w1, at::Tensor b1,
 
 #define LAUNCH_POOL(T, V)                                                       \
   se_pool_kernel<T, V><<<pool_grid, 256, pool_lds, stream>>>(                   \
-      (const T*)x.data_ptr(), s.data_ptr<float>(), C, HW, plan.log2_cpb,        \
+      (const T*)x.data_ptr(), s.data_ptr<float>(), C, HW, plan.cpb,        \
       plan.rows_per_chunk)
 #define POOL_VEC(T)                                                             \
   switch (vec) {                                                                \
@@ -384,7 +435,7 @@ std::vector<at::Tensor> se_bwd_reduce(at::Tensor dy, at::Tensor x, at::Tensor g)
 #define LAUNCH(T, V)                                                            \
   se_bwd_reduce_kernel<T, V><<<grid, 256, lds, stream>>>(                       \
       (const T*)dy.data_ptr(), (const T*)x.data_ptr(), g.data_ptr<float>(),     \
-      (T*)dx.data_ptr(), dg.data_ptr<float>(), C, HW, plan.log2_cpb,            \
+      (T*)dx.data_ptr(), dg.data_ptr<float>(), C, HW, plan.cpb,            \
       plan.rows_per_chunk)
 #define LV(T)                                                                   \
   switch (vec) {                                                                \
