@@ -407,14 +407,13 @@ template <typename T, int K, int VEC>
 __global__ void dw_bwd_weight_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                                      float* __restrict__ dw, int N, int C, int H, int W,
                                      int Ho, int Wo, int sh, int sw, int ph, int pw,
-                                     int log2_cpb, int rows_per_chunk) {
+                                     int cpb, int rows_per_chunk) {
   extern __shared__ float lds[];  // [blockDim.x * VEC]
-  const int cpb = 1 << log2_cpb;
-  const int slot = threadIdx.x & (cpb - 1);
-  const int rest = threadIdx.x >> log2_cpb;
+  const int slot = threadIdx.x % cpb;
+  const int rest = threadIdx.x / cpb;
   const int kh = rest % K;
   const int rg = rest / K;
-  const int nrg = (blockDim.x >> log2_cpb) / K;
+  const int nrg = (blockDim.x / cpb) / K;
   const int cv = C / VEC;
   const int cvec = blockIdx.x * cpb + slot;
   // no early return: every thread reaches the barriers below
@@ -485,18 +484,20 @@ __global__ void dw_bwd_weight_kernel(const T* __restrict__ dy, const T* __restri
   // still issued K*K*C atomics × 2048 blocks (≈13M) and fp32 global-atomic
   // throughput capped the kernel (~1.2 ms on C=288 k5); partial stores +
   // a tiny second-stage reduction run at stream rate.
-  float* my = lds + (size_t)threadIdx.x * VEC;
+  float* my = lds + (size_t)((kh + K * rg) * cpb + slot) * VEC;
   const int pow2 = 1 << (31 - __clz(nrg > 0 ? nrg : 1));
   float* part = dw + (long long)blockIdx.y * K * K * C;  // dw is [chunks,K,K,C]
   for (int kw = 0; kw < K; ++kw) {
     __syncthreads();
+    if (rg < nrg) {
 #pragma unroll
-    for (int i = 0; i < VEC; ++i) my[i] = acc[kw * VEC + i];
+      for (int i = 0; i < VEC; ++i) my[i] = acc[kw * VEC + i];
+    }
     __syncthreads();
     for (int step = pow2; step > 0; step >>= 1) {
       if (rg < step && rg + step < nrg) {
         const float* other =
-            lds + ((size_t)(slot + ((kh + K * (rg + step)) << log2_cpb))) * VEC;
+            lds + ((size_t)((kh + K * (rg + step)) * cpb + slot)) * VEC;
 #pragma unroll
         for (int i = 0; i < VEC; ++i) my[i] += other[i];
       }
@@ -624,14 +625,14 @@ template <typename T, int K, int VEC>
 void launch_bwd_weight(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw,
                        const Geom& g, hipStream_t stream) {
   const int cv = g.C / VEC;
-  // power-of-two channel slots per block, ≤64, sized to cover cv when small
-  // and capped so at least one (kh, row-group) pair fits: cpb * K ≤ 256
-  int log2_cpb = 0;
-  while ((1 << log2_cpb) < cv && log2_cpb < 6) ++log2_cpb;
-  while (log2_cpb > 0 && (256 >> log2_cpb) / K < 1) --log2_cpb;
-  const int cpb = 1 << log2_cpb;
-  const int grid_x = (cv + cpb - 1) / cpb;
-  const int nrg = (256 >> log2_cpb) / K;  // row-groups per block
+  // exact (non-pow2) balanced channel slots per block, capped so at least
+  // one (kh, row-group) pair fits (cpb * K <= 256) — the pow2 rounding both
+  // idled slots AND forced extra channel tiles, each of which re-reads the
+  // WHOLE dy+x (C=336 k5: 2 tiles of 32 instead of 1 of 42)
+  const int cpb_cap = std::min(64, 256 / K);
+  const int grid_x = (cv + cpb_cap - 1) / cpb_cap;
+  const int cpb = (cv + grid_x - 1) / grid_x;
+  const int nrg = (256 / cpb) / K;  // row-groups per block
   const long long rows_total = (long long)g.N * g.Ho;
   // aim for ~2048 blocks total to fill 256 CUs, but keep ≥~4 row-iterations
   // per thread so per-block atomics amortize
@@ -650,7 +651,7 @@ void launch_bwd_weight(const at::Tensor& dy, const at::Tensor& x, at::Tensor& dw
                         dw.options());
   dw_bwd_weight_kernel<T, K, VEC><<<grid, 256, lds, stream>>>(
       (const T*)dy.data_ptr(), (const T*)x.data_ptr(), (float*)part.data_ptr(), g.N, g.C,
-      g.H, g.W, g.Ho, g.Wo, g.sh, g.sw, g.ph, g.pw, log2_cpb, rows_per_chunk);
+      g.H, g.W, g.Ho, g.Wo, g.sh, g.sw, g.ph, g.pw, cpb, rows_per_chunk);
   const int gx = dfd_grid(kkc, 256, 1 << 20);
   int zsplit = (int)std::min<long long>((2048 + gx - 1) / gx, (chunks + 7) / 8);
   if (zsplit < 1) zsplit = 1;
