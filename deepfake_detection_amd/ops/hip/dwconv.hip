@@ -363,31 +363,24 @@ __global__ void dw_bwd_data_kernel(const T* __restrict__ dy, const T* __restrict
 #pragma unroll
     for (int i = 0; i < VEC; ++i) acc[i] = 0.f;
 
-    // only taps with (hi+ph-kh) % sh == 0 contribute: start at the right
-    // parity and stride by sh instead of testing a modulo per tap (this
-    // kernel only serves strided convs — stride-1 bwd-data reuses the
-    // forward kernel with flipped weights)
-    const int kh0 = (hi + ph) % sh;
-    const int kw0 = (wi + pw) % sw;
-    for (int kh = kh0; kh < K; kh += sh) {
-      const int ho = (hi + ph - kh) / sh;
-      if (ho < 0 || ho >= Ho) continue;
-      const T* gRow = dy + (((long long)n * Ho + ho) * Wo) * C + c;
-      // issue the whole row of valid (dy, w) pairs, then consume: per-tap
-      // guarded loads serialize behind s_waitcnt
-      TVec<T, VEC> gv[(K + 1) / 2 + 1], wv[(K + 1) / 2 + 1];
-      int nt = 0;
-      for (int kw = kw0; kw < K; kw += sw) {
-        const int wo = (wi + pw - kw) / sw;
-        if (wo < 0 || wo >= Wo) continue;
-        gv[nt] = vload<T, VEC>(gRow + (long long)wo * C);
-        wv[nt] = vload<T, VEC>(w + ((long long)kh * K + kw) * C + c);
-        ++nt;
-      }
-      for (int t = 0; t < nt; ++t)
+#pragma unroll
+    for (int kh = 0; kh < K; ++kh) {
+      const int th = hi + ph - kh;
+      if (th < 0 || th % sh) continue;
+      const int ho = th / sh;
+      if (ho >= Ho) continue;
+#pragma unroll
+      for (int kw = 0; kw < K; ++kw) {
+        const int tw = wi + pw - kw;
+        if (tw < 0 || tw % sw) continue;
+        const int wo = tw / sw;
+        if (wo >= Wo) continue;
+        const TVec<T, VEC> gv = vload<T, VEC>(dy + (((long long)n * Ho + ho) * Wo + wo) * C + c);
+        const TVec<T, VEC> wv = vload<T, VEC>(w + ((long long)kh * K + kw) * C + c);
 #pragma unroll
         for (int i = 0; i < VEC; ++i)
-          acc[i] += DfdCvt<T>::to_f32(gv[t].v[i]) * DfdCvt<T>::to_f32(wv[t].v[i]);
+          acc[i] += DfdCvt<T>::to_f32(gv.v[i]) * DfdCvt<T>::to_f32(wv.v[i]);
+      }
     }
     TVec<T, VEC> ov;
 #pragma unroll
