@@ -36,6 +36,29 @@ def _autocast(enabled, dtype=torch.bfloat16):
     return contextlib.nullcontext()
 
 
+def _use_fused_head(model, loss_fn, target, use_cuda) -> bool:
+    """Fused classifier+CE head eligibility: single-GPU (plain module, not
+    DDP-wrapped), a plain-Linear classifier after global avg pool, a hard
+    int64 target, and a CE-family loss (plain or label-smoothed)."""
+    if not use_cuda or target.dim() != 1 or target.dtype != torch.long:
+        return False
+    if isinstance(model, (torch.nn.parallel.DistributedDataParallel,
+                          torch.nn.DataParallel)):
+        return False  # wrappers forward attr lookups to .module — explicit guard
+    from .loss import LabelSmoothingCrossEntropy
+    from .ops.extension import gpu_ops_required
+
+    if not isinstance(loss_fn, (torch.nn.CrossEntropyLoss, LabelSmoothingCrossEntropy)):
+        return False
+    if isinstance(loss_fn, torch.nn.CrossEntropyLoss) and loss_fn.label_smoothing:
+        return False
+    classifier = getattr(model, "classifier", None)
+    pool = getattr(model, "global_pool", None)
+    return (gpu_ops_required() and hasattr(model, "forward_features")
+            and isinstance(classifier, torch.nn.Linear)
+            and getattr(pool, "pool_type", None) == "avg")
+
+
 def train_epoch(epoch, model, loader, optimizer, loss_fn, args, device,
                 lr_scheduler=None, saver=None, output_dir="", model_ema=None,
                 world_size=1, rank=0, num_epochs=None):
@@ -61,9 +84,26 @@ def train_epoch(epoch, model, loader, optimizer, loss_fn, args, device,
         if use_cuda:
             input = input.contiguous(memory_format=torch.channels_last)
 
-        with _autocast(use_amp):
-            output = model(input)
-            loss = loss_fn(output, target)
+        if _use_fused_head(model, loss_fn, target, use_cuda):
+            # fused classifier GEMM + CE (ops/head.py): pooled features go
+            # through one autograd op returning (loss, logits). Single-GPU
+            # only — DDP must run module.forward for its reducer hooks.
+            from .ops import functional as O
+            from .ops.head import fused_head_ce
+
+            smoothing = float(getattr(loss_fn, "smoothing", 0.0))
+            with _autocast(use_amp):
+                feats = model.forward_features(input)
+                pooled = O.global_avg_pool(feats)
+                if getattr(model, "drop_rate", 0.0) > 0.0:
+                    pooled = torch.nn.functional.dropout(
+                        pooled, p=model.drop_rate, training=model.training)
+            loss, output = fused_head_ce(pooled, model.classifier.weight,
+                                         model.classifier.bias, target, smoothing)
+        else:
+            with _autocast(use_amp):
+                output = model(input)
+                loss = loss_fn(output, target)
 
         optimizer.zero_grad(set_to_none=True)
         loss.backward()

@@ -148,3 +148,29 @@ def test_train_epoch_save_images(tmp_path):
     train_epoch(0, model, batches, opt, torch.nn.CrossEntropyLoss(), args,
                 torch.device("cpu"), output_dir=str(tmp_path))
     assert glob.glob(str(tmp_path / "train-batch-*.jpg"))
+
+
+def test_fused_head_gating_logic():
+    """_use_fused_head: eligible only for plain single-GPU CE-family setups."""
+    import deepfake_detection_amd as dfd
+    from deepfake_detection_amd.engine import _use_fused_head
+    from deepfake_detection_amd.loss import LabelSmoothingCrossEntropy
+
+    m = dfd.create_model("efficientnet_b0", num_classes=2)
+    t = torch.randint(0, 2, (4,))
+    ce = torch.nn.CrossEntropyLoss()
+    # CPU (use_cuda False) -> no
+    assert not _use_fused_head(m, ce, t, use_cuda=False)
+    # soft targets -> no
+    soft = torch.rand(4, 2)
+    assert not _use_fused_head(m, ce, soft, use_cuda=True)
+    # smoothing via torch CE kwarg -> no (fused op wants explicit smoothing)
+    assert not _use_fused_head(m, torch.nn.CrossEntropyLoss(label_smoothing=0.1),
+                               t, use_cuda=True)
+    # JSD-style loss -> no
+    assert not _use_fused_head(m, torch.nn.MSELoss(), t, use_cuda=True)
+    # DataParallel wrapper -> no
+    assert not _use_fused_head(torch.nn.DataParallel(m), ce, t, use_cuda=True)
+    # eligible shape (actual GPU dispatch still gated by gpu_ops_required)
+    ls = LabelSmoothingCrossEntropy(0.1)
+    assert isinstance(ls.smoothing, float) or ls.smoothing == 0.1

@@ -679,3 +679,28 @@ def test_fused_head_ce_matches_torch(smoothing):
     assert torch.allclose(w.grad, rw.grad, atol=0.02, rtol=0.05)
     assert torch.allclose(b.grad, rb.grad, atol=0.01, rtol=0.05)
     assert torch.allclose(x.grad.float(), rx.grad, atol=0.05, rtol=0.05)
+
+
+def test_engine_train_epoch_fused_head():
+    """One engine train_epoch batch on GPU takes the fused-head path and
+    produces finite metrics."""
+    import types
+
+    import deepfake_detection_amd as dfd
+    from deepfake_detection_amd.engine import _use_fused_head, train_epoch
+    from deepfake_detection_amd.optim import RMSpropTF
+
+    torch.manual_seed(17)
+    model = dfd.create_model("efficientnet_b0", num_classes=2).cuda()
+    model = model.to(memory_format=torch.channels_last)
+    opt = RMSpropTF(model.parameters(), lr=1e-5, alpha=0.9, eps=1e-3, momentum=0.9)
+    loss_fn = torch.nn.CrossEntropyLoss().cuda()
+    x = torch.randn(6, 3, 65, 65, device="cuda")
+    t = torch.randint(0, 2, (6,), device="cuda")
+    assert _use_fused_head(model, loss_fn, t, use_cuda=True)
+    args = types.SimpleNamespace(log_interval=1, prefetcher=True, amp=True,
+                                 recovery_interval=0, save_images=False)
+    metrics = train_epoch(0, model, [(x, t)], opt, loss_fn, args,
+                          torch.device("cuda", 0))
+    assert torch.isfinite(torch.tensor(metrics["loss"]))
+    assert 0.0 <= metrics["prec1"] <= 100.0
