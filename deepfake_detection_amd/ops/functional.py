@@ -66,19 +66,34 @@ def bn_act(x, bn, act: str = "silu", residual=None, drop_path_mask=None):
     if x.is_cuda and gpu_ops_required() and act in _FUSED_ACTS:
         from .bn_act import fused_bn_act
 
-        # producer-fused BN stats: the conv kernel that wrote x may have
-        # attached bucketed per-channel (sum, sumsq) partials (ops/pwconv.py)
-        stats = None
-        attached = getattr(x, "_dfd_bn_stats", None)
-        if attached is not None and bn.training:
-            buckets, m, c = attached
-            if c == x.shape[1] and m == x.shape[0] * x.shape[2] * x.shape[3]:
-                stats = buckets
-        return fused_bn_act(
-            x, bn.weight, bn.bias, bn.running_mean, bn.running_var,
-            bn.training, bn.momentum, bn.eps, act, residual, stats,
-            drop_path_mask,
-        )
+        # the kernel requires fp32 BN params/stats. A .half()'ed model (the
+        # reference's model_half.pth.tar inference path, test.py:44-47)
+        # carries fp16 BN params: cast read-only copies for EVAL (no
+        # write-back of running stats); a halved model in TRAINING falls
+        # through to the torch path below.
+        w, b = bn.weight, bn.bias
+        rm, rv = bn.running_mean, bn.running_var
+        if w is not None and w.dtype != torch.float32:
+            if bn.training:
+                w = None  # sentinel: use the eager fallback
+            else:
+                w, b = w.float(), b.float()
+                rm = rm.float() if rm is not None else rm
+                rv = rv.float() if rv is not None else rv
+        if w is not None or bn.weight is None:
+            # producer-fused BN stats: the conv kernel that wrote x may have
+            # attached bucketed per-channel (sum, sumsq) partials
+            stats = None
+            attached = getattr(x, "_dfd_bn_stats", None)
+            if attached is not None and bn.training:
+                buckets, m, c = attached
+                if c == x.shape[1] and m == x.shape[0] * x.shape[2] * x.shape[3]:
+                    stats = buckets
+            return fused_bn_act(
+                x, w, b, rm, rv,
+                bn.training, bn.momentum, bn.eps, act, residual, stats,
+                drop_path_mask,
+            )
     y = F.batch_norm(
         x, bn.running_mean, bn.running_var, bn.weight, bn.bias,
         bn.training, bn.momentum if bn.momentum is not None else 0.1, bn.eps,

@@ -704,3 +704,24 @@ def test_engine_train_epoch_fused_head():
                           torch.device("cuda", 0))
     assert torch.isfinite(torch.tensor(metrics["loss"]))
     assert 0.0 <= metrics["prec1"] <= 100.0
+
+
+def test_half_model_inference_path():
+    """The reference's model_half.pth.tar path (.half() model, test.py:44-47):
+    fused BN must accept fp16 BN params in eval (fp32 read-only casts)."""
+    import deepfake_detection_amd as dfd
+
+    torch.manual_seed(18)
+    m = dfd.create_model("efficientnet_b0", num_classes=2).cuda()
+    ref = dfd.create_model("efficientnet_b0", num_classes=2)
+    ref.load_state_dict(m.state_dict())
+    m = m.half().eval().to(memory_format=torch.channels_last)
+    ref = ref.eval()
+    x = torch.randn(2, 3, 65, 65)
+    xg = x.cuda().half().contiguous(memory_format=torch.channels_last)
+    with torch.no_grad():
+        y = m(xg)
+        yr = ref(x)
+    assert torch.isfinite(y.float()).all()
+    assert torch.allclose(y.float().cpu(), yr, atol=0.1, rtol=0.05), (
+        (y.float().cpu() - yr).abs().max().item())
