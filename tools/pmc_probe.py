@@ -56,6 +56,29 @@ def pw_fwd(k, n, h, b=384, iters=5):
     print("pw_fwd", k, n, h)
 
 
+def pw_wgrad(k, n, h, b=384, iters=5):
+    from deepfake_detection_amd.ops.extension import load_extension
+
+    ext = load_extension()
+    x = cl(torch.randn(b, k, h, h, device="cuda", dtype=torch.bfloat16))
+    dy = cl(torch.randn(b, n, h, h, device="cuda", dtype=torch.bfloat16))
+    for _ in range(iters):
+        ext.pw_conv2d_bwd_weight_mfma(dy, x)
+    torch.cuda.synchronize()
+    print("pw_wgrad", k, n, h)
+
+
+def stem_fwd(cin, n, img, b=384, iters=5):
+    from deepfake_detection_amd.ops.stemconv import stem_conv2d
+
+    x = cl(torch.randn(b, cin, img, img, device="cuda", dtype=torch.bfloat16))
+    w = torch.randn(n, cin, 3, 3, device="cuda", dtype=torch.bfloat16)
+    for _ in range(iters):
+        stem_conv2d(x, w, stride=2, padding=1)
+    torch.cuda.synchronize()
+    print("stem_fwd", cin, n, img)
+
+
 CASES = {
     "dw_fwd_k5": lambda: dw_fwd(5, 336, 38),
     "dw_fwd_k3": lambda: dw_fwd(3, 192, 75),
@@ -63,6 +86,8 @@ CASES = {
     "bn_bwd_672": lambda: bn_bwd(672, 19),
     "pw_fwd_expand": lambda: pw_fwd(192, 288, 38),
     "pw_fwd_lowk": lambda: pw_fwd(24, 144, 150),
+    "pw_wgrad_tall": lambda: pw_wgrad(24, 144, 150),
+    "stem_fwd_b4": lambda: stem_fwd(3, 48, 299),
 }
 
 if __name__ == "__main__":
