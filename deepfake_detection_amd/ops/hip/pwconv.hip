@@ -623,11 +623,8 @@ at::Tensor pw_conv2d_bwd_weight_mfma(at::Tensor dy, at::Tensor x) {
   const int ntiles = (N + WT - 1) / WT;
   const int ktiles = (K + WT - 1) / WT;
   const long long kn = (long long)ntiles * ktiles;
-  // enough m-chunks to fill the chip: ~4096 blocks total (256 CUs hold 4
-  // blocks each at this LDS size; the old 2048 target left small-kn shapes
-  // at ~11 of 32 waves/CU — PMC showed wgrad 74.9% wave-parked), but >=8
-  // stage iterations per block
-  long long chunks = (2 * kMaxGrid) / kn;
+  // enough m-chunks to fill the chip, but >=8 stage iterations per block
+  long long chunks = kMaxGrid / kn;
   const long long max_chunks = (M + 8 * TM - 1) / (8 * TM);
   if (chunks > max_chunks) chunks = max_chunks;
   if (chunks < 1) chunks = 1;
