@@ -153,3 +153,53 @@ def test_model_shapes_tool():
     # BN modules never run their own forward on this stack (the fused
     # HIP/CPU path consumes their parameters directly), so no "bn" rows
     assert "bn" not in totals
+
+
+def test_stem_weight_pack_layout():
+    """stemconv weight repack: [N, Kpad] with k = (kh*KW + kw)*Cin + c,
+    zero-padded to a 32 multiple."""
+    import torch
+
+    from deepfake_detection_amd.ops.stemconv import _pack_weight
+
+    w = torch.arange(2 * 3 * 3 * 3, dtype=torch.float32).reshape(2, 3, 3, 3)
+    p = _pack_weight(w)
+    assert p.shape == (2, 32) and p.dtype == torch.bfloat16
+    # element (n=1, kh=2, kw=1, c=0) -> k = (2*3+1)*3 + 0 = 21
+    assert p[1, 21].item() == w[1, 0, 2, 1].item()
+    assert (p[:, 27:] == 0).all()
+
+
+def test_conv_dispatch_classes():
+    """create_conv2d routes 1x1 -> PointwiseConv2d, k3 s2 small-Cin ->
+    StemConv2d, depthwise -> DepthwiseConv2d, else nn.Conv2d; all share the
+    nn.Conv2d state_dict."""
+    import torch.nn as nn
+
+    from deepfake_detection_amd.models.layers import (DepthwiseConv2d,
+                                                      PointwiseConv2d,
+                                                      StemConv2d, create_conv2d)
+
+    assert type(create_conv2d(64, 128, 1)) is PointwiseConv2d
+    assert type(create_conv2d(3, 48, 3, stride=2)) is StemConv2d
+    assert type(create_conv2d(64, 64, 3, depthwise=True)) is DepthwiseConv2d
+    assert type(create_conv2d(64, 128, 3, stride=2)) is nn.Conv2d  # Cin > 16
+    assert type(create_conv2d(64, 128, 3)) is nn.Conv2d
+    for cls_conv in (create_conv2d(64, 128, 1), create_conv2d(3, 48, 3, stride=2)):
+        sd = cls_conv.state_dict()
+        assert set(sd) == {"weight"}
+
+
+def test_pw_dispatch_gating():
+    import torch
+
+    from deepfake_detection_amd.ops.pwconv import pw_supported, pw_use_mfma
+
+    assert pw_use_mfma(64, 128)
+    assert not pw_use_mfma(63, 128)  # not 8-aligned
+    w = torch.zeros(128, 64, 1, 1)
+    x = torch.zeros(2, 64, 8, 8, dtype=torch.bfloat16)
+    assert pw_supported(x, w, 1, 0, 1, 1)
+    assert not pw_supported(x.float(), w, 1, 0, 1, 1)  # fp32 -> torch path
+    assert not pw_supported(x, w, 2, 0, 1, 1)          # strided
+    assert not pw_supported(x, torch.zeros(128, 64, 3, 3), 1, 1, 1, 1)
