@@ -258,16 +258,17 @@ __global__ void bn_act_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
   const long long r0 = (long long)blockIdx.y * rows_per_chunk;
   const long long r1 = min(r0 + rows_per_chunk, M);
   long long r = r0 + rg;
-  // 4-row batches: 8 loads in flight per iteration (wait_any was 52% at 2)
-  for (; r + 3 * (long long)nrg < r1; r += 4 * (long long)nrg) {
-    BVec<T, VEC> xv[4], rv[4], yv[4];
+  // 8-row batches: up to 16 loads in flight per iteration (wait_any was
+  // 52% at 2-row, ~50% at 4-row)
+  for (; r + 7 * (long long)nrg < r1; r += 8 * (long long)nrg) {
+    BVec<T, VEC> xv[8], rv[8], yv[8];
 #pragma unroll
-    for (int u = 0; u < 4; ++u) {
+    for (int u = 0; u < 8; ++u) {
       xv[u] = bvload<T, VEC>(x + (r + u * (long long)nrg) * C + c);
       if (RES) rv[u] = bvload<T, VEC>(res + (r + u * (long long)nrg) * C + c);
     }
 #pragma unroll
-    for (int u = 0; u < 4; ++u) {
+    for (int u = 0; u < 8; ++u) {
       const float keep = dp ? dp[(r + u * (long long)nrg) / hw] : 1.f;
 #pragma unroll
       for (int j = 0; j < VEC; ++j) {
