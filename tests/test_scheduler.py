@@ -92,3 +92,39 @@ def test_lr_noise_deterministic_and_bounded():
     assert lrs1[0] == 1.0 and lrs1[1] == 1.0  # before the window: no noise
     assert any(lr != 1.0 for lr in lrs1[2:])  # inside: noise applied
     assert all(0.5 < lr < 1.5 for lr in lrs1)  # bounded (pct=0.1 normal)
+
+
+def test_cyclic_cosine_reference_semantics():
+    """Pin the refactored cycle machinery to hand-computed reference values
+    (reference cosine_lr.py): t_mul growth, per-cycle decay, cycle_limit."""
+    import math
+
+    opt = _opt(lr=1.0)
+    s = CosineLRScheduler(opt, t_initial=4, t_mul=2.0, lr_min=0.1,
+                          decay_rate=0.5, cycle_limit=2)
+    # t=2: cycle 0 (t_i=4, t_curr=2): 0.1 + 0.5*(1-0.1)*(1+cos(pi/2))
+    assert abs(s._get_lr(2)[0] - (0.1 + 0.45 * (1 + math.cos(math.pi / 2)))) < 1e-9
+    # t=6: cycle 1 starts at 4, t_i=8, t_curr=2, gamma=0.5
+    lo, hi = 0.05, 0.5
+    assert abs(s._get_lr(6)[0] - (lo + 0.5 * (hi - lo) * (1 + math.cos(math.pi * 2 / 8)))) < 1e-9
+    # t=13: cycle 2 >= cycle_limit -> flat lr_min
+    assert abs(s._get_lr(13)[0] - 0.1) < 1e-12
+
+
+def test_cyclic_tanh_reference_semantics():
+    """Tanh specifics: non-prefix warmup ramps toward the in-cycle LR at
+    warmup_t; exhausted cycles hold lr_min * decay^cycle_limit."""
+    import math
+
+    opt = _opt(lr=1.0)
+    s = TanhLRScheduler(opt, t_initial=10, lb=-6.0, ub=4.0, lr_min=0.01,
+                        decay_rate=0.5, warmup_t=2, warmup_lr_init=0.001,
+                        cycle_limit=1)
+    # warmup target = _get_lr(2) of the decay curve
+    tr = 2 / 10
+    want2 = 0.01 + 0.5 * (1.0 - 0.01) * (1 - math.tanh(-6.0 * (1 - tr) + 4.0 * tr))
+    assert abs(s._get_lr(2)[0] - want2) < 1e-9
+    # warmup is linear from warmup_lr_init to that target
+    assert abs(s._get_lr(1)[0] - (0.001 + (want2 - 0.001) / 2)) < 1e-9
+    # past cycle_limit: floor keeps the final decay applied
+    assert abs(s._get_lr(25)[0] - 0.01 * 0.5) < 1e-12
