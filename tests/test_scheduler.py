@@ -116,6 +116,8 @@ def test_cyclic_tanh_reference_semantics():
     warmup_t; exhausted cycles hold lr_min * decay^cycle_limit."""
     import math
 
+    from deepfake_detection_amd.scheduler import TanhLRScheduler
+
     opt = _opt(lr=1.0)
     s = TanhLRScheduler(opt, t_initial=10, lb=-6.0, ub=4.0, lr_min=0.01,
                         decay_rate=0.5, warmup_t=2, warmup_lr_init=0.001,
