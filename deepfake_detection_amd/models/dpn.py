@@ -72,30 +72,28 @@ class InputBlock(nn.Module):
 
 
 class DualPathBlock(nn.Module):
+    """Dual-path unit: a residual lane of width num_1x1_c plus a dense lane
+    growing by `inc` per block. State carries as the (residual, dense)
+    tuple; module attribute names match the reference state dict
+    (c1x1_w_s{1,2} shortcut, c1x1_a -> c3x3_b -> c1x1_c main path, with the
+    'b' variant splitting the tail into c1x1_c1/c1x1_c2)."""
+
     def __init__(self, in_chs, num_1x1_a, num_3x3_b, num_1x1_c, inc, groups,
                  block_type="normal", b=False):
         super().__init__()
+        assert block_type in ("proj", "down", "normal")
         self.num_1x1_c = num_1x1_c
         self.inc = inc
         self.b = b
-        if block_type == "proj":
-            self.key_stride = 1
-            self.has_proj = True
-        elif block_type == "down":
-            self.key_stride = 2
-            self.has_proj = True
-        else:
-            assert block_type == "normal"
-            self.key_stride = 1
-            self.has_proj = False
+        self.key_stride = 2 if block_type == "down" else 1
+        self.has_proj = block_type != "normal"
 
         if self.has_proj:
-            if self.key_stride == 2:
-                self.c1x1_w_s2 = BnActConv2d(
-                    in_chs=in_chs, out_chs=num_1x1_c + 2 * inc, kernel_size=1, stride=2)
-            else:
-                self.c1x1_w_s1 = BnActConv2d(
-                    in_chs=in_chs, out_chs=num_1x1_c + 2 * inc, kernel_size=1, stride=1)
+            # the shortcut projection produces both lanes at once
+            proj = BnActConv2d(in_chs=in_chs, out_chs=num_1x1_c + 2 * inc,
+                               kernel_size=1, stride=self.key_stride)
+            self.add_module("c1x1_w_s2" if self.key_stride == 2 else "c1x1_w_s1",
+                            proj)
         self.c1x1_a = BnActConv2d(in_chs=in_chs, out_chs=num_1x1_a, kernel_size=1, stride=1)
         self.c3x3_b = BnActConv2d(
             in_chs=num_1x1_a, out_chs=num_3x3_b, kernel_size=3,
@@ -108,31 +106,26 @@ class DualPathBlock(nn.Module):
             self.c1x1_c = BnActConv2d(
                 in_chs=num_3x3_b, out_chs=num_1x1_c + inc, kernel_size=1, stride=1)
 
+    def _shortcut(self, joined, lanes):
+        if not self.has_proj:
+            return lanes  # identity: lanes pass through unchanged
+        proj = getattr(self, "c1x1_w_s2" if self.key_stride == 2 else "c1x1_w_s1")
+        both = proj(joined)
+        return both[:, : self.num_1x1_c], both[:, self.num_1x1_c:]
+
     def forward(self, x):
-        x_in = torch.cat(x, dim=1) if isinstance(x, tuple) else x
-        if self.has_proj:
-            if self.key_stride == 2:
-                x_s = self.c1x1_w_s2(x_in)
-            else:
-                x_s = self.c1x1_w_s1(x_in)
-            x_s1 = x_s[:, :self.num_1x1_c, :, :]
-            x_s2 = x_s[:, self.num_1x1_c:, :, :]
-        else:
-            x_s1 = x[0]
-            x_s2 = x[1]
-        x_in = self.c1x1_a(x_in)
-        x_in = self.c3x3_b(x_in)
+        lanes = x if isinstance(x, tuple) else (x[:, : self.num_1x1_c], x[:, self.num_1x1_c:])
+        joined = torch.cat(x, dim=1) if isinstance(x, tuple) else x
+        short_res, short_dense = self._shortcut(joined, lanes)
+
+        mid = self.c3x3_b(self.c1x1_a(joined))
         if self.b:
-            x_in = self.c1x1_c(x_in)
-            out1 = self.c1x1_c1(x_in)
-            out2 = self.c1x1_c2(x_in)
+            mid = self.c1x1_c(mid)
+            main_res, main_dense = self.c1x1_c1(mid), self.c1x1_c2(mid)
         else:
-            x_in = self.c1x1_c(x_in)
-            out1 = x_in[:, :self.num_1x1_c, :, :]
-            out2 = x_in[:, self.num_1x1_c:, :, :]
-        resid = x_s1 + out1
-        dense = torch.cat([x_s2, out2], dim=1)
-        return resid, dense
+            tail = self.c1x1_c(mid)
+            main_res, main_dense = tail[:, : self.num_1x1_c], tail[:, self.num_1x1_c:]
+        return short_res + main_res, torch.cat([short_dense, main_dense], dim=1)
 
 
 class DPN(nn.Module):
