@@ -115,6 +115,9 @@ def _build_parser():
     parser.add_argument("--blur_radiu", type=int, default=1)
     parser.add_argument("--blur_prob", type=float, default=0)
     parser.add_argument("--color-jitter", type=float, default=0.2)
+    parser.add_argument("--aa", type=str, default=None, metavar="NAME",
+                        help="AutoAugment policy ('v0', 'original', "
+                             "'rand-m9-mstd0.5', 'augmix-m3-w3')")
     parser.add_argument("--aug-splits", type=int, default=0)
     parser.add_argument("--reprob", type=float, default=0.0)
     parser.add_argument("--remode", type=str, default="pixel")
@@ -295,6 +298,7 @@ def main(rank, args, args_text, world_size=None, start_rank=0):
             is_training=True, use_prefetcher=args.prefetcher, re_prob=args.reprob,
             re_mode=args.remode, re_count=args.recount, re_split=args.resplit,
             re_max=args.remax, color_jitter=args.color_jitter,
+            auto_augment=args.aa, num_aug_splits=num_aug_splits,
             mean=data_config["mean"], std=data_config["std"], num_workers=args.workers,
             distributed=args.distributed, pin_memory=args.pin_mem,
             fp16=False, dtype="bfloat16" if args.amp else "float32",
