@@ -748,3 +748,37 @@ def test_runner_main_synthetic_gpu(tmp_path, monkeypatch):
     assert "state_dict_ema" in ck
     for v in ck["state_dict"].values():
         assert torch.isfinite(v.float()).all()
+
+
+def test_prefetch_loader_v3_gpu_stream_path():
+    """PrefetchLoader_v3 on GPU: side-HIP-stream H2D + fused uint8->bf16
+    normalize + GPU RandomErasing produce a correctly normalized
+    channels_last batch (the production input path, reference
+    loader.py:213-289)."""
+    import torch.utils.data as tud
+
+    from deepfake_detection_amd.data.loader import PrefetchLoader_v3, fast_collate
+
+    torch.manual_seed(19)
+
+    class _DS(tud.Dataset):
+        def __len__(self):
+            return 8
+
+        def __getitem__(self, i):
+            import numpy as np
+
+            rng = np.random.RandomState(i)
+            return rng.randint(0, 256, (12, 33, 35), dtype=np.uint8), i % 2
+
+    dl = tud.DataLoader(_DS(), batch_size=4, collate_fn=fast_collate)
+    pf = PrefetchLoader_v3(dl, img_num=4, dtype=torch.bfloat16, re_prob=0.5)
+    batches = list(pf)
+    assert len(batches) == 2
+    x, t = batches[0]
+    assert x.is_cuda and x.dtype == torch.bfloat16
+    assert x.is_contiguous(memory_format=torch.channels_last)
+    assert x.shape == (4, 12, 33, 35)
+    # normalization: uint8 [0,255] -> roughly [-3, 3] after mean/std
+    assert x.float().abs().max().item() < 4.0
+    assert t.is_cuda
