@@ -340,6 +340,102 @@ __global__ void dw_fwd_s1_stats_kernel(
 }
 
 // ---------------------------------------------------------------------------
+// stride-1 forward with LDS-STAGED weights (BASELINE "next levers" item):
+// the whole (K, K, C) packed weight is copied into LDS once per block and
+// read with DYNAMIC offsets — the compiler can neither hoist it into
+// registers (the k5 spill trap) nor send the per-tap reads through the
+// vmem queue, which the x loads saturate (74% SQ_WAIT_ANY). Used when the
+// weight fits a fraction of LDS that keeps >=3 blocks/CU resident.
+// ---------------------------------------------------------------------------
+template <typename T, int K, int VEC, int TW>
+__global__ void dw_fwd_s1_ldsw_kernel(const T* __restrict__ x, const T* __restrict__ w,
+                                      T* __restrict__ y, int N, int C, int H, int W,
+                                      int Ho, int Wo, int ph, int pw) {
+  extern __shared__ unsigned char smem_raw[];
+  T* wl = reinterpret_cast<T*>(smem_raw);  // [K*K*C]
+  const int wtot = K * K * C;
+  for (int i = threadIdx.x * VEC; i < wtot; i += blockDim.x * VEC) {
+    if (i + VEC <= wtot)
+      *reinterpret_cast<TVec<T, VEC>*>(&wl[i]) = vload<T, VEC>(w + i);
+    else
+      for (int e = i; e < wtot; ++e) wl[e] = w[e];
+  }
+  __syncthreads();
+
+  const int cv = C / VEC;
+  const int wt = (Wo + TW - 1) / TW;
+  const long long total = (long long)N * Ho * wt * cv;
+  for (long long idx = (long long)blockIdx.x * blockDim.x + threadIdx.x; idx < total;
+       idx += (long long)gridDim.x * blockDim.x) {
+    const int c = (int)(idx % cv) * VEC;
+    long long p = idx / cv;
+    const int wo0 = (int)(p % wt) * TW;
+    p /= wt;
+    const int ho = (int)(p % Ho);
+    const int n = (int)(p / Ho);
+
+    float acc[TW][VEC];
+#pragma unroll
+    for (int t = 0; t < TW; ++t)
+#pragma unroll
+      for (int i = 0; i < VEC; ++i) acc[t][i] = 0.f;
+
+    const int hi0 = ho - ph;
+    const int wi0 = wo0 - pw;
+    const bool interior = wi0 >= 0 && wi0 + K + TW - 2 < W;
+#pragma unroll
+    for (int kh = 0; kh < K; ++kh) {
+      const int hi = hi0 + kh;
+      if (hi < 0 || hi >= H) continue;
+      const T* xrow = x + (((long long)n * H + hi) * W) * C + c;
+      const T* wrow = wl + (kh * K) * C + c;
+      if (interior) {
+        TVec<T, VEC> xv[K + TW - 1];
+#pragma unroll
+        for (int col = 0; col < K + TW - 1; ++col)
+          xv[col] = vload<T, VEC>(xrow + (long long)(wi0 + col) * C);
+#pragma unroll
+        for (int col = 0; col < K + TW - 1; ++col)
+#pragma unroll
+          for (int t = 0; t < TW; ++t) {
+            const int kw = col - t;
+            if (kw < 0 || kw >= K) continue;
+            const TVec<T, VEC> wv = vload<T, VEC>(wrow + kw * C);
+#pragma unroll
+            for (int i = 0; i < VEC; ++i)
+              acc[t][i] += DfdCvt<T>::to_f32(xv[col].v[i]) * DfdCvt<T>::to_f32(wv.v[i]);
+          }
+      } else {
+#pragma unroll
+        for (int col = 0; col < K + TW - 1; ++col) {
+          const int wi = wi0 + col;
+          if (wi < 0 || wi >= W) continue;
+          const TVec<T, VEC> xv = vload<T, VEC>(xrow + (long long)wi * C);
+#pragma unroll
+          for (int t = 0; t < TW; ++t) {
+            const int kw = col - t;
+            if (kw < 0 || kw >= K) continue;
+            const TVec<T, VEC> wv = vload<T, VEC>(wrow + kw * C);
+#pragma unroll
+            for (int i = 0; i < VEC; ++i)
+              acc[t][i] += DfdCvt<T>::to_f32(xv.v[i]) * DfdCvt<T>::to_f32(wv.v[i]);
+          }
+        }
+      }
+    }
+    T* yrow = y + (((long long)n * Ho + ho) * Wo) * C + c;
+#pragma unroll
+    for (int t = 0; t < TW; ++t) {
+      if (wo0 + t >= Wo) break;
+      TVec<T, VEC> yv;
+#pragma unroll
+      for (int i = 0; i < VEC; ++i) yv.v[i] = DfdCvt<T>::from_f32(acc[t][i]);
+      vstore<T, VEC>(yrow + (long long)(wo0 + t) * C, yv);
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
 // backward data:
 // dx[n,hi,wi,c] = sum over (kh,kw) with hi = ho*sh-ph+kh solvable:
 //                 dy[n,ho,wo,c] * w[kh,kw,c]
@@ -552,6 +648,14 @@ void launch_fwd(const at::Tensor& x, const at::Tensor& w, at::Tensor& y, const G
     constexpr int TW = 4;
     const int wt = (g.Wo + TW - 1) / TW;
     const long long total = (long long)g.N * g.Ho * wt * (g.C / VEC);
+    const int wbytes = K * K * g.C * (int)x.element_size();
+    if (wbytes <= 48 * 1024) {  // >=3 blocks/CU with the staged weight
+      dw_fwd_s1_ldsw_kernel<T, K, VEC, TW>
+          <<<dfd_grid(total, block), block, wbytes, stream>>>(
+              (const T*)x.data_ptr(), (const T*)w.data_ptr(), (T*)y.data_ptr(),
+              g.N, g.C, g.H, g.W, g.Ho, g.Wo, g.ph, g.pw);
+      return;
+    }
     dw_fwd_s1_kernel<T, K, VEC, TW><<<dfd_grid(total, block), block, 0, stream>>>(
         (const T*)x.data_ptr(), (const T*)w.data_ptr(), (T*)y.data_ptr(), g.N, g.C, g.H,
         g.W, g.Ho, g.Wo, g.ph, g.pw);
