@@ -649,7 +649,7 @@ void launch_fwd(const at::Tensor& x, const at::Tensor& w, at::Tensor& y, const G
     const int wt = (g.Wo + TW - 1) / TW;
     const long long total = (long long)g.N * g.Ho * wt * (g.C / VEC);
     const int wbytes = K * K * g.C * (int)x.element_size();
-    if (wbytes <= 48 * 1024) {  // >=3 blocks/CU with the staged weight
+    if (wbytes <= 50 * 1024) {  // >=3 blocks/CU with the staged weight
       dw_fwd_s1_ldsw_kernel<T, K, VEC, TW>
           <<<dfd_grid(total, block), block, wbytes, stream>>>(
               (const T*)x.data_ptr(), (const T*)w.data_ptr(), (T*)y.data_ptr(),
