@@ -55,10 +55,8 @@ class _DwConv2d(torch.autograd.Function):
         C, _, K, _ = weight.shape
         w_packed = weight.reshape(C, K, K).permute(1, 2, 0).contiguous()
         stats = None
-        if (stats_out is not None and K == 3 and sh == 1 and sw == 1
+        if (stats_out is not None and K in (3, 5) and sh == 1 and sw == 1
                 and dw_stats_enabled()):
-            # k3 s1 only: the stats variant fixes each thread's channel; at
-            # k5 the register cost measured slower than the pass it saves
             stats = torch.zeros(STATS_BUCKETS, 2, C, device=x.device,
                                 dtype=torch.float32)
             stats_out.append(stats)

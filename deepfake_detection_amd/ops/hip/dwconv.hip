@@ -213,7 +213,8 @@ __global__ void dw_fwd_s1_stats_kernel(
     float* __restrict__ stats,  // [kDwStatsBuckets, 2, C]
     int N, int C, int H, int W, int Ho, int Wo, int ph, int pw,
     int cpb, long long rows_per_chunk) {
-  extern __shared__ float lds[];  // [256 * VEC]
+  extern __shared__ float lds[];  // [256 * VEC] fp32 + K*K*cpb*VEC bf16 w slice
+  T* wl = reinterpret_cast<T*>(lds + 256 * VEC);
   const int slot = threadIdx.x % cpb;
   const int rg = threadIdx.x / cpb;
   const int nrg = blockDim.x / cpb;
@@ -223,6 +224,18 @@ __global__ void dw_fwd_s1_stats_kernel(
   const int c = cvec * VEC;
   const int wt = (Wo + TW - 1) / TW;
   const long long rows = (long long)N * Ho * wt;
+
+  // stage THIS block's channel slice of the weights: fixed-c threads would
+  // otherwise make the compiler hoist the K*K tile into registers (the
+  // occupancy loss measured 2x slower, r02); dynamic LDS reads stay cheap.
+  const int c0 = blockIdx.x * cpb * VEC;
+  const int cw = min(cpb * VEC, C - c0);  // slice width (elements)
+  for (int i = threadIdx.x; i < K * K * cw; i += blockDim.x) {
+    const int tap = i / cw;
+    const int cc = i - tap * cw;
+    wl[tap * (cpb * VEC) + cc] = w[(long long)tap * C + c0 + cc];
+  }
+  __syncthreads();
 
   float s[VEC], q[VEC];
 #pragma unroll
@@ -251,7 +264,7 @@ __global__ void dw_fwd_s1_stats_kernel(
         const int hi = hi0 + kh;
         if (hi < 0 || hi >= H) continue;
         const T* xrow = x + (((long long)n * H + hi) * W) * C + c;
-        const T* wrow = w + ((long long)kh * K) * C + c;
+        const T* wrow = wl + (kh * K) * (cpb * VEC) + slot * VEC;
         if (interior) {
           TVec<T, VEC> xv[K + TW - 1];
 #pragma unroll
@@ -263,7 +276,7 @@ __global__ void dw_fwd_s1_stats_kernel(
             for (int t = 0; t < TW; ++t) {
               const int kw = col - t;
               if (kw < 0 || kw >= K) continue;
-              const TVec<T, VEC> wv = vload<T, VEC>(wrow + (long long)kw * C);
+              const TVec<T, VEC> wv = vload<T, VEC>(wrow + kw * (cpb * VEC));
 #pragma unroll
               for (int i = 0; i < VEC; ++i)
                 acc[t][i] += DfdCvt<T>::to_f32(xv[col].v[i]) * DfdCvt<T>::to_f32(wv.v[i]);
@@ -278,7 +291,7 @@ __global__ void dw_fwd_s1_stats_kernel(
             for (int t = 0; t < TW; ++t) {
               const int kw = col - t;
               if (kw < 0 || kw >= K) continue;
-              const TVec<T, VEC> wv = vload<T, VEC>(wrow + (long long)kw * C);
+              const TVec<T, VEC> wv = vload<T, VEC>(wrow + kw * (cpb * VEC));
 #pragma unroll
               for (int i = 0; i < VEC; ++i)
                 acc[t][i] += DfdCvt<T>::to_f32(xv.v[i]) * DfdCvt<T>::to_f32(wv.v[i]);
@@ -809,7 +822,7 @@ Geom make_geom(int64_t N, int64_t C, int64_t H, int64_t W, int64_t K, int64_t sh
 }  // namespace
 
 // x: NCHW logical, channels_last physical. w_packed: (K, K, C) same dtype.
-template <typename T>
+template <typename T, int K>
 void launch_fwd_s1_stats(const at::Tensor& x, const at::Tensor& w, at::Tensor& y,
                          at::Tensor& stats, const Geom& g, int vec, hipStream_t stream) {
   constexpr int TW = 4;
@@ -818,9 +831,11 @@ void launch_fwd_s1_stats(const at::Tensor& x, const at::Tensor& w, at::Tensor& y
   const int cv = g.C / vec;
   DfdPlan plan = dfd_plan(cv, rows);
   dim3 grid(plan.ctiles, plan.chunks);
-  const int lds = 256 * vec * sizeof(float);
+  // fp32 stats scratch + this block's bf16/fp16/fp32 weight slice
+  const int lds = 256 * vec * sizeof(float) +
+                  K * K * plan.cpb * vec * (int)x.element_size();
 #define DW_STATS_V(V)                                                          \
-  dw_fwd_s1_stats_kernel<T, 3, V, TW><<<grid, 256, lds, stream>>>(             \
+  dw_fwd_s1_stats_kernel<T, K, V, TW><<<grid, 256, lds, stream>>>(             \
       (const T*)x.data_ptr(), (const T*)w.data_ptr(), (T*)y.data_ptr(),        \
       stats.data_ptr<float>(), g.N, g.C, g.H, g.W, g.Ho, g.Wo, g.ph, g.pw,     \
       plan.cpb, plan.rows_per_chunk)
@@ -850,19 +865,25 @@ at::Tensor dw_conv2d_fwd(at::Tensor x, at::Tensor w_packed, int64_t sh, int64_t 
   const int Ki = (int)K;
   if (stats_opt.has_value()) {
     at::Tensor stats = *stats_opt;
-    TORCH_CHECK(Ki == 3 && sh == 1 && sw == 1,
-                "dwconv: stats epilogue only for k3 stride-1");
+    TORCH_CHECK((Ki == 3 || Ki == 5) && sh == 1 && sw == 1,
+                "dwconv: stats epilogue only for k3/k5 stride-1");
     TORCH_CHECK(stats.scalar_type() == at::kFloat && stats.is_contiguous() &&
                     stats.numel() == (long long)kDwStatsBuckets * 2 * g.C,
                 "dwconv: stats must be fp32 [64, 2, C]");
+#define DW_STATS_T(T)                                                          \
+    do {                                                                       \
+      if (Ki == 3)                                                             \
+        launch_fwd_s1_stats<T, 3>(x, w_packed, y, stats, g, vec, stream);      \
+      else                                                                     \
+        launch_fwd_s1_stats<T, 5>(x, w_packed, y, stats, g, vec, stream);      \
+    } while (0)
     switch (stype) {
-      case at::kBFloat16:
-        launch_fwd_s1_stats<__hip_bfloat16>(x, w_packed, y, stats, g, vec, stream);
-        break;
-      case at::kHalf: launch_fwd_s1_stats<__half>(x, w_packed, y, stats, g, vec, stream); break;
-      case at::kFloat: launch_fwd_s1_stats<float>(x, w_packed, y, stats, g, vec, stream); break;
+      case at::kBFloat16: DW_STATS_T(__hip_bfloat16); break;
+      case at::kHalf: DW_STATS_T(__half); break;
+      case at::kFloat: DW_STATS_T(float); break;
       default: TORCH_CHECK(false, "dwconv: unsupported dtype");
     }
+#undef DW_STATS_T
     return y;
   }
   switch (stype) {
