@@ -21,11 +21,13 @@ _SUPPORTED_K = (3, 5, 7, 9, 11)
 
 
 def dw_stats_enabled() -> bool:
-    """The k3 stats-epilogue variant is OFF by default: its fixed-channel
-    layout costs ~4 waves of occupancy (1451 vs 702 us on the bs768 k3
-    layers — r02 profile) and the dw kernels are latency-bound, so it loses
-    more than the separate bn_stats pass it saves. Kept behind
-    DFD_AMD_DW_STATS=1 for re-evaluation after a register-pressure fix."""
+    """The dw stats-epilogue variant stays OFF by default. Two designs were
+    measured (r02): fixed-channel threads with per-use weight loads (the
+    compiler hoists the K*K tile -> occupancy loss, 2x slower) and with an
+    LDS-staged weight slice (hoist gone, but the channel-blocked spatial
+    striding itself loses the wo-tile x locality: k5 C=336 1.89 vs 0.51 ms,
+    whole bench 2609 vs 2752 img/s) — both lose more than the bn_stats pass
+    they save. Kept correct and GPU-tested behind DFD_AMD_DW_STATS=1."""
     return os.environ.get("DFD_AMD_DW_STATS", "0") == "1"
 
 

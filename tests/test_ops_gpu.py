@@ -595,10 +595,14 @@ def test_dwconv_stats_epilogue_matches_direct(monkeypatch):
                           atol=0.5, rtol=1e-3)
     assert torch.allclose(buckets[:, 1].sum(0), (ys * ys).sum(dim=(0, 2, 3)),
                           atol=1.0, rtol=1e-3)
-    # k5 must NOT emit (excluded by design)
+    # k5 variant (LDS-staged weight slice) must also be numerically correct
     w5 = torch.randn(C, 1, 5, 5, device="cuda", dtype=torch.bfloat16)
     y5 = dw_conv2d(x, w5, None, 1, 2, 1, want_stats=True)
-    assert not hasattr(y5, "_dfd_bn_stats")
+    y5p = dw_conv2d(x, w5, None, 1, 2, 1)
+    assert torch.equal(y5, y5p)
+    b5, m5, c5 = y5._dfd_bn_stats
+    assert torch.allclose(b5[:, 0].sum(0), y5.float().sum(dim=(0, 2, 3)),
+                          atol=0.5, rtol=1e-3)
 
 
 @pytest.mark.parametrize("cin,n,img", [(3, 48, 61), (12, 256, 38)])
