@@ -73,14 +73,18 @@ def bn_act(x, bn, act: str = "silu", residual=None, drop_path_mask=None):
         # through to the torch path below.
         w, b = bn.weight, bn.bias
         rm, rv = bn.running_mean, bn.running_var
-        if w is not None and w.dtype != torch.float32:
+        halved = any(t is not None and t.dtype != torch.float32
+                     for t in (w, rm, rv))
+        use_fused = True
+        if halved:
             if bn.training:
-                w = None  # sentinel: use the eager fallback
+                use_fused = False  # halved training: eager fallback below
             else:
-                w, b = w.float(), b.float()
+                w = w.float() if w is not None else w
+                b = b.float() if b is not None else b
                 rm = rm.float() if rm is not None else rm
                 rv = rv.float() if rv is not None else rv
-        if w is not None or bn.weight is None:
+        if use_fused:
             # producer-fused BN stats: the conv kernel that wrote x may have
             # attached bucketed per-channel (sum, sumsq) partials
             stats = None
