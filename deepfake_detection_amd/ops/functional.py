@@ -8,7 +8,6 @@ arrives and the extension is missing, we raise (see ops/extension.py).
 import torch
 import torch.nn.functional as F
 
-from . import reference
 from .extension import gpu_ops_required
 
 _FUSED_ACTS = ("none", "relu", "silu")
