@@ -7,7 +7,6 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops import functional as O
 from .layers import SelectAdaptivePool2d
 from .registry import register_model
 

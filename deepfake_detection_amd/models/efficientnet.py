@@ -15,7 +15,6 @@ efficientnet_deepfake_v4: 62,373,826 params, blocks/stage [4,7,7,10,10,13,4]
 (SURVEY.md §2.3).
 """
 
-import torch
 import torch.nn as nn
 import torch.nn.functional as F
 

@@ -8,7 +8,6 @@ Fused* optimizers (:42-91) are replaced by our own fused HIP multi-tensor
 paths inside RMSpropTF / AdamW (always on for ROCm tensors).
 """
 
-import torch.nn as nn
 import torch.optim as optim
 
 from .adamw import AdamW
